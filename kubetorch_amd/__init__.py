@@ -1,0 +1,56 @@
+"""kubetorch_amd — an MI355X-native serverless ML dispatch framework.
+
+A from-scratch framework with the capabilities of run-house/kubetorch,
+re-designed for AMD Instinct MI355X (gfx950): PyTorch-ROCm + hand-written
+HIP/CDNA4 kernels for the compute path, RCCL over xGMI for collectives,
+amd.com/gpu scheduling for the control plane.
+
+Public API mirrors the reference's `kt` namespace (kt.fn, kt.cls,
+kt.Compute, ...; reference: python_client/kubetorch/__init__.py).
+Heavy client-side modules are imported lazily so that the training path
+(ops/models/parallel) does not pull in the control-plane stack.
+"""
+
+__version__ = "0.1.0"
+
+_LAZY = {
+    "fn": ("kubetorch_amd.client.fn", "fn"),
+    "cls": ("kubetorch_amd.client.cls", "cls"),
+    "app": ("kubetorch_amd.client.app", "app"),
+    "Compute": ("kubetorch_amd.compute.compute", "Compute"),
+    "Image": ("kubetorch_amd.compute.image", "Image"),
+    "Volume": ("kubetorch_amd.compute.volume", "Volume"),
+    "Secret": ("kubetorch_amd.compute.secret", "Secret"),
+    "Endpoint": ("kubetorch_amd.compute.endpoint", "Endpoint"),
+    "AutoscalingConfig": ("kubetorch_amd.compute.autoscaling", "AutoscalingConfig"),
+    "images": ("kubetorch_amd.compute.images", None),
+    "compute": ("kubetorch_amd.compute.decorators", "compute"),
+    "distribute": ("kubetorch_amd.compute.decorators", "distribute"),
+    "autoscale": ("kubetorch_amd.compute.decorators", "autoscale"),
+    "async_": ("kubetorch_amd.compute.decorators", "async_"),
+    "put": ("kubetorch_amd.data_store.commands", "put"),
+    "get": ("kubetorch_amd.data_store.commands", "get"),
+    "ls": ("kubetorch_amd.data_store.commands", "ls"),
+    "rm": ("kubetorch_amd.data_store.commands", "rm"),
+    "BroadcastWindow": ("kubetorch_amd.data_store.types", "BroadcastWindow"),
+    "pod_ips": ("kubetorch_amd.serving.discovery", "pod_ips"),
+    "ops": ("kubetorch_amd.ops", None),
+    "models": ("kubetorch_amd.models", None),
+    "parallel": ("kubetorch_amd.parallel", None),
+}
+
+
+def __getattr__(name):
+    if name in _LAZY:
+        import importlib
+
+        mod_name, attr = _LAZY[name]
+        mod = importlib.import_module(mod_name)
+        obj = mod if attr is None else getattr(mod, attr)
+        globals()[name] = obj
+        return obj
+    raise AttributeError(f"module 'kubetorch_amd' has no attribute {name!r}")
+
+
+def __dir__():
+    return sorted(list(globals()) + list(_LAZY))

@@ -1,0 +1,164 @@
+"""Llama-3 family, MI355X-first.
+
+The flagship model for bench.py (Llama-3-8B DDP bf16 tokens/sec — the
+BASELINE.json north-star config). Hot memory-bound ops run the gfx950 HIP
+kernels from kubetorch_amd.ops (RMSNorm, RoPE, SwiGLU, fused CE); GEMMs go
+to hipBLASLt via torch.nn.Linear; attention uses torch SDPA (flash) on ROCm.
+
+Reference parity note: the reference (run-house/kubetorch) ships no model
+code at all — it launches user training scripts. This
+module is the MI355X-native flagship workload those launchers run.
+"""
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from kubetorch_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    intermediate: int = 14336
+    vocab_size: int = 128256
+    max_seq_len: int = 8192
+    rope_base: float = 500000.0
+    norm_eps: float = 1e-5
+    init_std: float = 0.02
+
+    @property
+    def head_dim(self):
+        return self.dim // self.n_heads
+
+
+def llama3_8b(**overrides) -> LlamaConfig:
+    return LlamaConfig(**overrides)
+
+
+def llama_tiny(**overrides) -> LlamaConfig:
+    """Small config for tests / smoke (runs on CPU)."""
+    d = dict(dim=256, n_layers=2, n_heads=4, n_kv_heads=2, intermediate=512,
+             vocab_size=512, max_seq_len=256)
+    d.update(overrides)
+    return LlamaConfig(**d)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim, eps):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.rmsnorm(x, self.weight, self.eps)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        d, hd = cfg.dim, cfg.head_dim
+        self.n_heads, self.n_kv = cfg.n_heads, cfg.n_kv_heads
+        self.wqkv = nn.Linear(d, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
+        self.wo = nn.Linear(cfg.n_heads * hd, d, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, S, _ = x.shape
+        hd = self.cfg.head_dim
+        qkv = self.wqkv(x)
+        q, k, v = qkv.split(
+            [self.n_heads * hd, self.n_kv * hd, self.n_kv * hd], dim=-1
+        )
+        q = ops.rope(q.view(B, S, self.n_heads, hd), cos, sin)
+        k = ops.rope(k.view(B, S, self.n_kv, hd), cos, sin)
+        v = v.view(B, S, self.n_kv, hd)
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        if self.n_kv != self.n_heads:
+            try:
+                o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+            except (TypeError, RuntimeError):
+                rep = self.n_heads // self.n_kv
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+                o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        else:
+            o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        o = o.transpose(1, 2).reshape(B, S, -1)
+        return self.wo(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.w_gate_up = nn.Linear(cfg.dim, 2 * cfg.intermediate, bias=False)
+        self.w_down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
+
+    def forward(self, x):
+        return self.w_down(ops.swiglu(self.w_gate_up(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.attn = Attention(cfg)
+        self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.mlp = MLP(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        x = x + self.mlp(self.mlp_norm(x))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        cos, sin = ops.precompute_rope(cfg.max_seq_len, cfg.head_dim, cfg.rope_base)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.reset_parameters()
+
+    def _apply(self, fn, recurse=True):
+        # Keep the RoPE tables fp32 across model.to(bf16): recompute rather
+        # than round-trip through bf16 (precision).
+        ret = super()._apply(fn, recurse)
+        if self.rope_cos.dtype != torch.float32:
+            cos, sin = ops.precompute_rope(
+                self.cfg.max_seq_len, self.cfg.head_dim, self.cfg.rope_base,
+                device=self.rope_cos.device,
+            )
+            self.rope_cos, self.rope_sin = cos, sin
+        return ret
+
+    def reset_parameters(self):
+        std = self.cfg.init_std
+        for m in self.modules():
+            if isinstance(m, (nn.Linear, nn.Embedding)):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+            elif isinstance(m, RMSNorm):
+                nn.init.ones_(m.weight)
+
+    def forward(self, tokens):
+        S = tokens.shape[1]
+        x = self.embed(tokens)
+        cos = self.rope_cos[:S]
+        sin = self.rope_sin[:S]
+        for layer in self.layers:
+            x = layer(x, cos, sin)
+        return self.lm_head(self.norm(x))
+
+    def loss(self, tokens, targets):
+        """Forward + fused CE (logits buffer is consumed by the fused op)."""
+        logits = self.forward(tokens)
+        return ops.fused_cross_entropy(logits, targets)

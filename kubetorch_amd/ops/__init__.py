@@ -1,0 +1,242 @@
+"""kubetorch_amd.ops — MI355X-native fused ops with autograd.
+
+Dispatch policy:
+  * GPU tensors run the hand-written gfx950 HIP kernels (_hip_ops.so,
+    built in-tree by kubetorch_amd.ops.build). If the extension is missing
+    on a GPU box the ops raise — there is no silent eager fallback.
+  * CPU tensors run a plain fp32 PyTorch reference of the same op; this is
+    the numerics reference the GPU tests compare against, and what CPU-only
+    CI exercises.
+"""
+import importlib.util
+import os
+
+import torch
+
+_EXT = None
+_EXT_ERR = None
+
+
+def _ext():
+    """Load the in-tree HIP extension, raising loudly if unavailable."""
+    global _EXT, _EXT_ERR
+    if _EXT is not None:
+        return _EXT
+    if _EXT_ERR is not None:
+        raise _EXT_ERR
+    so = os.path.join(os.path.dirname(__file__), "_hip_ops.so")
+    if not os.path.exists(so):
+        _EXT_ERR = RuntimeError(
+            f"kubetorch_amd HIP extension not built: {so} missing. "
+            "Run `python -m kubetorch_amd.ops.build` (gfx950)."
+        )
+        raise _EXT_ERR
+    spec = importlib.util.spec_from_file_location("kubetorch_amd.ops._hip_ops", so)
+    mod = importlib.util.module_from_spec(spec)
+    try:
+        spec.loader.exec_module(mod)
+    except Exception as e:  # pragma: no cover
+        _EXT_ERR = RuntimeError(f"failed to load {so}: {e}")
+        raise _EXT_ERR
+    _EXT = mod
+    return _EXT
+
+
+def hip_available():
+    try:
+        _ext()
+        return True
+    except Exception:
+        return False
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+def _rmsnorm_ref_fwd(x, w, eps):
+    xf = x.float()
+    invrms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    y = (xf * invrms * w.float()).to(x.dtype)
+    return y, invrms.squeeze(-1).reshape(-1)
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if x.is_cuda:
+            y, invrms = _ext().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+        else:
+            y, invrms = _rmsnorm_ref_fwd(x, w, eps)
+        ctx.save_for_backward(x, w, invrms)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, invrms = ctx.saved_tensors
+        if x.is_cuda:
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), x.contiguous(),
+                                        w.contiguous(), invrms)
+        else:
+            H = x.shape[-1]
+            xf = x.float().reshape(-1, H)
+            dyf = dy.float().reshape(-1, H)
+            wf = w.float()
+            ir = invrms.reshape(-1, 1)
+            S = (dyf * wf * xf).sum(-1, keepdim=True)
+            dx = (ir * dyf * wf - xf * (ir ** 3) * S / H).to(x.dtype).reshape(x.shape)
+            dw = (dyf * xf * ir).sum(0).to(w.dtype)
+        return dx, dw, None
+
+
+def rmsnorm(x, w, eps=1e-5):
+    """y = x * rsqrt(mean(x^2, -1) + eps) * w (bf16 in/out, fp32 accum)."""
+    return _RMSNorm.apply(x, w, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE (Llama rotate-half)
+# ---------------------------------------------------------------------------
+def precompute_rope(seq_len, head_dim, base=500000.0, device=None):
+    """cos/sin tables [S, D/2] fp32 (host-side trig per CDNA guide App. B)."""
+    inv_freq = 1.0 / (
+        base ** (torch.arange(0, head_dim, 2, dtype=torch.float32, device=device) / head_dim)
+    )
+    t = torch.arange(seq_len, dtype=torch.float32, device=device)
+    freqs = torch.outer(t, inv_freq)  # [S, D/2]
+    return freqs.cos().contiguous(), freqs.sin().contiguous()
+
+
+def _rope_ref(x, cos, sin, sign):
+    # x: [B, S, Hh, D]
+    B, S, Hh, D = x.shape
+    xf = x.float()
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
+    c = cos[:S].view(1, S, 1, D // 2)
+    s = sin[:S].view(1, S, 1, D // 2) * sign
+    o1 = x1 * c - x2 * s
+    o2 = x2 * c + x1 * s
+    return torch.cat([o1, o2], dim=-1).to(x.dtype)
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        ctx.S = x.shape[1]
+        ctx.save_for_backward(cos, sin)
+        if x.is_cuda:
+            B, S, Hh, D = x.shape
+            return _ext().rope(x.contiguous().view(B * S, Hh, D), cos, sin, S, 1.0).view(x.shape)
+        return _rope_ref(x, cos, sin, 1.0)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        if dy.is_cuda:
+            B, S, Hh, D = dy.shape
+            dx = _ext().rope(dy.contiguous().view(B * S, Hh, D), cos, sin, S, -1.0).view(dy.shape)
+        else:
+            dx = _rope_ref(dy, cos, sin, -1.0)
+        return dx, None, None
+
+
+def rope(x, cos, sin):
+    """Apply rotate-half RoPE to x: [B, S, Hh, D] using [S, D/2] tables."""
+    return _Rope.apply(x, cos, sin)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+def _swiglu_ref_fwd(gu):
+    I = gu.shape[-1] // 2
+    g, u = gu[..., :I].float(), gu[..., I:].float()
+    return (torch.nn.functional.silu(g) * u).to(gu.dtype)
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        ctx.save_for_backward(gu)
+        if gu.is_cuda:
+            return _ext().swiglu_fwd(gu.contiguous())
+        return _swiglu_ref_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (gu,) = ctx.saved_tensors
+        if gu.is_cuda:
+            return _ext().swiglu_bwd(dout.contiguous(), gu.contiguous())
+        I = gu.shape[-1] // 2
+        g, u = gu[..., :I].float(), gu[..., I:].float()
+        do = dout.float()
+        sig = torch.sigmoid(g)
+        dg = do * u * sig * (1 + g * (1 - sig))
+        du = do * g * sig
+        return torch.cat([dg, du], dim=-1).to(gu.dtype)
+
+
+def swiglu(gate_up):
+    """out = silu(gate) * up with gate_up = [..., 2I] packed from one GEMM."""
+    return _SwiGLU.apply(gate_up)
+
+
+# ---------------------------------------------------------------------------
+# Fused cross entropy (destroys logits: overwritten with the gradient)
+# ---------------------------------------------------------------------------
+class _FusedCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        V = logits.shape[-1]
+        flat = logits.reshape(-1, V)
+        t = targets.reshape(-1)
+        n_valid = (t != ignore_index).sum().clamp(min=1)
+        if logits.is_cuda:
+            loss = _ext().cross_entropy_fwd_(flat, t.contiguous(), 1.0, ignore_index)
+            dlogits = flat  # overwritten in place by the kernel
+        else:
+            lf = flat.float()
+            logp = torch.log_softmax(lf, dim=-1)
+            loss = torch.nn.functional.nll_loss(
+                logp, t, reduction="none", ignore_index=ignore_index
+            )
+            dlogits = torch.softmax(lf, dim=-1)
+            valid = (t != ignore_index).unsqueeze(-1)
+            dlogits.scatter_add_(
+                1, t.clamp(min=0).unsqueeze(1), -torch.ones_like(t, dtype=torch.float32).unsqueeze(1)
+            )
+            dlogits = torch.where(valid, dlogits, torch.zeros_like(dlogits)).to(logits.dtype)
+        ctx.save_for_backward(dlogits, n_valid)
+        ctx.shape = logits.shape
+        return loss.sum() / n_valid.to(loss.dtype)
+
+    @staticmethod
+    def backward(ctx, gout):
+        dlogits, n_valid = ctx.saved_tensors
+        scale = (gout / n_valid).to(torch.float32)
+        return (dlogits * scale).to(dlogits.dtype).reshape(ctx.shape), None, None
+
+
+def fused_cross_entropy(logits, targets, ignore_index=-100):
+    """Mean CE over valid tokens. WARNING: logits buffer is overwritten with
+    the (unscaled) gradient on GPU — do not reuse logits after this call."""
+    return _FusedCE.apply(logits, targets, ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW on flat bf16 buckets (used by kubetorch_amd.parallel)
+# ---------------------------------------------------------------------------
+def adamw_(p, g, m, v, lr, beta1, beta2, eps, wd, step, grad_scale=1.0):
+    if p.is_cuda:
+        _ext().adamw_(p, g, m, v, lr, beta1, beta2, eps, wd, step, grad_scale)
+        return
+    # CPU reference (fp32 math, bf16 params)
+    gf = g.float() * grad_scale
+    pf = p.float()
+    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    mhat = m / bc1
+    vhat = v / bc2
+    pf = pf - lr * (mhat / (vhat.sqrt() + eps) + wd * pf)
+    p.copy_(pf.to(p.dtype))

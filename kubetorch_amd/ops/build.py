@@ -1,0 +1,112 @@
+"""In-tree hipcc build of the kubetorch_amd HIP extension (gfx950 only).
+
+Produces kubetorch_amd/ops/_hip_ops.so next to this file so the built
+artifact travels with the repo snapshot to GPU boxes (no JIT cache).
+
+Invoked by __graft_entry__.build() and by `python -m kubetorch_amd.ops.build`.
+"""
+import os
+import subprocess
+import sys
+import sysconfig
+
+OPS_DIR = os.path.dirname(os.path.abspath(__file__))
+HIP_DIR = os.path.join(OPS_DIR, "hip")
+BUILD_DIR = os.path.join(OPS_DIR, "_build")
+SO_PATH = os.path.join(OPS_DIR, "_hip_ops.so")
+ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+
+SOURCES = ["kernels.hip", "bindings.cpp"]
+
+
+def _torch_paths():
+    import torch
+
+    troot = os.path.dirname(torch.__file__)
+    includes = [
+        os.path.join(troot, "include"),
+        os.path.join(troot, "include", "torch", "csrc", "api", "include"),
+        sysconfig.get_paths()["include"],
+    ]
+    libdir = os.path.join(troot, "lib")
+    return includes, libdir
+
+
+def _common_flags(includes):
+    flags = [
+        f"--offload-arch={ARCH}",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-fno-gpu-rdc",
+        "-DUSE_ROCM=1",
+        "-DHIPBLAS_V2",
+        "-DCUDA_HAS_FP16=1",
+        "-D__HIP_NO_HALF_OPERATORS__=1",
+        "-D__HIP_NO_HALF_CONVERSIONS__=1",
+        "-DTORCH_API_INCLUDE_EXTENSION_H",
+        "-DTORCH_EXTENSION_NAME=_hip_ops",
+    ]
+    import torch
+
+    flags.append(
+        "-D_GLIBCXX_USE_CXX11_ABI=" + ("1" if torch._C._GLIBCXX_USE_CXX11_ABI else "0")
+    )
+    for inc in includes:
+        flags += ["-I", inc]
+    return flags
+
+
+def _needs_rebuild():
+    if not os.path.exists(SO_PATH):
+        return True
+    so_mtime = os.path.getmtime(SO_PATH)
+    for src in SOURCES + ["build.py"]:
+        p = os.path.join(HIP_DIR, src) if src != "build.py" else os.path.join(OPS_DIR, src)
+        if os.path.getmtime(p) > so_mtime:
+            return True
+    return False
+
+
+def build(verbose=True, force=False):
+    """Compile kernels.hip + bindings.cpp into _hip_ops.so (idempotent)."""
+    if not force and not _needs_rebuild():
+        if verbose:
+            print(f"[kubetorch_amd.ops.build] up to date: {SO_PATH}")
+        return SO_PATH
+    os.makedirs(BUILD_DIR, exist_ok=True)
+    includes, libdir = _torch_paths()
+    flags = _common_flags(includes)
+    hipcc = os.environ.get("HIPCC", "hipcc")
+    objs = []
+    for src in SOURCES:
+        obj = os.path.join(BUILD_DIR, os.path.splitext(src)[0] + ".o")
+        cmd = [hipcc, "-c", os.path.join(HIP_DIR, src), "-o", obj] + flags
+        if verbose:
+            print("[hipcc]", " ".join(cmd))
+        subprocess.run(cmd, check=True)
+        objs.append(obj)
+    link = (
+        [hipcc, "-shared", "-fPIC", "-o", SO_PATH]
+        + objs
+        + [
+            f"-L{libdir}",
+            f"-Wl,-rpath,{libdir}",
+            "-ltorch",
+            "-ltorch_hip",
+            "-ltorch_python",
+            "-lc10",
+            "-lc10_hip",
+            "-lamdhip64",
+        ]
+    )
+    if verbose:
+        print("[hipcc link]", " ".join(link))
+    subprocess.run(link, check=True)
+    if verbose:
+        print(f"[kubetorch_amd.ops.build] built {SO_PATH}")
+    return SO_PATH
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)

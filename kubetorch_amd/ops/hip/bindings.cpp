@@ -1,0 +1,165 @@
+// Torch bindings for the kubetorch_amd MI355X kernels (kernels.hip).
+// Host-only translation unit; compiled by hipcc against torch-ROCm's
+// native HIP API surface (c10/hip). No CUDA-compat shims.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
+extern "C" {
+void kt_rmsnorm_fwd(const void* x, const void* w, void* y, void* invrms,
+                    int N, int H, float eps, hipStream_t stream);
+void kt_rmsnorm_bwd(const void* dy, const void* x, const void* w,
+                    const void* invrms, void* dx, void* dw_partial, void* dw,
+                    int P, int N, int H, hipStream_t stream);
+void kt_rope(const void* x, void* o, const void* cost, const void* sint,
+             long total_quads, int S, int Hh, int D, float sign,
+             hipStream_t stream);
+void kt_swiglu_fwd(const void* gu, void* out, long N, int I,
+                   hipStream_t stream);
+void kt_swiglu_bwd(const void* dout, const void* gu, void* dgu, long N, int I,
+                   hipStream_t stream);
+void kt_cross_entropy_fwd(void* logits, const void* targets, void* loss,
+                          int N, int V, float scale, long ignore_index,
+                          hipStream_t stream);
+void kt_adamw(void* p, const void* g, void* m, void* v, long n, float lr,
+              float beta1, float beta2, float eps, float wd, float bc1,
+              float bc2, float grad_scale, hipStream_t stream);
+}
+
+namespace {
+
+#define CHECK_BF16_CONTIG(t)                                         \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                  \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+hipStream_t cur_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w,
+                                    double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  TORCH_CHECK(w.numel() == H, "weight shape mismatch");
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({N}, x.options().dtype(at::kFloat));
+  kt_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), invrms.data_ptr(),
+                 (int)N, H, (float)eps, cur_stream(x));
+  return {y, invrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                    const at::Tensor& w,
+                                    const at::Tensor& invrms) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto dx = at::empty_like(x);
+  auto dw = at::empty_like(w);
+  const int P = N < 512 ? (int)N : 512;
+  auto dw_partial = at::empty({P, H}, x.options().dtype(at::kFloat));
+  kt_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(), invrms.data_ptr(),
+                 dx.data_ptr(), dw_partial.data_ptr(), dw.data_ptr(), P,
+                 (int)N, H, cur_stream(x));
+  return {dx, dw};
+}
+
+at::Tensor rope(const at::Tensor& x, const at::Tensor& cost,
+                const at::Tensor& sint, int64_t S, double sign) {
+  // x: [T, Hh, D] with T = B*S; cos/sin: [S, D/2] fp32
+  CHECK_BF16_CONTIG(x);
+  TORCH_CHECK(x.dim() == 3, "rope expects [T, Hh, D]");
+  TORCH_CHECK(cost.scalar_type() == at::kFloat && cost.is_contiguous());
+  const int D = (int)x.size(2);
+  const int Hh = (int)x.size(1);
+  TORCH_CHECK(D % 8 == 0, "head dim must be a multiple of 8");
+  TORCH_CHECK(cost.size(0) >= S && cost.size(1) == D / 2, "cos table shape");
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto o = at::empty_like(x);
+  const long total_quads = (long)x.size(0) * Hh * (D / 8);
+  kt_rope(x.data_ptr(), o.data_ptr(), cost.data_ptr(), sint.data_ptr(),
+          total_quads, (int)S, Hh, D, (float)sign, cur_stream(x));
+  return o;
+}
+
+at::Tensor swiglu_fwd(const at::Tensor& gu) {
+  CHECK_BF16_CONTIG(gu);
+  const int twoI = (int)gu.size(-1);
+  TORCH_CHECK(twoI % 16 == 0, "2*I must be a multiple of 16");
+  const int I = twoI / 2;
+  const long N = gu.numel() / twoI;
+  c10::hip::OptionalHIPGuard guard(gu.device());
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = at::empty(sizes, gu.options());
+  kt_swiglu_fwd(gu.data_ptr(), out.data_ptr(), N, I, cur_stream(gu));
+  return out;
+}
+
+at::Tensor swiglu_bwd(const at::Tensor& dout, const at::Tensor& gu) {
+  CHECK_BF16_CONTIG(dout);
+  CHECK_BF16_CONTIG(gu);
+  const int twoI = (int)gu.size(-1);
+  const int I = twoI / 2;
+  const long N = gu.numel() / twoI;
+  c10::hip::OptionalHIPGuard guard(gu.device());
+  auto dgu = at::empty_like(gu);
+  kt_swiglu_bwd(dout.data_ptr(), gu.data_ptr(), dgu.data_ptr(), N, I,
+                cur_stream(gu));
+  return dgu;
+}
+
+at::Tensor cross_entropy_fwd_(at::Tensor logits, const at::Tensor& targets,
+                              double scale, int64_t ignore_index) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(targets.scalar_type() == at::kLong && targets.is_contiguous());
+  const int V = (int)logits.size(-1);
+  const long N = logits.numel() / V;
+  TORCH_CHECK(V % 8 == 0, "V must be a multiple of 8");
+  TORCH_CHECK(targets.numel() == N, "targets shape mismatch");
+  c10::hip::OptionalHIPGuard guard(logits.device());
+  auto loss = at::empty({N}, logits.options().dtype(at::kFloat));
+  kt_cross_entropy_fwd(logits.data_ptr(), targets.data_ptr(), loss.data_ptr(),
+                       (int)N, V, (float)scale, ignore_index,
+                       cur_stream(logits));
+  return loss;
+}
+
+void adamw_(at::Tensor p, const at::Tensor& g, at::Tensor m, at::Tensor v,
+            double lr, double beta1, double beta2, double eps, double wd,
+            int64_t step, double grad_scale) {
+  CHECK_BF16_CONTIG(p);
+  CHECK_BF16_CONTIG(g);
+  TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() &&
+              p.numel() == v.numel());
+  c10::hip::OptionalHIPGuard guard(p.device());
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  kt_adamw(p.data_ptr(), g.data_ptr(), m.data_ptr(), v.data_ptr(), p.numel(),
+           (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, bc1,
+           bc2, (float)grad_scale, cur_stream(p));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16)");
+  mod.def("rope", &rope, "RoPE rotate-half (bf16), sign=+1 fwd / -1 bwd");
+  mod.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (bf16)");
+  mod.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (bf16)");
+  mod.def("cross_entropy_fwd_", &cross_entropy_fwd_,
+          "Fused CE: returns per-row loss, overwrites logits with grad");
+  mod.def("adamw_", &adamw_, "Fused AdamW on a flat bf16 bucket");
+}

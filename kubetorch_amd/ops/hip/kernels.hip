@@ -1,0 +1,482 @@
+// MI355X (gfx950 / CDNA4) kernels for the kubetorch_amd compute path.
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  - wave = 64 lanes; block sizes are multiples of 64 (256 default).
+//  - all bf16 global traffic is vectorized as ushort8 (16 B/lane) --
+//    hipcc does NOT auto-vectorize scalar bf16 loads (Guideline 13).
+//  - memory-bound kernels cap the grid at ~2048 blocks and grid-stride.
+//  - reductions: wave shuffle (width 64) -> LDS across waves.
+//  - RoPE uses a host-precomputed cos/sin table (no device trig).
+//
+// These kernels replace the multi-kernel eager-PyTorch sequences for the
+// hot memory-bound ops of the Llama training step (RMSNorm, RoPE, SwiGLU,
+// fused cross-entropy, fused AdamW). GEMMs stay on hipBLASLt via torch.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef unsigned short u16;
+typedef ushort vec8u __attribute__((ext_vector_type(8)));
+typedef ushort vec4u __attribute__((ext_vector_type(4)));
+typedef float vec4f __attribute__((ext_vector_type(4)));
+
+#define WAVE 64
+
+__device__ __forceinline__ float bf2f(u16 x) {
+  unsigned int u = ((unsigned int)x) << 16;
+  return __uint_as_float(u);
+}
+__device__ __forceinline__ u16 f2bf(float f) {
+  // round-to-nearest-even bf16 conversion
+  unsigned int u = __float_as_uint(f);
+  unsigned int rounding = 0x7fff + ((u >> 16) & 1);
+  u += rounding;
+  return (u16)(u >> 16);
+}
+
+// Block-wide sum; every thread returns the total. Reusable across
+// iterations (trailing barrier protects the LDS scratch).
+__device__ __forceinline__ float block_reduce_sum(float v) {
+  __shared__ float s[16];
+#pragma unroll
+  for (int o = 32; o > 0; o >>= 1) v += __shfl_down(v, o, WAVE);
+  const int wid = threadIdx.x >> 6;
+  const int nw = blockDim.x >> 6;
+  if ((threadIdx.x & 63) == 0) s[wid] = v;
+  __syncthreads();
+  float t = 0.f;
+  for (int i = 0; i < nw; ++i) t += s[i];
+  __syncthreads();
+  return t;
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm forward: y = x * rsqrt(mean(x^2) + eps) * w ; saves invrms for bwd.
+// One block per row (grid-stride over rows). x re-read in pass 2 (row is hot
+// in L2; register-caching with runtime-indexed arrays would spill to scratch).
+// ---------------------------------------------------------------------------
+__global__ void rmsnorm_fwd_kernel(const u16* __restrict__ x,
+                                   const u16* __restrict__ w,
+                                   u16* __restrict__ y,
+                                   float* __restrict__ invrms,
+                                   int N, int H, float eps) {
+  const int nvec = H >> 3;
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const u16* xr = x + (size_t)row * H;
+    u16* yr = y + (size_t)row * H;
+    float ss = 0.f;
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(xv[j]);
+        ss += f * f;
+      }
+    }
+    ss = block_reduce_sum(ss);
+    const float ir = rsqrtf(ss / (float)H + eps);
+    if (threadIdx.x == 0 && invrms) invrms[row] = ir;
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+      vec8u wv = *reinterpret_cast<const vec8u*>(w + vI * 8);
+      vec8u ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = f2bf(bf2f(xv[j]) * ir * bf2f(wv[j]));
+      *reinterpret_cast<vec8u*>(yr + vI * 8) = ov;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm backward.
+//   dx_i = ir*(dy_i*w_i) - x_i * ir^3/H * S,  S = sum_j dy_j*w_j*x_j
+//   dw_j = sum_rows dy_j * x_j * ir   (accumulated per-block in LDS, written
+//          to a [gridDim.x, H] fp32 partial buffer; reduced by colsum below)
+// LDS budget: H fp32 <= 160KB -> H <= 40960 (Llama H=4096 -> 16 KB). Fine.
+// ---------------------------------------------------------------------------
+__global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
+                                   const u16* __restrict__ x,
+                                   const u16* __restrict__ w,
+                                   const float* __restrict__ invrms,
+                                   u16* __restrict__ dx,
+                                   float* __restrict__ dw_partial,
+                                   int N, int H) {
+  extern __shared__ float dwacc[];  // [H]
+  const int nvec = H >> 3;
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[i + j] = 0.f;
+  }
+  __syncthreads();
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const u16* dyr = dy + (size_t)row * H;
+    const u16* xr = x + (size_t)row * H;
+    u16* dxr = dx + (size_t)row * H;
+    const float ir = invrms[row];
+    float S = 0.f;
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u dyv = *reinterpret_cast<const vec8u*>(dyr + vI * 8);
+      vec8u wv = *reinterpret_cast<const vec8u*>(w + vI * 8);
+      vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) S += bf2f(dyv[j]) * bf2f(wv[j]) * bf2f(xv[j]);
+    }
+    S = block_reduce_sum(S);
+    const float k = ir * ir * ir * S / (float)H;
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u dyv = *reinterpret_cast<const vec8u*>(dyr + vI * 8);
+      vec8u wv = *reinterpret_cast<const vec8u*>(w + vI * 8);
+      vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+      vec8u ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float dyf = bf2f(dyv[j]);
+        float xf = bf2f(xv[j]);
+        ov[j] = f2bf(ir * dyf * bf2f(wv[j]) - xf * k);
+        dwacc[vI * 8 + j] += dyf * xf * ir;  // thread owns slot: no conflict
+      }
+      *reinterpret_cast<vec8u*>(dxr + vI * 8) = ov;
+    }
+    __syncthreads();
+  }
+  float* out = dw_partial + (size_t)blockIdx.x * H;
+  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[i + j] = dwacc[i + j];
+  }
+}
+
+// Column-sum of the [P, H] fp32 partial buffer -> dw [H] bf16.
+__global__ void colsum_bf16_kernel(const float* __restrict__ partial,
+                                   u16* __restrict__ out, int P, int H) {
+  int col = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (col >= H) return;
+  float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+  for (int p = 0; p < P; ++p) {
+    vec4f v = *reinterpret_cast<const vec4f*>(partial + (size_t)p * H + col);
+    acc0 += v[0]; acc1 += v[1]; acc2 += v[2]; acc3 += v[3];
+  }
+  vec4u o;
+  o[0] = f2bf(acc0); o[1] = f2bf(acc1); o[2] = f2bf(acc2); o[3] = f2bf(acc3);
+  *reinterpret_cast<vec4u*>(out + col) = o;
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (Llama rotate-half), in-place capable, fwd/bwd via sign.
+// x: [T, Hh, D] bf16 where T = B*S tokens (row-major), cos/sin: [S, D/2] fp32
+// host-precomputed (Appendix B: no on-device trig). Each thread rotates 4
+// pairs: 8 B from each half of the head dim.
+//   fwd: o1 = x1*c - x2*s ; o2 = x2*c + x1*s      (sign=+1)
+//   bwd: o1 = x1*c + x2*s ; o2 = x2*c - x1*s      (sign=-1)
+// ---------------------------------------------------------------------------
+__global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
+                            const float* __restrict__ cost,
+                            const float* __restrict__ sint,
+                            long total_quads, int S, int Hh, int D,
+                            float sign) {
+  const int half = D >> 1;
+  const int quads_per_head = D >> 3;  // 4 pairs per quad
+  const long quads_per_tok = (long)Hh * quads_per_head;
+  for (long q = (long)blockIdx.x * blockDim.x + threadIdx.x; q < total_quads;
+       q += (long)gridDim.x * blockDim.x) {
+    const long tok = q / quads_per_tok;
+    const int rem = (int)(q - tok * quads_per_tok);
+    const int h = rem / quads_per_head;
+    const int qi = rem - h * quads_per_head;
+    const int pos = (int)(tok % S);
+    const size_t base = ((size_t)tok * Hh + h) * D + qi * 4;
+    vec4u x1 = *reinterpret_cast<const vec4u*>(x + base);
+    vec4u x2 = *reinterpret_cast<const vec4u*>(x + base + half);
+    vec4f c = *reinterpret_cast<const vec4f*>(cost + (size_t)pos * half + qi * 4);
+    vec4f s = *reinterpret_cast<const vec4f*>(sint + (size_t)pos * half + qi * 4);
+    vec4u o1, o2;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float a = bf2f(x1[j]), b = bf2f(x2[j]);
+      o1[j] = f2bf(a * c[j] - sign * b * s[j]);
+      o2[j] = f2bf(b * c[j] + sign * a * s[j]);
+    }
+    *reinterpret_cast<vec4u*>(o + base) = o1;
+    *reinterpret_cast<vec4u*>(o + base + half) = o2;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: out = silu(gate) * up, with gate_up packed [N, 2I] from one GEMM.
+// ---------------------------------------------------------------------------
+__global__ void swiglu_fwd_kernel(const u16* __restrict__ gu,
+                                  u16* __restrict__ out, long N, int I) {
+  const long total = N * (I >> 3);
+  const int nvec = I >> 3;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long row = t / nvec;
+    const int vI = (int)(t - row * nvec);
+    const size_t gbase = (size_t)row * 2 * I + vI * 8;
+    vec8u g = *reinterpret_cast<const vec8u*>(gu + gbase);
+    vec8u u = *reinterpret_cast<const vec8u*>(gu + gbase + I);
+    vec8u o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      o[j] = f2bf(gf * sig * bf2f(u[j]));
+    }
+    *reinterpret_cast<vec8u*>(out + (size_t)row * I + vI * 8) = o;
+  }
+}
+
+__global__ void swiglu_bwd_kernel(const u16* __restrict__ dout,
+                                  const u16* __restrict__ gu,
+                                  u16* __restrict__ dgu, long N, int I) {
+  const long total = N * (I >> 3);
+  const int nvec = I >> 3;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long row = t / nvec;
+    const int vI = (int)(t - row * nvec);
+    const size_t gbase = (size_t)row * 2 * I + vI * 8;
+    vec8u g = *reinterpret_cast<const vec8u*>(gu + gbase);
+    vec8u u = *reinterpret_cast<const vec8u*>(gu + gbase + I);
+    vec8u dov = *reinterpret_cast<const vec8u*>(dout + (size_t)row * I + vI * 8);
+    vec8u dg, du;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float uf = bf2f(u[j]);
+      float dof = bf2f(dov[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      dg[j] = f2bf(dof * uf * sig * (1.f + gf * (1.f - sig)));
+      du[j] = f2bf(dof * silu);
+    }
+    *reinterpret_cast<vec8u*>(dgu + gbase) = dg;
+    *reinterpret_cast<vec8u*>(dgu + gbase + I) = du;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused cross entropy over bf16 logits [N, V]: computes per-row loss and
+// OVERWRITES logits with d(mean loss)/dlogits = (softmax - onehot) * scale.
+// Avoids materializing an fp32 softmax of the [N, 128256] logits (the torch
+// eager path costs ~5x the HBM traffic). One block per row; pass 1 = online
+// max+sumexp; pass 2 = grad write. ignore_index rows get loss 0 / grad 0.
+// ---------------------------------------------------------------------------
+__global__ void cross_entropy_fwd_kernel(u16* __restrict__ logits,
+                                         const long* __restrict__ targets,
+                                         float* __restrict__ loss,
+                                         int N, int V, float scale,
+                                         long ignore_index) {
+  __shared__ float s_m[16], s_s[16];
+  const int nvec = V >> 3;
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    u16* lr = logits + (size_t)row * V;
+    const long tgt = targets[row];
+    if (tgt == ignore_index) {
+      for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+        vec8u z = {0, 0, 0, 0, 0, 0, 0, 0};
+        *reinterpret_cast<vec8u*>(lr + vI * 8) = z;
+      }
+      if (threadIdx.x == 0) loss[row] = 0.f;
+      __syncthreads();
+      continue;
+    }
+    // pass 1: thread-local online max + sumexp
+    float m = -INFINITY, s = 0.f;
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u xv = *reinterpret_cast<const vec8u*>(lr + vI * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(xv[j]);
+        if (f > m) {
+          s = s * __expf(m - f) + 1.f;
+          m = f;
+        } else {
+          s += __expf(f - m);
+        }
+      }
+    }
+    // wave merge of (m, s)
+#pragma unroll
+    for (int o = 32; o > 0; o >>= 1) {
+      float m2 = __shfl_down(m, o, WAVE);
+      float s2 = __shfl_down(s, o, WAVE);
+      float mn = fmaxf(m, m2);
+      s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+      m = mn;
+    }
+    const int wid = threadIdx.x >> 6;
+    const int nw = blockDim.x >> 6;
+    if ((threadIdx.x & 63) == 0) { s_m[wid] = m; s_s[wid] = s; }
+    __syncthreads();
+    float M = -INFINITY;
+    for (int i = 0; i < nw; ++i) M = fmaxf(M, s_m[i]);
+    float Z = 0.f;
+    for (int i = 0; i < nw; ++i) Z += s_s[i] * __expf(s_m[i] - M);
+    __syncthreads();
+    const float logZ = __logf(Z) + M;
+    if (threadIdx.x == 0) loss[row] = logZ - bf2f(lr[tgt]);
+    __syncthreads();  // target logit must be read before pass 2 overwrites it
+    const float inv_Z = 1.f / Z;
+    // pass 2: grad in place
+    for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
+      vec8u xv = *reinterpret_cast<const vec8u*>(lr + vI * 8);
+      vec8u ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const long idx = (long)vI * 8 + j;
+        float p = __expf(bf2f(xv[j]) - M) * inv_Z;
+        if (idx == tgt) p -= 1.f;
+        ov[j] = f2bf(p * scale);
+      }
+      *reinterpret_cast<vec8u*>(lr + vI * 8) = ov;
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused AdamW over a flat bf16 param/grad bucket with fp32 m, v state.
+// One kernel per bucket (buckets are the DDP comm buckets), fp32 math,
+// decoupled weight decay. grad_scale folds in any 1/world_size factor.
+// Traffic: 22 B/param (vs ~5 separate eager kernels).
+// ---------------------------------------------------------------------------
+__global__ void adamw_kernel(u16* __restrict__ p, const u16* __restrict__ g,
+                             float* __restrict__ m, float* __restrict__ v,
+                             long n, float lr, float beta1, float beta2,
+                             float eps, float wd, float bc1, float bc2,
+                             float grad_scale) {
+  const long nvec = n >> 3;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < nvec;
+       t += (long)gridDim.x * blockDim.x) {
+    const size_t i = (size_t)t * 8;
+    vec8u pv = *reinterpret_cast<const vec8u*>(p + i);
+    vec8u gv = *reinterpret_cast<const vec8u*>(g + i);
+    vec4f m0 = *reinterpret_cast<const vec4f*>(m + i);
+    vec4f m1 = *reinterpret_cast<const vec4f*>(m + i + 4);
+    vec4f v0 = *reinterpret_cast<const vec4f*>(v + i);
+    vec4f v1 = *reinterpret_cast<const vec4f*>(v + i + 4);
+    vec8u po;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(gv[j]) * grad_scale;
+      float pf = bf2f(pv[j]);
+      float mj = (j < 4) ? m0[j] : m1[j - 4];
+      float vj = (j < 4) ? v0[j] : v1[j - 4];
+      mj = beta1 * mj + (1.f - beta1) * gf;
+      vj = beta2 * vj + (1.f - beta2) * gf * gf;
+      float mhat = mj / bc1;
+      float vhat = vj / bc2;
+      pf = pf - lr * (mhat / (sqrtf(vhat) + eps) + wd * pf);
+      po[j] = f2bf(pf);
+      if (j < 4) { m0[j] = mj; v0[j] = vj; }
+      else { m1[j - 4] = mj; v1[j - 4] = vj; }
+    }
+    *reinterpret_cast<vec8u*>(p + i) = po;
+    *reinterpret_cast<vec4f*>(m + i) = m0;
+    *reinterpret_cast<vec4f*>(m + i + 4) = m1;
+    *reinterpret_cast<vec4f*>(v + i) = v0;
+    *reinterpret_cast<vec4f*>(v + i + 4) = v1;
+  }
+}
+
+// Tail handler for n not divisible by 8 (scalar; runs in the same launch).
+__global__ void adamw_tail_kernel(u16* p, const u16* g, float* m, float* v,
+                                  long start, long n, float lr, float beta1,
+                                  float beta2, float eps, float wd, float bc1,
+                                  float bc2, float grad_scale) {
+  long i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  float gf = bf2f(g[i]) * grad_scale;
+  float pf = bf2f(p[i]);
+  float mj = beta1 * m[i] + (1.f - beta1) * gf;
+  float vj = beta2 * v[i] + (1.f - beta2) * gf * gf;
+  m[i] = mj; v[i] = vj;
+  float mhat = mj / bc1, vhat = vj / bc2;
+  p[i] = f2bf(pf - lr * (mhat / (sqrtf(vhat) + eps) + wd * pf));
+}
+
+// ---------------------------------------------------------------------------
+// C launchers
+// ---------------------------------------------------------------------------
+static inline int grid_for(long work, int block, int cap = 2048) {
+  long g = (work + block - 1) / block;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+extern "C" {
+
+void kt_rmsnorm_fwd(const void* x, const void* w, void* y, void* invrms,
+                    int N, int H, float eps, hipStream_t stream) {
+  int grid = N < 2048 ? N : 2048;
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const u16*)x, (const u16*)w, (u16*)y, (float*)invrms, N,
+                     H, eps);
+}
+
+void kt_rmsnorm_bwd(const void* dy, const void* x, const void* w,
+                    const void* invrms, void* dx, void* dw_partial,
+                    void* dw, int P, int N, int H, hipStream_t stream) {
+  int grid = N < P ? N : P;
+  size_t lds = (size_t)H * sizeof(float);
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), lds, stream,
+                     (const u16*)dy, (const u16*)x, (const u16*)w,
+                     (const float*)invrms, (u16*)dx, (float*)dw_partial, N, H);
+  int cgrid = (H / 4 + 255) / 256;
+  hipLaunchKernelGGL(colsum_bf16_kernel, dim3(cgrid), dim3(256), 0, stream,
+                     (const float*)dw_partial, (u16*)dw, grid, H);
+}
+
+void kt_rope(const void* x, void* o, const void* cost, const void* sint,
+             long total_quads, int S, int Hh, int D, float sign,
+             hipStream_t stream) {
+  int grid = grid_for(total_quads, 256);
+  hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const u16*)x, (u16*)o, (const float*)cost,
+                     (const float*)sint, total_quads, S, Hh, D, sign);
+}
+
+void kt_swiglu_fwd(const void* gu, void* out, long N, int I,
+                   hipStream_t stream) {
+  int grid = grid_for(N * (I >> 3), 256);
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const u16*)gu, (u16*)out, N, I);
+}
+
+void kt_swiglu_bwd(const void* dout, const void* gu, void* dgu, long N, int I,
+                   hipStream_t stream) {
+  int grid = grid_for(N * (I >> 3), 256);
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const u16*)dout, (const u16*)gu, (u16*)dgu, N, I);
+}
+
+void kt_cross_entropy_fwd(void* logits, const void* targets, void* loss, int N,
+                          int V, float scale, long ignore_index,
+                          hipStream_t stream) {
+  int grid = N < 2048 ? N : 2048;
+  hipLaunchKernelGGL(cross_entropy_fwd_kernel, dim3(grid), dim3(256), 0,
+                     stream, (u16*)logits, (const long*)targets, (float*)loss,
+                     N, V, scale, ignore_index);
+}
+
+void kt_adamw(void* p, const void* g, void* m, void* v, long n, float lr,
+              float beta1, float beta2, float eps, float wd, float bc1,
+              float bc2, float grad_scale, hipStream_t stream) {
+  long nvec = n >> 3;
+  if (nvec > 0) {
+    int grid = grid_for(nvec, 256);
+    hipLaunchKernelGGL(adamw_kernel, dim3(grid), dim3(256), 0, stream, (u16*)p,
+                       (const u16*)g, (float*)m, (float*)v, n, lr, beta1,
+                       beta2, eps, wd, bc1, bc2, grad_scale);
+  }
+  long tail = n - (nvec << 3);
+  if (tail > 0) {
+    hipLaunchKernelGGL(adamw_tail_kernel, dim3(1), dim3(64), 0, stream,
+                       (u16*)p, (const u16*)g, (float*)m, (float*)v, nvec << 3,
+                       n, lr, beta1, beta2, eps, wd, bc1, bc2, grad_scale);
+  }
+}
+
+}  // extern "C"

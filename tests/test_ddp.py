@@ -1,0 +1,100 @@
+"""Multi-process FlatDDP test on gloo, world_size=2 (CPU).
+
+Verifies the distributed path is correct by construction: two ranks with
+different data shards must produce identical post-step params, equal to a
+single-process run on the combined batch (grad averaging).
+"""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from kubetorch_amd.models import Llama, llama_tiny
+from kubetorch_amd.parallel import FlatDDP
+
+PORT = 29511
+
+
+def _make(seed=0):
+    torch.manual_seed(seed)
+    cfg = llama_tiny(n_layers=1, dim=128, intermediate=256, vocab_size=256,
+                     n_heads=4, n_kv_heads=2)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.bfloat16)
+    try:
+        model = Llama(cfg)
+    finally:
+        torch.set_default_dtype(prev)
+    return model, cfg
+
+
+def _data(cfg):
+    g = torch.Generator().manual_seed(7)
+    x = torch.randint(0, cfg.vocab_size, (4, 32), generator=g)
+    y = torch.randint(0, cfg.vocab_size, (4, 32), generator=g)
+    return x, y
+
+
+def _worker(rank, world, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(PORT)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        model, cfg = _make()
+        eng = FlatDDP(model, lr=1e-2, bucket_mb=1)
+        eng.broadcast_params(src=0)
+        x, y = _data(cfg)
+        # shard the batch across ranks
+        xs = x[rank * 2:(rank + 1) * 2]
+        ys = y[rank * 2:(rank + 1) * 2]
+        for _ in range(2):
+            loss = model.loss(xs, ys)
+            loss.backward()
+            eng.step()
+        flat = torch.cat([b.flat_param.float() for b in eng.buckets])
+        q.put((rank, flat))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_flatddp_two_ranks_match_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, flat = q.get()
+        results[rank] = flat
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+
+    # ranks must agree exactly (same reduced grads, same update)
+    torch.testing.assert_close(results[0], results[1], rtol=0, atol=0)
+
+    # and match a single-process run over the full batch
+    model, cfg = _make()
+    eng = FlatDDP(model, lr=1e-2, bucket_mb=1)
+    x, y = _data(cfg)
+    for _ in range(2):
+        # average of shard losses == mean over full batch here since shards
+        # are equal-sized; emulate grad averaging by running both shards
+        # and averaging grads manually
+        l0 = model.loss(x[:2], y[:2])
+        l0.backward()
+        g0 = [b.flat_grad.clone() for b in eng.buckets]
+        eng.zero_grad()
+        l1 = model.loss(x[2:], y[2:])
+        l1.backward()
+        for b, g in zip(eng.buckets, g0):
+            b.flat_grad.add_(g)
+        # emulate SUM all-reduce then 1/world scale inside step
+        eng._world = 2
+        eng.step()
+        eng._world = 1
+    single = torch.cat([b.flat_param.float() for b in eng.buckets])
+    torch.testing.assert_close(results[0], single, rtol=2e-2, atol=2e-2)

@@ -1,0 +1,226 @@
+"""Numerics tests for kubetorch_amd.ops.
+
+CPU tests check the wrapper/autograd semantics against independent plain
+PyTorch fp32 implementations. GPU tests (-m gpu) compare the gfx950 HIP
+kernels against the same fp32 references (reference parity model:
+SURVEY.md §4 — kernel vs plain fp32 torch).
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from kubetorch_amd import ops
+
+BF16 = torch.bfloat16
+
+
+def devices():
+    d = ["cpu"]
+    if torch.cuda.is_available():
+        d.append("cuda")
+    return d
+
+
+def _assert_close(a, b, rtol=2e-2, atol=2e-2, msg=""):
+    torch.testing.assert_close(a.float(), b.float(), rtol=rtol, atol=atol, msg=msg)
+
+
+# ---------------------------------------------------------------------------
+# fp32 references (independent of ops.py internals)
+# ---------------------------------------------------------------------------
+def ref_rmsnorm(x, w, eps):
+    xf = x.float()
+    return xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps) * w.float()
+
+
+def ref_rope(x, cos, sin):
+    B, S, H, D = x.shape
+    xf = x.float()
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
+    c = cos[:S].view(1, S, 1, -1)
+    s = sin[:S].view(1, S, 1, -1)
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1)
+
+
+def ref_swiglu(gu):
+    I = gu.shape[-1] // 2
+    g, u = gu[..., :I].float(), gu[..., I:].float()
+    return F.silu(g) * u
+
+
+def _run_fwd_bwd(fn, *inputs):
+    ins = [t.detach().clone().requires_grad_(t.is_floating_point()) for t in inputs]
+    out = fn(*ins)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+    grads = [t.grad for t in ins if t.is_floating_point()]
+    return out.detach(), grads, gout
+
+
+class TestRMSNorm:
+    def test_fwd_bwd_cpu(self):
+        self._check("cpu")
+
+    @pytest.mark.gpu
+    def test_fwd_bwd_gpu(self):
+        self._check("cuda")
+
+    def _check(self, device):
+        torch.manual_seed(0)
+        N, H = 33, 256
+        x = torch.randn(N, H, dtype=BF16, device=device)
+        w = torch.randn(H, dtype=BF16, device=device)
+        gout = None
+
+        def mine(xi, wi):
+            return ops.rmsnorm(xi, wi, 1e-5)
+
+        y, (dx, dw), gout = _run_fwd_bwd(mine, x, w)
+
+        xr = x.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        yr = ref_rmsnorm(xr, wr, 1e-5)
+        yr.backward(gout.float())
+        _assert_close(y, yr, msg="rmsnorm fwd")
+        _assert_close(dx, xr.grad, msg="rmsnorm dx")
+        _assert_close(dw, wr.grad, rtol=3e-2, atol=3e-1, msg="rmsnorm dw")
+
+    @pytest.mark.gpu
+    def test_gpu_large_row(self):
+        torch.manual_seed(1)
+        N, H = 4096, 4096
+        x = torch.randn(N, H, dtype=BF16, device="cuda")
+        w = torch.randn(H, dtype=BF16, device="cuda")
+        y = ops.rmsnorm(x, w, 1e-5)
+        _assert_close(y, ref_rmsnorm(x, w, 1e-5), msg="rmsnorm 4096")
+
+
+class TestRope:
+    def _check(self, device):
+        torch.manual_seed(0)
+        B, S, H, D = 2, 64, 4, 64
+        cos, sin = ops.precompute_rope(S, D, base=10000.0, device=device)
+        x = torch.randn(B, S, H, D, dtype=BF16, device=device)
+
+        y, (dx,), gout = _run_fwd_bwd(lambda xi: ops.rope(xi, cos, sin), x)
+        xr = x.float().requires_grad_(True)
+        yr = ref_rope(xr, cos, sin)
+        yr.backward(gout.float())
+        _assert_close(y, yr, msg="rope fwd")
+        _assert_close(dx, xr.grad, msg="rope bwd")
+
+    def test_cpu(self):
+        self._check("cpu")
+
+    @pytest.mark.gpu
+    def test_gpu(self):
+        self._check("cuda")
+
+
+class TestSwiGLU:
+    def _check(self, device):
+        torch.manual_seed(0)
+        N, I = 65, 128
+        gu = torch.randn(N, 2 * I, dtype=BF16, device=device)
+        y, (dgu,), gout = _run_fwd_bwd(lambda g: ops.swiglu(g), gu)
+        gr = gu.float().requires_grad_(True)
+        yr = ref_swiglu(gr)
+        yr.backward(gout.float())
+        _assert_close(y, yr, msg="swiglu fwd")
+        _assert_close(dgu, gr.grad, msg="swiglu bwd")
+
+    def test_cpu(self):
+        self._check("cpu")
+
+    @pytest.mark.gpu
+    def test_gpu(self):
+        self._check("cuda")
+
+
+class TestFusedCE:
+    def _check(self, device, ignore_some=False):
+        torch.manual_seed(0)
+        N, V = 128, 1024
+        logits = torch.randn(N, V, dtype=BF16, device=device) * 4
+        targets = torch.randint(0, V, (N,), device=device)
+        if ignore_some:
+            targets[::5] = -100
+
+        lg = logits.detach().clone().requires_grad_(True)
+        loss = ops.fused_cross_entropy(lg, targets)
+        loss.backward()
+        dl = lg.grad
+
+        lr = logits.float().requires_grad_(True)
+        loss_ref = F.cross_entropy(lr, targets, ignore_index=-100)
+        loss_ref.backward()
+        _assert_close(loss, loss_ref, rtol=1e-2, atol=1e-2, msg="ce loss")
+        _assert_close(dl, lr.grad, rtol=5e-2, atol=1e-3, msg="ce grad")
+
+    def test_cpu(self):
+        self._check("cpu")
+
+    def test_cpu_ignore(self):
+        self._check("cpu", ignore_some=True)
+
+    @pytest.mark.gpu
+    def test_gpu(self):
+        self._check("cuda")
+
+    @pytest.mark.gpu
+    def test_gpu_ignore(self):
+        self._check("cuda", ignore_some=True)
+
+    @pytest.mark.gpu
+    def test_gpu_large_vocab(self):
+        torch.manual_seed(2)
+        N, V = 64, 128256
+        logits = torch.randn(N, V, dtype=BF16, device="cuda") * 3
+        targets = torch.randint(0, V, (N,), device="cuda")
+        ref = F.cross_entropy(logits.float(), targets)
+        lg = logits.clone().requires_grad_(True)
+        loss = ops.fused_cross_entropy(lg, targets)
+        _assert_close(loss, ref, rtol=1e-2, atol=1e-2, msg="ce large vocab")
+
+
+class TestAdamW:
+    def _check(self, device):
+        torch.manual_seed(0)
+        n = 1000 + 3  # exercise the tail kernel
+        p = torch.randn(n, dtype=BF16, device=device)
+        g = torch.randn(n, dtype=BF16, device=device)
+        m = torch.zeros(n, dtype=torch.float32, device=device)
+        v = torch.zeros(n, dtype=torch.float32, device=device)
+        p2, m2, v2 = p.float().clone(), m.clone(), v.clone()
+
+        for step in (1, 2, 3):
+            ops.adamw_(p, g, m, v, 1e-2, 0.9, 0.95, 1e-8, 0.1, step, 0.5)
+            # reference in fp32 (mirrors bf16 param rounding each step)
+            gf = g.float() * 0.5
+            m2.mul_(0.9).add_(gf, alpha=0.1)
+            v2.mul_(0.95).addcmul_(gf, gf, value=0.05)
+            mh = m2 / (1 - 0.9 ** step)
+            vh = v2 / (1 - 0.95 ** step)
+            p2 = (p2 - 1e-2 * (mh / (vh.sqrt() + 1e-8) + 0.1 * p2)).to(BF16).float()
+
+        _assert_close(p, p2, rtol=1e-2, atol=1e-2, msg="adamw params")
+        _assert_close(m, m2, rtol=1e-2, atol=1e-3, msg="adamw m")
+        _assert_close(v, v2, rtol=1e-2, atol=1e-4, msg="adamw v")
+
+    def test_cpu(self):
+        self._check("cpu")
+
+    @pytest.mark.gpu
+    def test_gpu(self):
+        self._check("cuda")
+
+
+@pytest.mark.gpu
+def test_hip_extension_required_on_gpu():
+    """On a GPU box the HIP extension must be present and be the code path
+    that runs (no silent eager fallback)."""
+    assert ops.hip_available(), "HIP extension missing on GPU box"
+    x = torch.randn(8, 64, dtype=BF16, device="cuda")
+    w = torch.ones(64, dtype=BF16, device="cuda")
+    y = ops.rmsnorm(x, w)
+    assert y.is_cuda and y.dtype == BF16
