@@ -4,8 +4,10 @@
 
 #include <torch/extension.h>
 
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+// torch-ROCm registers GPU tensors under DeviceType::CUDA; its native HIP
+// guard/stream accessors for that device type are the "masquerading" ones.
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 extern "C" {
 void kt_rmsnorm_fwd(const void* x, const void* w, void* y, void* invrms,
@@ -36,7 +38,7 @@ namespace {
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
 
 hipStream_t cur_stream(const at::Tensor& t) {
-  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(t.device().index()).stream();
 }
 
 std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w,
@@ -47,7 +49,7 @@ std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w,
   const long N = x.numel() / H;
   TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
   TORCH_CHECK(w.numel() == H, "weight shape mismatch");
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto y = at::empty_like(x);
   auto invrms = at::empty({N}, x.options().dtype(at::kFloat));
   kt_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), invrms.data_ptr(),
@@ -63,7 +65,7 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   CHECK_BF16_CONTIG(w);
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto dx = at::empty_like(x);
   auto dw = at::empty_like(w);
   const int P = N < 512 ? (int)N : 512;
@@ -84,7 +86,7 @@ at::Tensor rope(const at::Tensor& x, const at::Tensor& cost,
   const int Hh = (int)x.size(1);
   TORCH_CHECK(D % 8 == 0, "head dim must be a multiple of 8");
   TORCH_CHECK(cost.size(0) >= S && cost.size(1) == D / 2, "cos table shape");
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto o = at::empty_like(x);
   const long total_quads = (long)x.size(0) * Hh * (D / 8);
   kt_rope(x.data_ptr(), o.data_ptr(), cost.data_ptr(), sint.data_ptr(),
@@ -98,7 +100,7 @@ at::Tensor swiglu_fwd(const at::Tensor& gu) {
   TORCH_CHECK(twoI % 16 == 0, "2*I must be a multiple of 16");
   const int I = twoI / 2;
   const long N = gu.numel() / twoI;
-  c10::hip::OptionalHIPGuard guard(gu.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(gu.device());
   auto sizes = gu.sizes().vec();
   sizes.back() = I;
   auto out = at::empty(sizes, gu.options());
@@ -112,7 +114,7 @@ at::Tensor swiglu_bwd(const at::Tensor& dout, const at::Tensor& gu) {
   const int twoI = (int)gu.size(-1);
   const int I = twoI / 2;
   const long N = gu.numel() / twoI;
-  c10::hip::OptionalHIPGuard guard(gu.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(gu.device());
   auto dgu = at::empty_like(gu);
   kt_swiglu_bwd(dout.data_ptr(), gu.data_ptr(), dgu.data_ptr(), N, I,
                 cur_stream(gu));
@@ -127,7 +129,7 @@ at::Tensor cross_entropy_fwd_(at::Tensor logits, const at::Tensor& targets,
   const long N = logits.numel() / V;
   TORCH_CHECK(V % 8 == 0, "V must be a multiple of 8");
   TORCH_CHECK(targets.numel() == N, "targets shape mismatch");
-  c10::hip::OptionalHIPGuard guard(logits.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(logits.device());
   auto loss = at::empty({N}, logits.options().dtype(at::kFloat));
   kt_cross_entropy_fwd(logits.data_ptr(), targets.data_ptr(), loss.data_ptr(),
                        (int)N, V, (float)scale, ignore_index,
@@ -143,7 +145,7 @@ void adamw_(at::Tensor p, const at::Tensor& g, at::Tensor m, at::Tensor v,
   TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() &&
               p.numel() == v.numel());
-  c10::hip::OptionalHIPGuard guard(p.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(p.device());
   const float bc1 = 1.f - powf((float)beta1, (float)step);
   const float bc2 = 1.f - powf((float)beta2, (float)step);
   kt_adamw(p.data_ptr(), g.data_ptr(), m.data_ptr(), v.data_ptr(), p.numel(),
