@@ -296,13 +296,14 @@ __global__ void cross_entropy_fwd_kernel(u16* __restrict__ logits,
         }
       }
     }
-    // wave merge of (m, s)
+    // wave merge of (m, s); threads with no elements carry (m=-inf, s=0)
+    // and merging two of those must not compute exp(-inf - -inf) = NaN.
 #pragma unroll
     for (int o = 32; o > 0; o >>= 1) {
       float m2 = __shfl_down(m, o, WAVE);
       float s2 = __shfl_down(s, o, WAVE);
       float mn = fmaxf(m, m2);
-      s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+      if (mn > -INFINITY) s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
       m = mn;
     }
     const int wid = threadIdx.x >> 6;
@@ -312,7 +313,8 @@ __global__ void cross_entropy_fwd_kernel(u16* __restrict__ logits,
     float M = -INFINITY;
     for (int i = 0; i < nw; ++i) M = fmaxf(M, s_m[i]);
     float Z = 0.f;
-    for (int i = 0; i < nw; ++i) Z += s_s[i] * __expf(s_m[i] - M);
+    for (int i = 0; i < nw; ++i)
+      if (s_m[i] > -INFINITY) Z += s_s[i] * __expf(s_m[i] - M);
     __syncthreads();
     const float logZ = __logf(Z) + M;
     if (threadIdx.x == 0) loss[row] = logZ - bf2f(lr[tgt]);
