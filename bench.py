@@ -33,6 +33,8 @@ def main():
                     choices=["llama3-8b", "tiny"])
     ap.add_argument("--bucket-mb", type=int, default=256)
     ap.add_argument("--lr", type=float, default=1e-4)
+    ap.add_argument("--sdpa", type=str, default="flash",
+                    choices=["flash", "efficient", "math"])
     args = ap.parse_args()
 
     from kubetorch_amd.models import Llama, llama3_8b, llama_tiny
@@ -69,8 +71,24 @@ def main():
     tokens = torch.randint(0, cfg.vocab_size, (B, S + 1), generator=gen).to(dev)
     x, y = tokens[:, :-1].contiguous(), tokens[:, 1:].contiguous()
 
+    from contextlib import nullcontext
+
+    if dev.type == "cuda":
+        from torch.nn.attention import SDPBackend, sdpa_kernel
+
+        backend = {"flash": SDPBackend.FLASH_ATTENTION,
+                   "efficient": SDPBackend.EFFICIENT_ATTENTION,
+                   "math": SDPBackend.MATH}[args.sdpa]
+
+        def sdpa_ctx():
+            return sdpa_kernel(backend)
+    else:
+        def sdpa_ctx():
+            return nullcontext()
+
     def one_step():
-        loss = model.loss(x, y)
+        with sdpa_ctx():
+            loss = model.loss(x, y)
         loss.backward()
         engine.step()
         return loss
@@ -98,6 +116,10 @@ def main():
     if world > 1:
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
     elapsed = float(t.item())
+
+    if dev.type == "cuda":
+        log(f"[bench rank{rank}] peak HBM: "
+            f"{torch.cuda.max_memory_allocated(dev)/2**30:.1f} GiB")
 
     tokens_per_step = B * S * world
     toks_per_sec = tokens_per_step * args.steps / elapsed
