@@ -33,7 +33,9 @@ def main():
                     choices=["llama3-8b", "tiny"])
     ap.add_argument("--bucket-mb", type=int, default=256)
     ap.add_argument("--lr", type=float, default=1e-4)
-    ap.add_argument("--sdpa", type=str, default="flash",
+    # AOTriton's "efficient" kernels beat its flash kernels on gfx950 by
+    # ~14% end-to-end (profiles/r01_sdpa_ab.md) -> default efficient.
+    ap.add_argument("--sdpa", type=str, default="efficient",
                     choices=["flash", "efficient", "math"])
     args = ap.parse_args()
 

@@ -147,18 +147,27 @@ __global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
 }
 
 // Column-sum of the [P, H] fp32 partial buffer -> dw [H] bf16.
+// One block per 4-column group (H/4 blocks keep the chip busy; the previous
+// thread-per-column layout launched only H/1024 blocks and left 255/256 CUs
+// idle), threads split the P rows, block-reduce per column.
 __global__ void colsum_bf16_kernel(const float* __restrict__ partial,
                                    u16* __restrict__ out, int P, int H) {
-  int col = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  const int col = blockIdx.x * 4;
   if (col >= H) return;
-  float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-  for (int p = 0; p < P; ++p) {
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  for (int p = threadIdx.x; p < P; p += blockDim.x) {
     vec4f v = *reinterpret_cast<const vec4f*>(partial + (size_t)p * H + col);
-    acc0 += v[0]; acc1 += v[1]; acc2 += v[2]; acc3 += v[3];
+    a0 += v[0]; a1 += v[1]; a2 += v[2]; a3 += v[3];
   }
-  vec4u o;
-  o[0] = f2bf(acc0); o[1] = f2bf(acc1); o[2] = f2bf(acc2); o[3] = f2bf(acc3);
-  *reinterpret_cast<vec4u*>(out + col) = o;
+  a0 = block_reduce_sum(a0);
+  a1 = block_reduce_sum(a1);
+  a2 = block_reduce_sum(a2);
+  a3 = block_reduce_sum(a3);
+  if (threadIdx.x == 0) {
+    vec4u o;
+    o[0] = f2bf(a0); o[1] = f2bf(a1); o[2] = f2bf(a2); o[3] = f2bf(a3);
+    *reinterpret_cast<vec4u*>(out + col) = o;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -426,7 +435,7 @@ void kt_rmsnorm_bwd(const void* dy, const void* x, const void* w,
   hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), lds, stream,
                      (const u16*)dy, (const u16*)x, (const u16*)w,
                      (const float*)invrms, (u16*)dx, (float*)dw_partial, N, H);
-  int cgrid = (H / 4 + 255) / 256;
+  int cgrid = (H + 3) / 4;
   hipLaunchKernelGGL(colsum_bf16_kernel, dim3(cgrid), dim3(256), 0, stream,
                      (const float*)dw_partial, (u16*)dw, grid, H);
 }

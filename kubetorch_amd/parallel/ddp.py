@@ -76,7 +76,8 @@ class FlatDDP:
     """
 
     def __init__(self, model, lr=3e-4, betas=(0.9, 0.95), eps=1e-8,
-                 weight_decay=0.1, bucket_mb=256, process_group=None):
+                 weight_decay=0.1, bucket_mb=256, process_group=None,
+                 overlap_optimizer=True):
         self.model = model
         self.lr = lr
         self.betas = betas
@@ -87,6 +88,14 @@ class FlatDDP:
         self._world = (
             dist.get_world_size(process_group) if dist.is_initialized() else 1
         )
+        # Overlap mode: fused AdamW for a bucket is launched on a side HIP
+        # stream as soon as the bucket's grads are final (and all-reduced),
+        # running concurrently with the remaining backward. Safe because a
+        # param's value is only read by its own layer's backward, which has
+        # completed by the time its post-accumulate hook fires.
+        self._overlap = overlap_optimizer and torch.cuda.is_available()
+        self._opt_stream = torch.cuda.Stream() if self._overlap else None
+        self._step_started = False
         params = [p for p in model.parameters() if p.requires_grad]
         if not params:
             raise ValueError("model has no trainable parameters")
@@ -141,28 +150,59 @@ class FlatDDP:
             view.add_(p.grad)
             p.grad = view
         b.pending -= 1
-        if b.pending == 0 and self._world > 1:
+        if b.pending == 0:
+            self._bucket_ready(b)
+
+    def _bucket_ready(self, b):
+        if self._world > 1:
             b.work = dist.all_reduce(
                 b.flat_grad, op=dist.ReduceOp.SUM, group=self.pg, async_op=True
+            )
+        if not self._overlap:
+            return
+        if not self._step_started:
+            self._step_started = True
+            self.step_count += 1
+        # run this bucket's AdamW on the side stream, ordered after the
+        # grad-producing kernels (event) and the all-reduce (work.wait()).
+        ev = torch.cuda.Event()
+        ev.record(torch.cuda.current_stream())
+        with torch.cuda.stream(self._opt_stream):
+            self._opt_stream.wait_event(ev)
+            if b.work is not None:
+                b.work.wait()  # syncs the side stream with the RCCL stream
+                b.work = None
+            ops.adamw_(
+                b.flat_param, b.flat_grad, b.m, b.v, self.lr,
+                self.betas[0], self.betas[1], self.eps, self.weight_decay,
+                self.step_count, 1.0 / self._world,
             )
 
     # -- optimizer side ------------------------------------------------------
     @torch.no_grad()
     def step(self, lr=None):
-        """Wait for outstanding all-reduces, run fused AdamW, reset grads."""
+        """Finish the update: in overlap mode just join the side stream and
+        reset; otherwise wait for all-reduces and run fused AdamW here.
+        NOTE (overlap mode): set the LR for step N before its backward."""
         if lr is not None:
             self.lr = lr
-        self.step_count += 1
-        grad_scale = 1.0 / self._world
-        for b in self.buckets:
-            if b.work is not None:
-                b.work.wait()
-                b.work = None
-            ops.adamw_(
-                b.flat_param, b.flat_grad, b.m, b.v, self.lr,
-                self.betas[0], self.betas[1], self.eps, self.weight_decay,
-                self.step_count, grad_scale,
-            )
+        if self._overlap:
+            if not self._step_started:  # backward produced no grads
+                self.step_count += 1
+            torch.cuda.current_stream().wait_stream(self._opt_stream)
+            self._step_started = False
+        else:
+            self.step_count += 1
+            grad_scale = 1.0 / self._world
+            for b in self.buckets:
+                if b.work is not None:
+                    b.work.wait()
+                    b.work = None
+                ops.adamw_(
+                    b.flat_param, b.flat_grad, b.m, b.v, self.lr,
+                    self.betas[0], self.betas[1], self.eps, self.weight_decay,
+                    self.step_count, grad_scale,
+                )
         for b in self.buckets:
             b.flat_grad.zero_()
             b.pending = len(b.params)
