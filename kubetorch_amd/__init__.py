@@ -43,14 +43,19 @@ _LAZY = {
 def __getattr__(name):
     if name in _LAZY:
         import importlib
+        import sys
 
         mod_name, attr = _LAZY[name]
         mod = importlib.import_module(mod_name)
         obj = mod if attr is None else getattr(mod, attr)
-        globals()[name] = obj
+        # NB: use the module dict directly — importing the kubetorch_amd.globals
+        # submodule shadows the globals() builtin in this namespace.
+        sys.modules[__name__].__dict__[name] = obj
         return obj
     raise AttributeError(f"module 'kubetorch_amd' has no attribute {name!r}")
 
 
 def __dir__():
-    return sorted(list(globals()) + list(_LAZY))
+    import sys
+
+    return sorted(list(sys.modules[__name__].__dict__) + list(_LAZY))

@@ -1,0 +1,57 @@
+"""App: deploy an arbitrary CLI/HTTP app (reference parity:
+resources/compute/app.py). The command ships as the pod's main process; an
+optional health path gates readiness."""
+import os
+
+from kubetorch_amd.client.module import Module, sanitize_name
+from kubetorch_amd.compute.compute import Compute
+from kubetorch_amd.config import config
+
+
+class App(Module):
+    module_type = "app"
+
+    def __init__(self, command, name=None, compute=None, health_path=None,
+                 port=None):
+        pointers = {
+            "name": name or "app",
+            "file_path": "",
+            "rel_path": "",
+            "project_root": os.getcwd(),
+        }
+        super().__init__(pointers, name=name, compute=compute)
+        self.command = command
+        self.health_path = health_path
+        self.port = port
+
+    def metadata(self):
+        md = super().metadata()
+        md["command"] = self.command
+        md["health_path"] = self.health_path
+        md["port"] = self.port
+        return md
+
+    def to(self, compute: Compute = None, **kw):
+        if compute is not None:
+            self.compute = compute
+        if self.compute is None:
+            self.compute = Compute(cpus=1)
+        # app-mode pods run the user command directly
+        if self.compute._raw_manifest is None and not self.compute.local:
+            manifest = self.compute.to_manifest(self.name,
+                                                username=config.username)
+            container = manifest["spec"]["template"]["spec"]["containers"][0]
+            container["command"] = ["bash", "-lc", self.command]
+            self.compute._raw_manifest = manifest
+        return super().to()
+
+    def _wait_ready(self, timeout=None, reloaded=False):
+        if self.compute.local:
+            return  # local driver runs the command; no HTTP server to poll
+        if self.health_path:
+            super()._wait_ready(timeout=timeout or 900, reloaded=reloaded)
+
+
+def app(command, name=None, health_path=None, port=None):
+    return App(command, name=sanitize_name(name) if name else None,
+               health_path=health_path, port=port)

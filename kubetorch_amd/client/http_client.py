@@ -1,0 +1,150 @@
+"""Client->pod HTTP calls with remote-exception reconstruction and optional
+live log streaming. (Reference parity: serving/http_client.py.)"""
+import base64
+import json
+import pickle
+import threading
+import uuid
+
+import httpx
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.exceptions import reconstruct_exception
+
+_client = None
+_client_lock = threading.Lock()
+
+
+def shared_client():
+    global _client
+    with _client_lock:
+        if _client is None:
+            _client = httpx.Client(
+                timeout=httpx.Timeout(C.HTTP_TIMEOUT, connect=15),
+                limits=httpx.Limits(max_connections=100),
+            )
+        return _client
+
+
+class LogStreamer:
+    """Background thread tailing /logs/tail for a request id and printing
+    lines as they arrive (poll-based tail; sub-second latency)."""
+
+    def __init__(self, base_url, request_id, printer=print):
+        self.base_url = base_url
+        self.request_id = request_id
+        self.printer = printer
+        self._stop = threading.Event()
+        self._since = None
+        self._thread = threading.Thread(target=self._run, daemon=True)
+
+    def start(self):
+        try:
+            r = shared_client().get(self.base_url + "/logs/tail",
+                                    params={"since": 0, "limit": 1})
+            self._since = r.json().get("seq", 0)
+        except Exception:
+            self._since = 0
+        self._thread.start()
+        return self
+
+    def stop(self, drain=True):
+        if drain:
+            self._poll_once()
+        self._stop.set()
+
+    def _poll_once(self):
+        try:
+            r = shared_client().get(
+                self.base_url + "/logs/tail",
+                params={"since": self._since, "request_id": self.request_id},
+                timeout=5,
+            )
+            for e in r.json().get("entries", []):
+                self.printer(f"[remote {e['source']}] {e['line']}")
+                self._since = max(self._since, e["seq"] + 1)
+        except Exception:
+            pass
+
+    def _run(self):
+        while not self._stop.wait(0.25):
+            self._poll_once()
+
+
+class HTTPClient:
+    def __init__(self, base_url, name):
+        self.base_url = base_url.rstrip("/")
+        self.name = name
+
+    def is_ready(self, launch_id=None, timeout=5):
+        try:
+            r = shared_client().get(
+                self.base_url + "/ready",
+                params={"launch_id": launch_id} if launch_id else {},
+                timeout=timeout,
+            )
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    def call(self, args=(), kwargs=None, method=None, serialization="pickle",
+             stream_logs=False, timeout=None, workers=None,
+             restart_procs=False, request_id=None):
+        rid = request_id or uuid.uuid4().hex
+        url = f"{self.base_url}/call/{self.name}"
+        if method:
+            url += f"/{method}"
+        params = {}
+        if workers is not None:
+            params["workers"] = (workers if isinstance(workers, str)
+                                 else json.dumps(workers))
+        if restart_procs:
+            params["restart_procs"] = "true"
+        headers = {"X-Request-ID": rid, "X-Serialization": serialization}
+        if serialization == "pickle":
+            body = {"body": base64.b64encode(
+                pickle.dumps((tuple(args), kwargs or {}))).decode()}
+        else:
+            body = {"args": list(args), "kwargs": kwargs or {}}
+        streamer = None
+        if stream_logs:
+            streamer = LogStreamer(self.base_url, rid).start()
+        try:
+            r = shared_client().post(
+                url, json=body, params=params, headers=headers,
+                timeout=timeout or C.HTTP_TIMEOUT,
+            )
+        finally:
+            if streamer:
+                streamer.stop()
+        try:
+            data = r.json()
+        except json.JSONDecodeError:
+            r.raise_for_status()
+            raise
+        if r.status_code != 200 or "error" in data:
+            raise reconstruct_exception(data.get("error", {"message": r.text}))
+        if "result_pickle" in data:
+            return pickle.loads(base64.b64decode(data["result_pickle"]))
+        if serialization == "pickle":
+            return pickle.loads(base64.b64decode(data["result"]))
+        return data["result"]
+
+    def reload(self, metadata, launch_id):
+        r = shared_client().post(
+            self.base_url + "/reload",
+            json={"metadata": metadata, "launch_id": launch_id},
+            timeout=C.LAUNCH_TIMEOUT,
+        )
+        data = r.json()
+        if r.status_code != 200:
+            raise reconstruct_exception(data.get("error", {}))
+        return data
+
+    def logs(self, since=0, request_id=None, limit=1000):
+        r = shared_client().get(
+            self.base_url + "/logs/tail",
+            params={"since": since, "limit": limit,
+                    **({"request_id": request_id} if request_id else {})},
+        )
+        return r.json().get("entries", [])

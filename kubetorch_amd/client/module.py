@@ -1,0 +1,160 @@
+"""Module: shared deploy/call machinery for Fn/Cls/App.
+
+Deploy path (reference parity: module.py:486 Module.to + compute.py:2006):
+  1. name resolution with username prefix
+  2. sync the working dir to the data store (k8s) — skipped in local mode
+  3. build module metadata (pointers + dispatch + procs)
+  4. controller /deploy: apply manifest, hot-reload warm pods, ack barrier
+  5. wait for /ready?launch_id on the service
+The hot loop (re-.to() with changed code) reuses warm pods: rsync delta +
+reload broadcast, no pod restart."""
+import base64
+import pickle
+import re
+import time
+import uuid
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.client.http_client import HTTPClient, shared_client
+from kubetorch_amd.compute.compute import Compute
+from kubetorch_amd.config import config
+from kubetorch_amd.exceptions import LaunchError
+from kubetorch_amd.globals import controller_client
+
+
+def sanitize_name(name):
+    name = re.sub(r"[^a-z0-9-]", "-", name.lower()).strip("-")
+    return name[:63]
+
+
+class Module:
+    module_type = "fn"
+
+    def __init__(self, pointers, name=None, compute=None, init_args=None):
+        self.pointers = pointers
+        self._name = name
+        self.compute = compute
+        self.init_args = init_args
+        self.service_hosts = []
+        self.launch_id = None
+        self._http = None
+        self.stream_logs = bool(config.get("stream_logs"))
+
+    @property
+    def name(self):
+        base = self._name or self.pointers["name"]
+        return sanitize_name(f"{config.username}-{base}")
+
+    @property
+    def namespace(self):
+        return (self.compute.namespace if self.compute else config.namespace)
+
+    # -- deploy ----------------------------------------------------------------
+    def metadata(self):
+        md = {
+            "module_name": self.name,
+            "callable_name": self.pointers["name"],
+            "module_type": self.module_type,
+            "file_path": self.pointers["file_path"],
+            "rel_path": self.pointers.get("rel_path"),
+            "project_root": self.pointers["project_root"],
+            "workdir_key": f"{self.namespace}/{self.name}/workdir",
+            "distributed_config": (self.compute.distributed_config
+                                   if self.compute else None),
+            "image_setup": (self.compute.image_setup_contents()
+                            if self.compute else ""),
+        }
+        if self.init_args is not None:
+            md["init_args"] = base64.b64encode(
+                pickle.dumps(self.init_args)).decode()
+        return md
+
+    def _sync_workdir(self, md):
+        if self.compute is not None and self.compute.local:
+            return  # same filesystem: pods import the original paths
+        try:
+            from kubetorch_amd.data_store import commands as ds
+
+            ds.put(md["workdir_key"], src=self.pointers["project_root"])
+        except Exception:
+            pass  # no data store deployed: pods must have the code baked in
+
+    def to(self, compute: Compute = None, init_args=None):
+        """Deploy (or hot-reload) this module onto the compute."""
+        if compute is not None:
+            self.compute = compute
+        if init_args is not None:
+            self.init_args = init_args
+        if self.compute is None:
+            self.compute = Compute(cpus=1)
+        t0 = time.time()
+        md = self.metadata()
+        self._sync_workdir(md)
+        launch_id = uuid.uuid4().hex[:12]
+        manifest = self.compute.to_manifest(self.name, username=config.username,
+                                            module=self.pointers["name"])
+        resp = controller_client().deploy(
+            name=self.name, namespace=self.namespace, manifest=manifest,
+            metadata=md, launch_id=launch_id,
+            service_config={"kind": self.compute.kind},
+            timeout=self.compute.launch_timeout,
+        )
+        self.launch_id = resp.get("launch_id", launch_id)
+        self.service_hosts = resp.get("hosts") or []
+        self._http = None
+        self._wait_ready(timeout=self.compute.launch_timeout,
+                         reloaded=bool(resp.get("reloaded_pods")))
+        elapsed = time.time() - t0
+        if self.stream_logs:
+            print(f"[kt] {self.name} ready in {elapsed:.2f}s "
+                  f"({len(self.service_hosts) or self.compute.replicas} pods)")
+        return self
+
+    async def to_async(self, compute=None, init_args=None):
+        import asyncio
+
+        return await asyncio.to_thread(self.to, compute, init_args)
+
+    def _base_url(self):
+        from kubetorch_amd.globals import service_url
+
+        return service_url(self.name, self.namespace, self.service_hosts)
+
+    @property
+    def http(self) -> HTTPClient:
+        if self._http is None:
+            self._http = HTTPClient(self._base_url(), self.pointers["name"])
+        return self._http
+
+    def _wait_ready(self, timeout=C.LAUNCH_TIMEOUT, reloaded=False):
+        """Poll /ready?launch_id until the pod finished loading this deploy.
+        If pods were hot-reloaded through the controller ack barrier, they are
+        ready by construction, but poll once to verify."""
+        deadline = time.time() + timeout
+        delay = 0.05
+        while time.time() < deadline:
+            if self.http.is_ready(launch_id=None if not reloaded else self.launch_id):
+                # accept pods that don't carry a launch_id (direct-env launch)
+                return
+            time.sleep(delay)
+            delay = min(delay * 1.5, 2.0)
+        raise LaunchError(f"service {self.name} not ready after {timeout}s")
+
+    # -- lifecycle ---------------------------------------------------------------
+    def teardown(self):
+        controller_client().delete_workload(self.name, self.namespace)
+        self.service_hosts = []
+        self._http = None
+
+    def logs(self, since=0, limit=1000):
+        return self.http.logs(since=since, limit=limit)
+
+    def workload(self):
+        return controller_client().get_workload(self.name, self.namespace)
+
+    def _call(self, args, kwargs, method=None, **opts):
+        return self.http.call(
+            args=args, kwargs=kwargs, method=method,
+            stream_logs=opts.pop("stream_logs", self.stream_logs),
+            **opts,
+        )

@@ -1,0 +1,137 @@
+"""Compute: the resource spec a Module deploys onto.
+
+MI355X-first: `gpus=` maps to amd.com/gpu, `gpu_type=` to the AMD device
+plugin's product label, worker image defaults to PyTorch-ROCm. (Reference
+parity: resources/compute/compute.py — 2798 LoC there; the manifest
+mechanics live in provisioning/manifests.py here.)"""
+import copy
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.compute.autoscaling import AutoscalingConfig
+from kubetorch_amd.compute.images import DEFAULT_PYTORCH_ROCM
+from kubetorch_amd.config import config
+from kubetorch_amd.provisioning import manifests as M
+
+
+class Compute:
+    def __init__(self, cpus=None, memory=None, gpus=0, gpu_type=None,
+                 image=None, env=None, namespace=None, volumes=None,
+                 secrets=None, shared_memory="8Gi", node_selector=None,
+                 annotations=None, inactivity_ttl=None, queue=None,
+                 launch_timeout=C.LAUNCH_TIMEOUT, endpoint=None,
+                 local=None):
+        self.cpus = cpus
+        self.memory = memory
+        self.gpus = gpus
+        self.gpu_type = gpu_type
+        self.image = image  # Image object or image id string
+        self.env = dict(env or {})
+        self.namespace = namespace or config.namespace
+        self.volumes = list(volumes or [])
+        self.secrets = list(secrets or [])
+        self.shared_memory = shared_memory
+        self.node_selector = node_selector
+        self.annotations = dict(annotations or {})
+        self.inactivity_ttl = inactivity_ttl
+        self.queue = queue
+        self.launch_timeout = launch_timeout
+        self.endpoint = endpoint
+        self.local = config.local_mode if local is None else local
+        self.kind = "deployment"
+        self.replicas = 1
+        self.distributed_config = None
+        self.autoscaling = None
+        self._raw_manifest = None
+
+    # -- fluent configuration --------------------------------------------------
+    def distribute(self, framework="pytorch", workers=1, num_proc=None,
+                   quorum_timeout=C.QUORUM_TIMEOUT):
+        """Configure the SPMD launcher: `workers` pods x `num_proc` local
+        ranks (auto = one per MI355X GPU via torch.cuda.device_count())."""
+        if framework == "ray":
+            self.kind = "raycluster"
+            self.replicas = workers
+            self.distributed_config = {"type": "ray", "workers": workers}
+            return self
+        self.replicas = workers
+        self.distributed_config = {
+            "type": framework,
+            "workers": workers,
+            "num_proc": num_proc,
+            "quorum_timeout": quorum_timeout,
+        }
+        return self
+
+    def autoscale(self, autoscaling=None, **kw):
+        self.kind = "knative"
+        self.autoscaling = autoscaling or AutoscalingConfig(**kw)
+        return self
+
+    @classmethod
+    def from_manifest(cls, manifest, namespace=None, **kw):
+        """BYO compute: a prebuilt manifest (Deployment / PyTorchJob / ...)."""
+        comp = cls(namespace=namespace or manifest.get("metadata", {})
+                   .get("namespace", "default"), **kw)
+        comp._raw_manifest = copy.deepcopy(manifest)
+        kind = manifest.get("kind", "Deployment").lower()
+        comp.kind = kind
+        if kind == "pytorchjob":
+            specs = manifest.get("spec", {}).get("pytorchReplicaSpecs", {})
+            workers = sum(s.get("replicas", 0) for s in specs.values())
+            if workers > 1:
+                comp.distributed_config = {
+                    "type": "pytorch", "workers": workers,
+                    "num_proc": int(manifest["spec"].get("nprocPerNode", 1) or 1),
+                }
+        return comp
+
+    # -- manifest --------------------------------------------------------------
+    def image_id(self):
+        if self.image is None:
+            return config.get("image") or DEFAULT_PYTORCH_ROCM
+        return getattr(self.image, "image_id", None) or str(self.image)
+
+    def pod_env(self):
+        env = dict(self.env)
+        if self.inactivity_ttl:
+            env["KT_INACTIVITY_TTL"] = str(self.inactivity_ttl)
+        return env
+
+    def to_manifest(self, service_name, username=None, module=None):
+        if self._raw_manifest is not None:
+            return copy.deepcopy(self._raw_manifest)
+        pod_kw = dict(
+            env=self.pod_env(), cpus=self.cpus, memory=self.memory,
+            gpus=self.gpus, gpu_type=self.gpu_type,
+            shared_memory=self.shared_memory, volumes=self.volumes,
+            secrets=self.secrets, node_selector=self.node_selector,
+        )
+        ann = dict(self.annotations)
+        if self.inactivity_ttl:
+            ann[C.INACTIVITY_TTL_ANNOTATION] = str(self.inactivity_ttl)
+        if self.kind == "knative":
+            return M.build_knative_manifest(
+                service_name, self.namespace, self.image_id(),
+                autoscaling=self.autoscaling, username=username,
+                annotations=ann, **pod_kw)
+        if self.kind == "raycluster":
+            return M.build_raycluster_manifest(
+                service_name, self.namespace, self.image_id(),
+                workers=max(0, self.replicas - 1), username=username, **pod_kw)
+        if self.kind == "pytorchjob":
+            dc = self.distributed_config or {}
+            return M.build_pytorchjob_manifest(
+                service_name, self.namespace, self.image_id(),
+                workers=dc.get("workers", 1),
+                num_proc=dc.get("num_proc") or 1,
+                username=username, queue=self.queue, **pod_kw)
+        return M.build_deployment_manifest(
+            service_name, self.namespace, self.image_id(),
+            replicas=self.replicas, username=username, module=module,
+            annotations=ann, queue=self.queue, **pod_kw)
+
+    def image_setup_contents(self):
+        img = self.image
+        if img is not None and hasattr(img, "contents"):
+            return img.contents()
+        return ""

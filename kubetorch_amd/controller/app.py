@@ -1,0 +1,228 @@
+"""kubetorch_amd controller: workload registry + pod push hub + TTL reaper.
+
+The reference ships this as a closed-source image
+(ghcr.io/run-house/kubetorch-controller); this is a
+from-scratch implementation. Endpoints:
+
+    POST /controller/deploy           apply manifest + register + reload pods
+    POST /controller/workload         register-only (BYO / selector-only)
+    GET  /controller/workloads/{ns}
+    GET|DELETE /controller/workload/{ns}/{name}
+    POST /controller/pods/stream      pod registration -> NDJSON push stream
+    POST /controller/pods/reload_ack  ack barrier for hot reload
+    GET  /controller/debug/connections
+    GET  /health
+
+Push semantics: one asyncio queue per connected pod; deploy broadcasts
+{action: reload, metadata, launch_id} to the service's pods and waits for
+every ack before returning (reference: provisioning/design.md:59-103).
+Single-process by design — the hub state is in-memory (the reference's
+controller has the same 1-worker constraint, design.md:372).
+"""
+import asyncio
+import json
+import os
+import time
+import uuid
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.controller.drivers import K8sDriver, LocalDriver
+
+
+class Hub:
+    def __init__(self):
+        self.workloads = {}       # (ns, name) -> workload dict
+        self.pod_queues = {}      # pod_name -> asyncio.Queue
+        self.pod_info = {}        # pod_name -> registration dict
+        self.acks = {}            # launch_id -> {"pending": set, "event": Event}
+        mode = os.environ.get("KT_CONTROLLER_DRIVER", "auto")
+        k8s = K8sDriver()
+        if mode == "local" or (mode == "auto" and not k8s.available()):
+            self.driver = LocalDriver()
+            self.driver_name = "local"
+        else:
+            self.driver = k8s
+            self.driver_name = "k8s"
+
+    def set_url(self, url):
+        if isinstance(self.driver, LocalDriver):
+            self.driver.controller_url = url
+
+    def pods_for_service(self, service, namespace):
+        return [p for p, info in self.pod_info.items()
+                if info.get("service_name") == service
+                and info.get("namespace") == namespace]
+
+
+HUB = Hub()
+app = FastAPI()
+
+
+@app.get("/health")
+def health():
+    return {"status": "ok", "driver": HUB.driver_name}
+
+
+@app.post("/controller/deploy")
+async def deploy(request: Request):
+    """Apply the manifest and hot-reload the workload's pods.
+    Body: {namespace, name, manifest, metadata, service_config}."""
+    body = await request.json()
+    ns = body.get("namespace", "default")
+    name = body["name"]
+    manifest = body.get("manifest") or {}
+    metadata = body.get("metadata") or {}
+    launch_id = body.get("launch_id") or uuid.uuid4().hex[:12]
+
+    key = (ns, name)
+    existing = HUB.workloads.get(key)
+    HUB.workloads[key] = {
+        "name": name,
+        "namespace": ns,
+        "manifest": manifest,
+        "metadata": metadata,
+        "service_config": body.get("service_config") or {},
+        "launch_id": launch_id,
+        "created": existing["created"] if existing else time.time(),
+        "updated": time.time(),
+    }
+
+    hosts = []
+    if manifest:
+        try:
+            hosts = await asyncio.to_thread(
+                HUB.driver.apply, manifest, ns, metadata, launch_id
+            )
+        except Exception as e:
+            return JSONResponse(
+                {"error": f"driver apply failed: {e}"}, status_code=500
+            )
+
+    # hot-reload connected pods of this service and wait for acks
+    pods = HUB.pods_for_service(name, ns)
+    acked = []
+    if pods:
+        ev = asyncio.Event()
+        HUB.acks[launch_id] = {"pending": set(pods), "event": ev}
+        msg = {"action": "reload", "metadata": metadata, "launch_id": launch_id}
+        for p in pods:
+            await HUB.pod_queues[p].put(msg)
+        try:
+            await asyncio.wait_for(ev.wait(), timeout=C.RELOAD_ACK_TIMEOUT)
+            acked = pods
+        except asyncio.TimeoutError:
+            acked = [p for p in pods
+                     if p not in HUB.acks[launch_id]["pending"]]
+        finally:
+            HUB.acks.pop(launch_id, None)
+
+    return {"ok": True, "launch_id": launch_id, "hosts": hosts,
+            "reloaded_pods": acked, "driver": HUB.driver_name}
+
+
+@app.post("/controller/workload")
+async def register_workload(request: Request):
+    body = await request.json()
+    ns = body.get("namespace", "default")
+    name = body["name"]
+    HUB.workloads[(ns, name)] = {
+        "name": name, "namespace": ns,
+        "manifest": body.get("manifest") or {},
+        "metadata": body.get("metadata") or {},
+        "service_config": body.get("service_config") or {},
+        "launch_id": body.get("launch_id"),
+        "created": time.time(), "updated": time.time(),
+    }
+    return {"ok": True}
+
+
+@app.get("/controller/workloads/{ns}")
+def list_workloads(ns: str):
+    return {"workloads": [w for (n, _), w in HUB.workloads.items() if n == ns]}
+
+
+@app.get("/controller/workload/{ns}/{name}")
+def get_workload(ns: str, name: str):
+    w = HUB.workloads.get((ns, name))
+    if not w:
+        return JSONResponse({"error": "not found"}, status_code=404)
+    pods = HUB.driver.pods(name, ns) if hasattr(HUB.driver, "pods") else []
+    return {**w, "pods": pods}
+
+
+@app.delete("/controller/workload/{ns}/{name}")
+async def delete_workload(ns: str, name: str):
+    HUB.workloads.pop((ns, name), None)
+    try:
+        await asyncio.to_thread(HUB.driver.delete, name, ns)
+    except Exception:
+        pass
+    return {"ok": True}
+
+
+@app.post("/controller/pods/stream")
+async def pod_stream(request: Request):
+    """Pod registration; the response is an endless NDJSON stream of pushes.
+    First push is the current module metadata (request_metadata contract)."""
+    reg = await request.json()
+    pod_name = reg["pod_name"]
+    q = asyncio.Queue()
+    HUB.pod_queues[pod_name] = q
+    HUB.pod_info[pod_name] = reg
+
+    if reg.get("request_metadata"):
+        w = HUB.workloads.get((reg.get("namespace", "default"),
+                               reg.get("service_name")))
+        if w:
+            await q.put({"action": "metadata", "metadata": w["metadata"],
+                         "launch_id": w.get("launch_id")})
+
+    async def gen():
+        try:
+            while True:
+                try:
+                    msg = await asyncio.wait_for(q.get(), timeout=15.0)
+                except asyncio.TimeoutError:
+                    msg = {"action": "ping"}
+                yield json.dumps(msg) + "\n"
+        finally:
+            HUB.pod_queues.pop(pod_name, None)
+            HUB.pod_info.pop(pod_name, None)
+
+    return StreamingResponse(gen(), media_type="application/x-ndjson")
+
+
+@app.post("/controller/pods/reload_ack")
+async def reload_ack(request: Request):
+    body = await request.json()
+    rec = HUB.acks.get(body.get("launch_id"))
+    if rec:
+        rec["pending"].discard(body.get("pod_name"))
+        if not rec["pending"]:
+            rec["event"].set()
+    return {"ok": True}
+
+
+@app.get("/controller/debug/connections")
+def debug_connections():
+    return {"pods": list(HUB.pod_info.values()),
+            "workloads": [f"{ns}/{n}" for ns, n in HUB.workloads]}
+
+
+def main():
+    import argparse
+
+    import uvicorn
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=C.CONTROLLER_PORT)
+    ap.add_argument("--host", default="0.0.0.0")
+    args = ap.parse_args()
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

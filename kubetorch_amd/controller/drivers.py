@@ -1,0 +1,167 @@
+"""Controller drivers: how workload manifests become running pods.
+
+* K8sDriver  — applies manifests through kubectl / the K8s API (in-cluster).
+* LocalDriver — "pods" are local http_server subprocesses on localhost
+  ports. This is the fake-cluster mode that makes the whole control plane
+  testable end-to-end without Kubernetes (and the `kt` local dev story);
+  the reference only fakes the supervisor layer (LOCAL_IPS), we fake the
+  pod layer too.
+"""
+import json
+import os
+import shutil
+import socket
+import subprocess
+import sys
+import time
+
+from kubetorch_amd import constants as C
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+class LocalPod:
+    def __init__(self, name, port, proc):
+        self.name = name
+        self.port = port
+        self.proc = proc
+
+    @property
+    def host(self):
+        return f"127.0.0.1:{self.port}"
+
+    def alive(self):
+        return self.proc.poll() is None
+
+    def kill(self):
+        if self.alive():
+            self.proc.terminate()
+            try:
+                self.proc.wait(5)
+            except subprocess.TimeoutExpired:
+                self.proc.kill()
+
+
+class LocalDriver:
+    """Spawns one http_server subprocess per replica; KT_LOCAL_IPS carries
+    the peer list (discovery), KT_CONTROLLER_URL points back at us."""
+
+    def __init__(self, controller_url=None):
+        self.controller_url = controller_url
+        self.services = {}  # (ns, name) -> [LocalPod]
+
+    def apply(self, manifest, namespace, metadata=None, launch_id=None):
+        name = manifest["metadata"]["name"]
+        replicas = manifest.get("spec", {}).get("replicas", 1)
+        key = (namespace, name)
+        pods = self.services.get(key, [])
+        alive = [p for p in pods if p.alive()]
+        if len(alive) == replicas:
+            return [p.host for p in alive]  # warm pods: reload only
+        for p in pods:
+            p.kill()
+        ports = [_free_port() for _ in range(replicas)]
+        peer_list = ",".join(f"127.0.0.1:{p}" for p in ports)
+        master_port = _free_port()  # one rendezvous port for the whole service
+        new_pods = []
+        md = metadata or {}
+        for i, port in enumerate(ports):
+            env = dict(os.environ)
+            env.update({
+                "KT_SERVER_PORT": str(port),
+                "KT_SELF_HOST": f"127.0.0.1:{port}",
+                C.ENV_LOCAL_IPS: peer_list,
+                C.ENV_SERVICE_NAME: name,
+                "POD_NAME": f"{name}-{i}",
+                "POD_NAMESPACE": namespace,
+                "POD_IP": "127.0.0.1",
+                "KT_MASTER_PORT": str(master_port),
+            })
+            if launch_id:
+                env[C.ENV_LAUNCH_ID] = str(launch_id)
+            if self.controller_url:
+                env["KT_CONTROLLER_URL"] = self.controller_url
+            for k, v in (md.get("env") or {}).items():
+                env[k] = str(v)
+            # module metadata -> env contract (same as controller push)
+            if md.get("file_path"):
+                env[C.ENV_FILE_PATH] = md["file_path"]
+            if md.get("project_root"):
+                env[C.ENV_PROJECT_ROOT] = md["project_root"]
+            if md.get("callable_name"):
+                env[C.ENV_CALLABLE_NAME] = md["callable_name"]
+                env[C.ENV_MODULE_NAME] = md.get("module_name", md["callable_name"])
+            if md.get("module_type"):
+                env[C.ENV_MODULE_TYPE] = md["module_type"]
+            if md.get("init_args"):
+                env[C.ENV_INIT_ARGS] = md["init_args"]
+            if md.get("distributed_config"):
+                env[C.ENV_DISTRIBUTED_CONFIG] = json.dumps(md["distributed_config"])
+            proc = subprocess.Popen(
+                [sys.executable, "-m", "kubetorch_amd.serving.http_server",
+                 "--port", str(port), "--host", "127.0.0.1"],
+                env=env,
+                stdout=subprocess.DEVNULL,
+                stderr=subprocess.DEVNULL,
+            )
+            new_pods.append(LocalPod(f"{name}-{i}", port, proc))
+        self.services[key] = new_pods
+        return [p.host for p in new_pods]
+
+    def delete(self, name, namespace):
+        for p in self.services.pop((namespace, name), []):
+            p.kill()
+
+    def pods(self, name, namespace):
+        return [p.host for p in self.services.get((namespace, name), [])
+                if p.alive()]
+
+    def teardown_all(self):
+        for key in list(self.services):
+            self.delete(key[1], key[0])
+
+
+class K8sDriver:
+    """Applies manifests via kubectl (the controller pod has RBAC for this;
+    reference ships this logic inside its closed-source controller image)."""
+
+    def __init__(self, kubectl="kubectl"):
+        self.kubectl = kubectl
+
+    def available(self):
+        return shutil.which(self.kubectl) is not None
+
+    def apply(self, manifest, namespace, metadata=None, launch_id=None):
+        payload = json.dumps(manifest)
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "apply", "-f", "-"],
+            input=payload.encode(), check=True, capture_output=True,
+        )
+        return []
+
+    def delete(self, name, namespace, kind="deployment"):
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "delete", kind, name,
+             "--ignore-not-found"],
+            check=True, capture_output=True,
+        )
+
+    def pods(self, name, namespace):
+        out = subprocess.run(
+            [self.kubectl, "-n", namespace, "get", "pods", "-l",
+             f"{C.SERVICE_LABEL}={name}", "-o", "json"],
+            check=True, capture_output=True,
+        )
+        items = json.loads(out.stdout).get("items", [])
+        hosts = []
+        for it in items:
+            ip = it.get("status", {}).get("podIP")
+            if ip:
+                hosts.append(f"{ip}:{C.SERVER_PORT}")
+        return hosts

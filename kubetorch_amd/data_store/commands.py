@@ -1,0 +1,172 @@
+"""kt.put / kt.get / kt.ls / kt.rm — the user-facing data-store facade.
+
+Routes by payload type (reference parity: data_store/data_store_cmds.py):
+  * GPU tensors / state dicts -> gpu_store (hipIpc registration + RCCL
+    transfer through the per-node pod-data-server)
+  * filesystem paths          -> file store (HTTP store service, or the
+    local directory store when no store service is configured)
+"""
+import io
+import os
+import shutil
+import tarfile
+
+import httpx
+
+from kubetorch_amd import constants as C
+
+LOCAL_STORE_ROOT = os.environ.get(
+    "KT_STORE_ROOT", os.path.expanduser("~/.ktamd/store"))
+
+
+def _store_url():
+    return os.environ.get("KT_STORE_URL")
+
+
+def _is_gpu_data(obj):
+    try:
+        import torch
+
+        if isinstance(obj, torch.Tensor):
+            return obj.is_cuda
+        if isinstance(obj, dict) and obj:
+            vals = list(obj.values())
+            return all(isinstance(v, torch.Tensor) and v.is_cuda for v in vals)
+    except ImportError:
+        pass
+    return False
+
+
+def _local_path(key):
+    p = os.path.abspath(os.path.join(LOCAL_STORE_ROOT, key.strip("/")))
+    if not p.startswith(os.path.abspath(LOCAL_STORE_ROOT)):
+        raise ValueError("key escapes store root")
+    return p
+
+
+def put(key, src, window=None, timeout=C.HTTP_TIMEOUT):
+    """Store a file/dir (by path) or publish GPU tensors under `key`."""
+    if _is_gpu_data(src):
+        from kubetorch_amd.data_store import gpu_store
+
+        return gpu_store.put(key, src, window=window)
+    src = os.path.expanduser(str(src))
+    if not os.path.exists(src):
+        raise FileNotFoundError(src)
+    url = _store_url()
+    if url is None:
+        dest = _local_path(key)
+        if os.path.isdir(src):
+            if os.path.exists(dest):
+                shutil.rmtree(dest)
+            shutil.copytree(src, dest,
+                            ignore=shutil.ignore_patterns(
+                                "__pycache__", ".git", "*.pyc"))
+        else:
+            os.makedirs(os.path.dirname(dest), exist_ok=True)
+            shutil.copy2(src, dest)
+        return {"key": key, "locale": "local-dir"}
+    if os.path.isdir(src):
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w:gz") as tar:
+            tar.add(src, arcname=".",
+                    filter=lambda ti: None if "__pycache__" in ti.name
+                    or "/.git/" in ti.name or ti.name.endswith(".pyc") else ti)
+        r = httpx.put(f"{url}/files/{key}", content=buf.getvalue(),
+                      headers={"X-KT-Tar": "1"}, timeout=timeout)
+    else:
+        with open(src, "rb") as f:
+            r = httpx.put(f"{url}/files/{key}", content=f.read(), timeout=timeout)
+    r.raise_for_status()
+    return {"key": key, "locale": "store"}
+
+
+def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
+    """Fetch a stored file/dir to `dest` path, or receive GPU tensors into
+    `dest` tensors/state dict."""
+    if dest is not None and _is_gpu_data(dest):
+        from kubetorch_amd.data_store import gpu_store
+
+        return gpu_store.get(key, dest, window=window)
+    url = _store_url()
+    if url is None:
+        srcp = _local_path(key)
+        if not os.path.exists(srcp):
+            raise KeyError(f"no such key: {key}")
+        if dest is None:
+            return srcp
+        dest = os.path.expanduser(dest)
+        if os.path.isdir(srcp):
+            if os.path.exists(dest):
+                shutil.rmtree(dest)
+            shutil.copytree(srcp, dest)
+        else:
+            os.makedirs(os.path.dirname(dest) or ".", exist_ok=True)
+            shutil.copy2(srcp, dest)
+        return dest
+    r = httpx.get(f"{url}/files/{key}", timeout=timeout)
+    if r.status_code == 404:
+        raise KeyError(f"no such key: {key}")
+    r.raise_for_status()
+    dest = os.path.expanduser(dest or os.path.basename(key))
+    if r.headers.get("X-KT-Tar") == "1":
+        os.makedirs(dest, exist_ok=True)
+        with tarfile.open(fileobj=io.BytesIO(r.content), mode="r:gz") as tar:
+            tar.extractall(dest)  # noqa: S202
+    else:
+        os.makedirs(os.path.dirname(dest) or ".", exist_ok=True)
+        with open(dest, "wb") as f:
+            f.write(r.content)
+    return dest
+
+
+def ls(prefix=""):
+    url = _store_url()
+    if url is None:
+        base = _local_path(prefix) if prefix else LOCAL_STORE_ROOT
+        out = []
+        if os.path.isdir(base):
+            for root, _d, files in os.walk(base):
+                for f in files:
+                    full = os.path.join(root, f)
+                    out.append({"key": os.path.relpath(full, LOCAL_STORE_ROOT),
+                                "size": os.path.getsize(full)})
+        return out
+    r = httpx.get(f"{url}/ls", params={"prefix": prefix})
+    r.raise_for_status()
+    return r.json()["entries"]
+
+
+def rm(key):
+    url = _store_url()
+    if url is None:
+        p = _local_path(key)
+        if os.path.isdir(p):
+            shutil.rmtree(p, ignore_errors=True)
+        elif os.path.exists(p):
+            os.remove(p)
+        return {"ok": True}
+    r = httpx.delete(f"{url}/files/{key}")
+    r.raise_for_status()
+    return r.json()
+
+
+def sync_workdir_from_store():
+    """Pod-side: pull the synced working dir for this workload and remap the
+    callable pointers to it. Returns the local root or None."""
+    key = os.environ.get("KT_WORKDIR_KEY")
+    if not key:
+        return None
+    if _store_url() is None and not os.path.isdir(_local_path(key)):
+        return None  # local mode without a store: original paths are valid
+    root = os.environ.get("KT_WORKDIR_BASE", "/workdir")
+    dest = os.path.join(root, key.strip("/").replace("/", "_"))
+    try:
+        get(key, dest)
+    except KeyError:
+        return None
+    rel = os.environ.get("KT_REL_PATH")
+    if rel:
+        os.environ[C.ENV_PROJECT_ROOT] = dest
+        os.environ[C.ENV_FILE_PATH] = os.path.join(dest, rel)
+    return dest

@@ -1,0 +1,178 @@
+"""Namespace data-store service: file store + key metadata + log store.
+
+One pod per namespace (reference runs rsyncd+metadata+Loki images; this is a
+single from-scratch FastAPI service):
+    PUT  /files/{key...}       upload (tar stream or single file)
+    GET  /files/{key...}       download
+    GET  /ls?prefix=           list keys
+    DELETE /files/{key...}
+    POST /meta/{key...}        publish key metadata (e.g. GPU source ip)
+    GET  /meta/{key...}
+    DELETE /meta/{key...}
+    POST /logs/push            pod log batches (ring per service)
+    GET  /logs/tail            query by service/request_id
+"""
+import io
+import json
+import os
+import tarfile
+import threading
+import time
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, Response, StreamingResponse
+
+from kubetorch_amd import constants as C
+
+DATA_ROOT = os.environ.get("KT_STORE_ROOT",
+                           os.path.expanduser("~/.ktamd/store"))
+
+app = FastAPI()
+_meta = {}
+_meta_lock = threading.Lock()
+_logs = {}
+_logs_lock = threading.Lock()
+_LOG_RING = 50000
+
+
+def _path_for(key):
+    p = os.path.abspath(os.path.join(DATA_ROOT, key.strip("/")))
+    if not p.startswith(os.path.abspath(DATA_ROOT)):
+        raise ValueError("key escapes store root")
+    return p
+
+
+@app.get("/health")
+def health():
+    return {"status": "ok", "root": DATA_ROOT}
+
+
+@app.put("/files/{key:path}")
+async def put_file(key: str, request: Request):
+    path = _path_for(key)
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    body = await request.body()
+    if request.headers.get("X-KT-Tar") == "1":
+        os.makedirs(path, exist_ok=True)
+        with tarfile.open(fileobj=io.BytesIO(body), mode="r:gz") as tar:
+            tar.extractall(path)  # noqa: S202 - trusted in-cluster clients
+    else:
+        with open(path, "wb") as f:
+            f.write(body)
+    return {"ok": True, "bytes": len(body)}
+
+
+@app.get("/files/{key:path}")
+def get_file(key: str):
+    path = _path_for(key)
+    if os.path.isdir(path):
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w:gz") as tar:
+            tar.add(path, arcname=".")
+        return Response(buf.getvalue(), media_type="application/gzip",
+                        headers={"X-KT-Tar": "1"})
+    if not os.path.exists(path):
+        return JSONResponse({"error": "not found"}, status_code=404)
+
+    def stream():
+        with open(path, "rb") as f:
+            while chunk := f.read(1 << 20):
+                yield chunk
+
+    return StreamingResponse(stream(), media_type="application/octet-stream")
+
+
+@app.get("/ls")
+def ls(prefix: str = ""):
+    base = _path_for(prefix) if prefix else DATA_ROOT
+    out = []
+    if os.path.isdir(base):
+        for root, _dirs, files in os.walk(base):
+            for f in files:
+                full = os.path.join(root, f)
+                out.append({
+                    "key": os.path.relpath(full, DATA_ROOT),
+                    "size": os.path.getsize(full),
+                    "mtime": os.path.getmtime(full),
+                })
+    elif os.path.exists(base):
+        out.append({"key": prefix, "size": os.path.getsize(base),
+                    "mtime": os.path.getmtime(base)})
+    return {"entries": out}
+
+
+@app.delete("/files/{key:path}")
+def rm(key: str):
+    import shutil
+
+    path = _path_for(key)
+    if os.path.isdir(path):
+        shutil.rmtree(path, ignore_errors=True)
+    elif os.path.exists(path):
+        os.remove(path)
+    return {"ok": True}
+
+
+@app.post("/meta/{key:path}")
+async def put_meta(key: str, request: Request):
+    body = await request.json()
+    with _meta_lock:
+        _meta[key] = {**body, "updated": time.time()}
+    return {"ok": True}
+
+
+@app.get("/meta/{key:path}")
+def get_meta(key: str):
+    with _meta_lock:
+        m = _meta.get(key)
+    if m is None:
+        return JSONResponse({"error": "not found"}, status_code=404)
+    return m
+
+
+@app.delete("/meta/{key:path}")
+def del_meta(key: str):
+    with _meta_lock:
+        _meta.pop(key, None)
+    return {"ok": True}
+
+
+@app.post("/logs/push")
+async def logs_push(request: Request):
+    body = await request.json()
+    service = body.get("service", "unknown")
+    with _logs_lock:
+        ring = _logs.setdefault(service, [])
+        ring.extend(body.get("entries", []))
+        if len(ring) > _LOG_RING:
+            _logs[service] = ring[-_LOG_RING:]
+    return {"ok": True}
+
+
+@app.get("/logs/tail")
+def logs_tail(service: str, request_id: str = None, since_ts: float = 0,
+              limit: int = 1000):
+    with _logs_lock:
+        ring = list(_logs.get(service, []))
+    out = [e for e in ring if e.get("ts", 0) >= since_ts
+           and (request_id is None or e.get("request_id") == request_id)]
+    return {"entries": out[-limit:]}
+
+
+def main():
+    import argparse
+
+    import uvicorn
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=C.DATA_STORE_PORT)
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--root", default=DATA_ROOT)
+    args = ap.parse_args()
+    globals()["DATA_ROOT"] = args.root
+    os.makedirs(args.root, exist_ok=True)
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

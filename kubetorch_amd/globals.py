@@ -1,0 +1,109 @@
+"""Process-wide singletons: ControllerClient (retrying HTTP client of the
+controller API) and service URL resolution (in-cluster vs local vs
+port-forward). (Reference parity: python_client/kubetorch/globals.py.)"""
+import os
+import threading
+import time
+
+import httpx
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.config import config
+
+RETRY_STATUS = (502, 503)
+MAX_ATTEMPTS = 5
+
+
+class ControllerClient:
+    """HTTP client of the controller with retry/backoff on 502/503 and
+    connection errors (reference: globals.py:372-501)."""
+
+    def __init__(self, base_url=None):
+        self._base_url = base_url
+        self._client = httpx.Client(timeout=C.HTTP_TIMEOUT)
+
+    @property
+    def base_url(self):
+        if self._base_url:
+            return self._base_url
+        url = config.api_url
+        if url:
+            self._base_url = url.rstrip("/")
+            return self._base_url
+        # local mode: in-process controller
+        from kubetorch_amd.controller.local import ensure_local_controller
+
+        self._base_url = ensure_local_controller()
+        return self._base_url
+
+    def _request(self, method, path, **kw):
+        last = None
+        backoff = 0.5
+        for _ in range(MAX_ATTEMPTS):
+            try:
+                r = self._client.request(method, self.base_url + path, **kw)
+                if r.status_code in RETRY_STATUS:
+                    last = RuntimeError(f"{r.status_code}: {r.text[:200]}")
+                else:
+                    return r
+            except (httpx.ConnectError, httpx.ReadTimeout, httpx.RemoteProtocolError) as e:
+                last = e
+            time.sleep(backoff)
+            backoff = min(backoff * 2, 8)
+        raise RuntimeError(f"controller unreachable after {MAX_ATTEMPTS} attempts: {last}")
+
+    def deploy(self, name, namespace, manifest=None, metadata=None,
+               service_config=None, launch_id=None, timeout=None):
+        r = self._request(
+            "POST", "/controller/deploy",
+            json={"name": name, "namespace": namespace, "manifest": manifest,
+                  "metadata": metadata, "service_config": service_config,
+                  "launch_id": launch_id},
+            timeout=timeout or C.LAUNCH_TIMEOUT,
+        )
+        data = r.json()
+        if r.status_code != 200:
+            from kubetorch_amd.exceptions import LaunchError
+
+            raise LaunchError(str(data))
+        return data
+
+    def register_workload(self, name, namespace, **body):
+        return self._request(
+            "POST", "/controller/workload",
+            json={"name": name, "namespace": namespace, **body},
+        ).json()
+
+    def get_workload(self, name, namespace):
+        r = self._request("GET", f"/controller/workload/{namespace}/{name}")
+        return r.json() if r.status_code == 200 else None
+
+    def list_workloads(self, namespace):
+        return self._request("GET", f"/controller/workloads/{namespace}").json()
+
+    def delete_workload(self, name, namespace):
+        return self._request(
+            "DELETE", f"/controller/workload/{namespace}/{name}"
+        ).json()
+
+
+_controller = None
+_lock = threading.Lock()
+
+
+def controller_client() -> ControllerClient:
+    global _controller
+    with _lock:
+        if _controller is None:
+            _controller = ControllerClient()
+        return _controller
+
+
+def service_url(service_name, namespace, hosts=None):
+    """Resolve the base URL for a deployed service. Local mode / explicit
+    hosts -> first pod; in-cluster -> the K8s Service DNS."""
+    if hosts:
+        return f"http://{hosts[0]}"
+    if os.environ.get("KUBERNETES_SERVICE_HOST"):
+        return f"http://{service_name}.{namespace}.svc.cluster.local:{C.SERVER_PORT}"
+    return f"http://{service_name}.{namespace}:{C.SERVER_PORT}"

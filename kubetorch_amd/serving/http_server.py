@@ -1,0 +1,412 @@
+"""Worker pod HTTP server (FastAPI).
+
+Endpoints (reference parity: serving/http_server.py):
+    GET  /health                     liveness
+    GET  /ready?launch_id=X          503 until the reload for X completed
+    GET  /metrics                    prometheus exposition
+    GET  /logs/tail?since=&request_id=   ring-buffer tail (JSON)
+    WS   /logs/ws                    live log stream
+    POST /call/{name}[/{method}]     run the configured callable
+    POST /spmd/subcall               peer-to-peer distributed fan-in
+    POST /reload                     hot reload (also driven by controller WS)
+    GET  /app/status                 app-mode process status
+
+The server process never runs user code: calls route through the supervisor
+into spawn-method subprocesses (process_pool.py). Distributed calls fan out
+per the supervisor (supervisors.py).
+"""
+import asyncio
+import base64
+import json
+import os
+import pickle
+import threading
+import time
+import uuid
+from contextlib import asynccontextmanager
+
+from fastapi import FastAPI, Request, Response, WebSocket
+from fastapi.responses import JSONResponse
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.exceptions import package_exception
+from kubetorch_amd.serving import log_capture, metrics
+from kubetorch_amd.serving.supervisors import supervisor_factory
+
+STATE = {
+    "supervisor": None,
+    "lock": threading.Lock(),
+    "launch_id": None,
+    "started_at": None,
+    "terminating": False,
+    "app_proc": None,
+    "load_error": None,
+}
+
+
+def _distributed_config():
+    raw = os.environ.get(C.ENV_DISTRIBUTED_CONFIG)
+    if not raw:
+        return {}
+    try:
+        return json.loads(raw)
+    except json.JSONDecodeError:
+        return {}
+
+
+def get_supervisor(recreate=False):
+    with STATE["lock"]:
+        if STATE["supervisor"] is not None and not recreate:
+            return STATE["supervisor"]
+        if STATE["supervisor"] is not None:
+            try:
+                STATE["supervisor"].cleanup()
+            except Exception:
+                pass
+            STATE["supervisor"] = None
+        cfg = _distributed_config()
+        dtype = cfg.get("type")
+        kw = {}
+        if dtype not in (None, "", "local"):
+            kw = {
+                "num_workers": cfg.get("workers"),
+                "num_proc": cfg.get("num_proc"),
+                "quorum_timeout": cfg.get("quorum_timeout", C.QUORUM_TIMEOUT),
+            }
+        else:
+            kw = {"num_proc": cfg.get("num_proc") or 1}
+        try:
+            STATE["supervisor"] = supervisor_factory(dtype, **kw)
+            STATE["load_error"] = None
+        except Exception as e:
+            STATE["load_error"] = package_exception(e)
+            raise
+        return STATE["supervisor"]
+
+
+def apply_metadata(md: dict):
+    """Apply module metadata pushed by the controller (env-var contract)."""
+    mapping = {
+        "module_name": C.ENV_MODULE_NAME,
+        "callable_name": C.ENV_CALLABLE_NAME,
+        "file_path": C.ENV_FILE_PATH,
+        "project_root": C.ENV_PROJECT_ROOT,
+        "init_args": C.ENV_INIT_ARGS,
+        "module_type": C.ENV_MODULE_TYPE,
+        "service_name": C.ENV_SERVICE_NAME,
+        "service_dns": C.ENV_SERVICE_DNS,
+        "workdir_key": "KT_WORKDIR_KEY",
+        "rel_path": "KT_REL_PATH",
+    }
+    for key, env in mapping.items():
+        if key in md and md[key] is not None:
+            os.environ[env] = str(md[key])
+    if md.get("distributed_config") is not None:
+        os.environ[C.ENV_DISTRIBUTED_CONFIG] = json.dumps(md["distributed_config"])
+
+
+def do_reload(md: dict, launch_id=None):
+    """Hot reload: apply metadata, re-sync code, recreate supervisor.
+    The launch_id is only set after success so /ready gates correctly."""
+    apply_metadata(md or {})
+    try:
+        from kubetorch_amd.data_store import commands as ds
+
+        ds.sync_workdir_from_store()
+    except Exception:
+        pass
+    from kubetorch_amd.serving import loading
+
+    loading.clear_cache()
+    get_supervisor(recreate=True)
+    if launch_id:
+        STATE["launch_id"] = str(launch_id)
+        os.environ[C.ENV_LAUNCH_ID] = str(launch_id)
+
+
+# -- controller push channel (pod side) ---------------------------------------
+async def _controller_ws_loop(url):
+    """Register with the controller hub and stream metadata/reload pushes.
+    Implemented as a long-lived chunked-HTTP stream (NDJSON) rather than a
+    raw WebSocket: same push semantics + ack barrier, reconnect w/ backoff.
+    (Reference parity: http_server.py:206-497 ControllerWebSocket.)"""
+    import httpx
+
+    backoff = 1.0
+    pod = {
+        "pod_name": os.environ.get("POD_NAME", os.uname().nodename),
+        "pod_ip": os.environ.get("POD_IP", "127.0.0.1"),
+        "namespace": os.environ.get("POD_NAMESPACE", "default"),
+        "service_name": os.environ.get(C.ENV_SERVICE_NAME, ""),
+        "port": int(os.environ.get("KT_SERVER_PORT", C.SERVER_PORT)),
+        "request_metadata": True,
+    }
+    while not STATE["terminating"]:
+        try:
+            async with httpx.AsyncClient(timeout=None) as client:
+                async with client.stream("POST", url, json=pod) as resp:
+                    backoff = 1.0
+                    async for line in resp.aiter_lines():
+                        if not line.strip():
+                            continue
+                        msg = json.loads(line)
+                        action = msg.get("action")
+                        if action in ("metadata", "reload"):
+                            lid = msg.get("launch_id")
+                            await asyncio.to_thread(
+                                do_reload, msg.get("metadata", {}),
+                                lid if action == "reload" else None,
+                            )
+                            if action == "reload":
+                                ack_url = url.rsplit("/", 1)[0] + "/reload_ack"
+                                await client.post(ack_url, json={
+                                    "pod_name": pod["pod_name"],
+                                    "launch_id": lid,
+                                })
+        except Exception:
+            await asyncio.sleep(backoff)
+            backoff = min(backoff * 2, 30.0)
+
+
+@asynccontextmanager
+async def lifespan(app: FastAPI):
+    log_capture.install()
+    STATE["started_at"] = time.time()
+    pusher = metrics.MetricsPusher()
+    pusher.start()
+    stop_ev = threading.Event()
+    ws_task = None
+    controller_url = os.environ.get("KT_CONTROLLER_URL")
+    if controller_url:
+        ws_task = asyncio.create_task(
+            _controller_ws_loop(controller_url.rstrip("/") + "/controller/pods/stream")
+        )
+    # initial launch_id from env (set by the launcher/local driver)
+    STATE["launch_id"] = os.environ.get(C.ENV_LAUNCH_ID)
+    # eagerly build the supervisor if a callable is configured; startup
+    # errors are surfaced at call time, not crash time (reference behavior)
+    if os.environ.get(C.ENV_FILE_PATH):
+        try:
+            get_supervisor()
+        except Exception:
+            pass
+
+    def _drain_loop():
+        # follow the CURRENT supervisor's log queue (reloads swap the pool)
+        while not stop_ev.is_set():
+            sup = STATE.get("supervisor")
+            q = sup.pool.log_q if sup is not None else None
+            if q is None:
+                stop_ev.wait(0.5)
+                continue
+            try:
+                item = q.get(timeout=0.5)
+            except Exception:
+                continue
+            if item:
+                log_capture.RING.append(
+                    item.get("line", ""), source=item.get("source", "worker"),
+                    request_id=item.get("request_id"))
+
+    threading.Thread(target=_drain_loop, daemon=True).start()
+    yield
+    STATE["terminating"] = True
+    stop_ev.set()
+    if ws_task:
+        ws_task.cancel()
+    pusher.stop()
+    if STATE["supervisor"] is not None:
+        STATE["supervisor"].cleanup()
+
+
+app = FastAPI(lifespan=lifespan)
+
+
+@app.middleware("http")
+def metrics_middleware(request: Request, call_next):
+    async def inner():
+        t0 = time.time()
+        path = request.url.path.split("/")[1] or "root"
+        metrics.ACTIVE_REQUESTS.inc()
+        try:
+            resp = await call_next(request)
+            metrics.HTTP_REQUESTS.labels(path=path, status=resp.status_code).inc()
+            return resp
+        finally:
+            metrics.ACTIVE_REQUESTS.dec()
+            metrics.HTTP_DURATION.labels(path=path).observe(time.time() - t0)
+            if path in ("call", "spmd"):
+                metrics.touch_activity()
+
+    return inner()
+
+
+@app.get("/health")
+def health():
+    return {"status": "ok", "uptime": time.time() - (STATE["started_at"] or time.time())}
+
+
+@app.get("/ready")
+def ready(launch_id: str = None):
+    if STATE["terminating"]:
+        return JSONResponse({"ready": False, "reason": "terminating"}, status_code=503)
+    if launch_id and STATE["launch_id"] != launch_id:
+        return JSONResponse(
+            {"ready": False, "reason": f"launch {STATE['launch_id']} != {launch_id}"},
+            status_code=503,
+        )
+    return {"ready": True, "launch_id": STATE["launch_id"]}
+
+
+@app.get("/metrics")
+def metrics_route():
+    return Response(metrics.exposition(), media_type="text/plain; version=0.0.4")
+
+
+@app.get("/logs/tail")
+def logs_tail(since: int = 0, request_id: str = None, limit: int = 1000):
+    return {"entries": log_capture.RING.tail(since, request_id, limit),
+            "seq": log_capture.RING.seq}
+
+
+@app.websocket("/logs/ws")
+async def logs_ws(ws: WebSocket):
+    await ws.accept()
+    params = ws.query_params
+    since = int(params.get("since", 0))
+    request_id = params.get("request_id")
+    try:
+        while True:
+            entries = log_capture.RING.tail(since, request_id)
+            for e in entries:
+                await ws.send_json(e)
+                since = e["seq"] + 1
+            got = await asyncio.to_thread(log_capture.RING.wait_for, since, 5.0)
+            if not got:
+                await ws.send_json({"keepalive": True})
+    except Exception:
+        pass
+
+
+def _validate_name(name):
+    configured = os.environ.get(C.ENV_CALLABLE_NAME) or os.environ.get(C.ENV_MODULE_NAME)
+    return configured is None or name in (configured, os.environ.get(C.ENV_MODULE_NAME))
+
+
+async def _parse_call(request: Request):
+    ser = request.headers.get("X-Serialization", "json")
+    body = await request.body()
+    if ser == "pickle":
+        payload = json.loads(body)
+        return payload["body"], ser
+    payload = json.loads(body) if body else {}
+    args = payload.get("args", [])
+    kwargs = payload.get("kwargs", {})
+    return base64.b64encode(pickle.dumps((tuple(args), kwargs))).decode(), ser
+
+
+def _serialize_result(result, ser):
+    if ser == "pickle":
+        return {"result": base64.b64encode(pickle.dumps(result)).decode()}
+    try:
+        json.dumps(result)
+        return {"result": result}
+    except (TypeError, ValueError):
+        return {"result_pickle": base64.b64encode(pickle.dumps(result)).decode()}
+
+
+@app.post("/call/{name}")
+@app.post("/call/{name}/{method}")
+async def run_callable(name: str, request: Request, method: str = None,
+                       workers: str = None, restart_procs: bool = False):
+    if not _validate_name(name):
+        return JSONResponse(
+            {"error": {"error_type": "KeyError",
+                       "message": f"callable {name!r} not deployed here "
+                                  f"(configured: {os.environ.get(C.ENV_CALLABLE_NAME)})",
+                       "traceback": ""}},
+            status_code=404,
+        )
+    if STATE["load_error"]:
+        return JSONResponse({"error": STATE["load_error"]}, status_code=500)
+    rid = request.headers.get("X-Request-ID") or uuid.uuid4().hex
+    token = log_capture.request_id_var.set(rid)
+    try:
+        sup = get_supervisor()
+        body_b64, ser = await _parse_call(request)
+        sel = workers
+        if workers and workers not in ("all", "any", "ready"):
+            sel = json.loads(workers)
+        result = await asyncio.to_thread(
+            sup.call, serialized_body=body_b64, method=method,
+            workers=sel or "all", restart_procs=restart_procs,
+        ) if sup.distributed else await asyncio.to_thread(
+            _call_simple, sup, body_b64, method, rid,
+        )
+        return _serialize_result(result, ser)
+    except BaseException as e:  # noqa: BLE001
+        return JSONResponse({"error": package_exception(e)}, status_code=500)
+    finally:
+        log_capture.request_id_var.reset(token)
+
+
+def _call_simple(sup, body_b64, method, rid):
+    from kubetorch_amd.serving.supervisors import _decode_resp
+
+    resp = sup.pool.submit(0, body_b64, method=method, request_id=rid).result(
+        C.HTTP_TIMEOUT * 10
+    )
+    return _decode_resp(resp)
+
+
+@app.post("/spmd/subcall")
+async def spmd_subcall(request: Request):
+    payload = await request.json()
+    try:
+        sup = get_supervisor()
+        if "hosts" in payload and payload["hosts"]:
+            os.environ["KT_SPMD_HOSTS"] = json.dumps(payload["hosts"])
+        result = await asyncio.to_thread(
+            sup.call, serialized_body=payload["body"],
+            method=payload.get("method"), distributed_subcall=True,
+        )
+        return {"result": base64.b64encode(pickle.dumps(result)).decode()}
+    except BaseException as e:  # noqa: BLE001
+        return JSONResponse({"error": package_exception(e)}, status_code=500)
+
+
+@app.post("/reload")
+async def reload_route(request: Request):
+    payload = await request.json()
+    try:
+        await asyncio.to_thread(do_reload, payload.get("metadata", {}),
+                                payload.get("launch_id"))
+        return {"ok": True, "launch_id": STATE["launch_id"]}
+    except BaseException as e:  # noqa: BLE001
+        return JSONResponse({"error": package_exception(e)}, status_code=500)
+
+
+@app.get("/app/status")
+def app_status():
+    proc = STATE.get("app_proc")
+    if proc is None:
+        return {"running": False}
+    rc = proc.poll()
+    return {"running": rc is None, "returncode": rc}
+
+
+def main():
+    import argparse
+
+    import uvicorn
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int,
+                    default=int(os.environ.get("KT_SERVER_PORT", C.SERVER_PORT)))
+    ap.add_argument("--host", default="0.0.0.0")
+    args = ap.parse_args()
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

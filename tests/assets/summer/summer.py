@@ -1,0 +1,48 @@
+import os
+
+
+def summer(a, b):
+    print(f"summing {a}+{b}")
+    return a + b
+
+
+def rank_env():
+    return {
+        "rank": int(os.environ.get("RANK", "-1")),
+        "world_size": int(os.environ.get("WORLD_SIZE", "-1")),
+        "local_rank": int(os.environ.get("LOCAL_RANK", "-1")),
+        "node_rank": int(os.environ.get("NODE_RANK", "-1")),
+        "master_addr": os.environ.get("MASTER_ADDR"),
+        "pod_ips": os.environ.get("POD_IPS", ""),
+    }
+
+
+def boom():
+    raise ValueError("intentional failure")
+
+
+def gloo_allreduce(x):
+    """DDP-style allreduce through the SPMD env contract (gloo on CPU)."""
+    import torch
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        dist.init_process_group("gloo")
+    t = torch.tensor([float(x)])
+    dist.all_reduce(t)
+    out = t.item()
+    rank = dist.get_rank()
+    dist.destroy_process_group()
+    return {"rank": rank, "sum": out}
+
+
+class Counter:
+    def __init__(self, start=0):
+        self.value = start
+
+    def add(self, n=1):
+        self.value += n
+        return self.value
+
+    def get(self):
+        return self.value

@@ -1,0 +1,118 @@
+"""End-to-end tests through the full stack on the local driver:
+client API -> controller -> subprocess "pods" (http_server) -> supervisor ->
+worker processes. No Kubernetes required (BASELINE config 1 analog:
+hello_world via kt.fn().to(kt.Compute(cpus=1))), incl. warm-call RTT and the
+hot-reload loop."""
+import os
+import sys
+import time
+
+import pytest
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
+os.environ["KT_LOCAL_MODE"] = "true"
+os.environ["KT_USERNAME"] = "citest"
+
+import kubetorch_amd as kt  # noqa: E402
+from tests.assets.summer import summer as summer_mod  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def remote_fn():
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    yield f
+    f.teardown()
+
+
+def test_hello_world_and_warm_rtt(remote_fn):
+    assert remote_fn(1, 2) == 3
+    # warm remote-call RTT (BASELINE metric 2): well under 1 s
+    t0 = time.perf_counter()
+    n = 10
+    for i in range(n):
+        assert remote_fn(i, i) == 2 * i
+    rtt = (time.perf_counter() - t0) / n
+    print(f"warm RTT: {rtt*1000:.1f} ms")
+    assert rtt < 1.0, f"warm RTT too slow: {rtt:.3f}s"
+
+
+def test_remote_exception(remote_fn):
+    boom = kt.fn(summer_mod.boom).to(kt.Compute(cpus=1))
+    try:
+        with pytest.raises(ValueError, match="intentional failure") as ei:
+            boom()
+        assert hasattr(ei.value, "remote_traceback")
+        assert "boom" in ei.value.remote_traceback
+    finally:
+        boom.teardown()
+
+
+def test_log_streaming(remote_fn):
+    remote_fn(40, 2)
+    deadline = time.time() + 5
+    while time.time() < deadline:
+        entries = remote_fn.logs()
+        if any("summing 40+2" in e["line"] for e in entries):
+            return
+        time.sleep(0.2)
+    raise AssertionError(f"log line not found in {entries}")
+
+
+def test_hot_reload_loop(remote_fn):
+    """Re-.to() with warm pods must NOT respawn pods and must stay fast
+    (the reference's 1-3 s hot loop; local driver should be much faster)."""
+    first_hosts = list(remote_fn.service_hosts)
+    t0 = time.perf_counter()
+    remote_fn.to()
+    elapsed = time.perf_counter() - t0
+    assert remote_fn.service_hosts == first_hosts, "pods were respawned"
+    assert remote_fn(2, 3) == 5
+    print(f"hot reload: {elapsed:.2f}s")
+    assert elapsed < 3.0, f"hot loop too slow: {elapsed:.2f}s"
+
+
+def test_remote_cls():
+    c = kt.cls(summer_mod.Counter, init_args=((), {"start": 10})).to(
+        kt.Compute(cpus=1))
+    try:
+        assert c.add(5) == 15
+        assert c.add(1) == 16
+        assert c.get() == 16  # state persists in the worker process
+    finally:
+        c.teardown()
+
+
+def test_workload_registry(remote_fn):
+    w = remote_fn.workload()
+    assert w is not None
+    assert w["metadata"]["callable_name"] == "summer"
+
+
+class TestDistributed:
+    @pytest.fixture(scope="class")
+    def dist_fn(self):
+        f = kt.fn(summer_mod.rank_env).to(
+            kt.Compute(cpus=1).distribute("pytorch", workers=2, num_proc=2))
+        yield f
+        f.teardown()
+
+    def test_all_ranks_env(self, dist_fn):
+        results = dist_fn()
+        assert isinstance(results, list) and len(results) == 4
+        ranks = sorted(r["rank"] for r in results)
+        assert ranks == [0, 1, 2, 3]
+        assert all(r["world_size"] == 4 for r in results)
+        assert all(r["master_addr"] == "127.0.0.1" for r in results)
+        assert sorted({r["node_rank"] for r in results}) == [0, 1]
+
+    def test_gloo_allreduce(self, dist_fn):
+        """User code runs torch.distributed through the env contract."""
+        ddp = kt.fn(summer_mod.gloo_allreduce).to(
+            kt.Compute(cpus=1).distribute("pytorch", workers=2, num_proc=1))
+        try:
+            results = ddp(3, kt_restart_procs=True, kt_timeout=180)
+            assert len(results) == 2
+            assert all(r["sum"] == 6.0 for r in results)
+            assert sorted(r["rank"] for r in results) == [0, 1]
+        finally:
+            ddp.teardown()
