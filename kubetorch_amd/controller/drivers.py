@@ -87,6 +87,7 @@ class LocalDriver:
                 env[C.ENV_LAUNCH_ID] = str(launch_id)
             if self.controller_url:
                 env["KT_CONTROLLER_URL"] = self.controller_url
+                env["KT_PEERS_URL"] = self.controller_url
             for k, v in (md.get("env") or {}).items():
                 env[k] = str(v)
             # module metadata -> env contract (same as controller push)

@@ -1,0 +1,174 @@
+"""GPU tensor store facade: kt.put/get for CUDA tensors and state dicts.
+
+put(key, tensor|state_dict): register the live tensors with the node's
+pod-data-server (hipIpc, zero-copy) and publish {key -> source host} to the
+metadata service. get(key, dest): same-node -> IPC device copy inside the
+daemon; cross-node -> per-transfer RCCL broadcast between daemons. Packed
+mode (BroadcastWindow.pack) moves a whole same-dtype state dict as one flat
+buffer = one collective (xGMI-friendly). Reference parity:
+data_store/gpu_transfer.py + pod_data_server client."""
+import json
+import os
+
+import httpx
+
+from kubetorch_amd import constants as C
+from kubetorch_amd.data_store.pod_data_server import PodDataClient, _my_ip
+
+_client = None
+
+
+def _pd_client():
+    global _client
+    if _client is None:
+        _client = PodDataClient()
+    return _client
+
+
+def _my_host(cli):
+    port = cli.ping().get("tcp_port", C.GPU_DATA_SERVER_TCP_PORT)
+    return f"{_my_ip()}:{port}"
+
+
+# -- metadata ------------------------------------------------------------------
+def _meta_root():
+    root = os.path.join(
+        os.environ.get("KT_STORE_ROOT", os.path.expanduser("~/.ktamd/store")),
+        ".meta")
+    os.makedirs(root, exist_ok=True)
+    return root
+
+
+def _meta_path(key):
+    return os.path.join(_meta_root(), key.strip("/").replace("/", "_") + ".json")
+
+
+def publish_meta(key, meta):
+    url = os.environ.get("KT_STORE_URL")
+    if url:
+        httpx.post(f"{url}/meta/{key}", json=meta, timeout=10).raise_for_status()
+    else:
+        with open(_meta_path(key), "w") as f:
+            json.dump(meta, f)
+
+
+def get_meta(key):
+    url = os.environ.get("KT_STORE_URL")
+    if url:
+        r = httpx.get(f"{url}/meta/{key}", timeout=10)
+        if r.status_code == 404:
+            raise KeyError(key)
+        r.raise_for_status()
+        return r.json()
+    p = _meta_path(key)
+    if not os.path.exists(p):
+        raise KeyError(key)
+    with open(p) as f:
+        return json.load(f)
+
+
+def _flatten(sd):
+    return {k: sd[k] for k in sorted(sd)}
+
+
+# -- public api ----------------------------------------------------------------
+def put(key, src, window=None):
+    """Publish a tensor or state dict. Tensors stay in the owner's memory
+    (zero-copy hipIpc registration); only metadata goes to the store."""
+    import torch
+
+    cli = _pd_client()
+    if isinstance(src, torch.Tensor):
+        cli.register(key, src)
+        publish_meta(key, {"kind": "tensor",
+                           "host": _my_host(cli),
+                           "meta": {"shape": tuple(src.shape),
+                                    "dtype": str(src.dtype).split(".")[-1]}})
+        return {"key": key, "locale": "local"}
+    sd = _flatten(src)
+    pack = bool(window and getattr(window, "pack", False))
+    if pack:
+        dtypes = {t.dtype for t in sd.values()}
+        if len(dtypes) != 1:
+            raise ValueError("packed mode requires a single dtype")
+        flat = torch.cat([t.reshape(-1) for t in sd.values()])
+        cli.register(f"{key}/__packed__", flat)
+    else:
+        for sub, t in sd.items():
+            cli.register(f"{key}/{sub}", t)
+    publish_meta(key, {
+        "kind": "state_dict",
+        "host": _my_host(cli),
+        "packed": pack,
+        "entries": {sub: {"shape": tuple(t.shape),
+                          "dtype": str(t.dtype).split(".")[-1],
+                          "numel": t.numel()}
+                    for sub, t in sd.items()},
+    })
+    return {"key": key, "locale": "local", "packed": pack}
+
+
+def get(key, dest, window=None):
+    """Receive into pre-allocated dest tensor / state dict (shapes must
+    match the published metadata)."""
+    import torch
+
+    cli = _pd_client()
+    meta = get_meta(key)
+    source = meta["host"]
+    local = source.split(":")[0] in (_my_ip(), "127.0.0.1")
+
+    def fetch(k, d):
+        if local:
+            cli.get_local(k, d)
+        else:
+            cli.fetch_remote(k, d, source)
+
+    if isinstance(dest, torch.Tensor):
+        if meta["kind"] != "tensor":
+            raise TypeError(f"{key} is a {meta['kind']}, dest is a tensor")
+        fetch(key, dest)
+        return dest
+    sd = _flatten(dest)
+    entries = meta["entries"]
+    missing = set(sd) - set(entries)
+    if missing:
+        raise KeyError(f"dest keys not in stored state dict: {sorted(missing)}")
+    if meta.get("packed"):
+        total = sum(e["numel"] for e in entries.values())
+        dt = getattr(torch, next(iter(entries.values()))["dtype"])
+        first = next(iter(sd.values()))
+        flat = torch.empty(total, dtype=dt, device=first.device)
+        fetch(f"{key}/__packed__", flat)
+        off = 0
+        for sub, e in entries.items():  # sorted publish order
+            n = e["numel"]
+            if sub in sd:
+                sd[sub].copy_(flat[off:off + n].view(e["shape"]))
+            off += n
+        return dest
+    for sub in sd:
+        fetch(f"{key}/{sub}", sd[sub])
+    return dest
+
+
+def rm(key):
+    cli = _pd_client()
+    try:
+        meta = get_meta(key)
+    except KeyError:
+        return
+    if meta["kind"] == "tensor":
+        cli.unregister(key)
+    else:
+        if meta.get("packed"):
+            cli.unregister(f"{key}/__packed__")
+        for sub in meta.get("entries", {}):
+            cli.unregister(f"{key}/{sub}")
+    url = os.environ.get("KT_STORE_URL")
+    if url:
+        httpx.delete(f"{url}/meta/{key}", timeout=10)
+    else:
+        p = _meta_path(key)
+        if os.path.exists(p):
+            os.remove(p)

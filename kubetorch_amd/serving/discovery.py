@@ -24,6 +24,21 @@ def _resolve_dns(hostname, port):
 
 
 def current_peers(service_name=None, namespace=None, port=C.SERVER_PORT):
+    peers_url = os.environ.get("KT_PEERS_URL")
+    if peers_url:
+        # live peer list from the controller (local driver / BYO): reflects
+        # pod death, unlike the static KT_LOCAL_IPS env
+        try:
+            import httpx
+
+            svc = service_name or os.environ.get(C.ENV_SERVICE_NAME)
+            ns = namespace or os.environ.get("POD_NAMESPACE", "default")
+            r = httpx.get(f"{peers_url}/controller/workload/{ns}/{svc}",
+                          timeout=5)
+            if r.status_code == 200:
+                return sorted(r.json().get("pods", []))
+        except Exception:
+            pass
     local = os.environ.get(C.ENV_LOCAL_IPS)
     if local:
         return sorted(p.strip() for p in local.split(",") if p.strip())

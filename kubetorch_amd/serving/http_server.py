@@ -115,6 +115,10 @@ def do_reload(md: dict, launch_id=None):
         ds.sync_workdir_from_store()
     except Exception:
         pass
+    if md and md.get("image_setup"):
+        from kubetorch_amd.serving import image_setup
+
+        image_setup.cached_image_setup(md["image_setup"], app_state=STATE)
     from kubetorch_amd.serving import loading
 
     loading.clear_cache()

@@ -236,6 +236,19 @@ class SPMDSupervisor(DistributedSupervisor):
             return self._run_local_ranks(body, method, hosts, node_rank, timeout)
 
         hosts = self._ordered_hosts()
+        if workers == "any":
+            hosts = hosts[:1]
+        elif workers == "ready":
+            from kubetorch_amd.serving.remote_pool import get_client
+
+            def _up(h):
+                try:
+                    return get_client().get(f"http://{h}/health",
+                                            timeout=3).status_code == 200
+                except Exception:
+                    return False
+
+            hosts = [hosts[0]] + [h for h in hosts[1:] if _up(h)]
         if isinstance(workers, (list, tuple)):
             sel = []
             for w in workers:
@@ -279,6 +292,14 @@ class SPMDSupervisor(DistributedSupervisor):
             for f in done:
                 exc = f.exception()
                 if exc is not None:
+                    # a failed peer connection usually means a dead pod:
+                    # force a membership check before reporting (reference:
+                    # spmd_supervisor.py:413-429)
+                    changed = self.monitor.check_now()
+                    if aborted:
+                        raise aborted[0]
+                    if changed is not None:
+                        raise changed
                     raise exc
             results = []
             local_results = local_fut.result()

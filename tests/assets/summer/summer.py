@@ -46,3 +46,10 @@ class Counter:
 
     def get(self):
         return self.value
+
+
+def slow_echo(x, delay=10):
+    import time
+
+    time.sleep(delay)
+    return x

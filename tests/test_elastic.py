@@ -1,0 +1,56 @@
+"""Elastic fault handling (BASELINE config 4 analog, CPU/local driver):
+kill 1 of 2 worker pods mid-call -> WorkerMembershipChanged aborts the
+in-flight distributed call; re-deploy re-provisions the pod and the next
+call re-forms the process group (per-call rendezvous = elastic re-join)."""
+import os
+import sys
+import threading
+import time
+
+import pytest
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
+os.environ["KT_LOCAL_MODE"] = "true"
+os.environ["KT_USERNAME"] = "faulttest"
+
+import kubetorch_amd as kt  # noqa: E402
+from kubetorch_amd.exceptions import WorkerMembershipChanged  # noqa: E402
+from tests.assets.summer import summer as summer_mod  # noqa: E402
+
+
+@pytest.mark.timeout(300)
+def test_kill_pod_mid_call_then_rejoin():
+    f = kt.fn(summer_mod.slow_echo).to(
+        kt.Compute(cpus=1).distribute("pytorch", workers=2, num_proc=1,
+                                      quorum_timeout=30))
+    try:
+        # healthy distributed call
+        assert f(1, delay=0) == [1, 1]
+
+        # start a long call, then kill the non-coordinator pod
+        result = {}
+
+        def call():
+            try:
+                result["value"] = f(2, delay=30, kt_timeout=120)
+            except BaseException as e:  # noqa: BLE001
+                result["error"] = e
+
+        t = threading.Thread(target=call)
+        t.start()
+        time.sleep(2.0)
+        from kubetorch_amd.controller.app import HUB
+
+        pods = HUB.driver.services[("default", f.name)]
+        pods[1].kill()
+        t.join(90)
+        assert not t.is_alive(), "call did not abort after pod death"
+        assert "error" in result, f"expected abort, got {result}"
+        err = result["error"]
+        assert isinstance(err, WorkerMembershipChanged), repr(err)
+
+        # re-deploy: controller re-provisions (respawn), next call succeeds
+        f.to()
+        assert f(3, delay=0, kt_restart_procs=True) == [3, 3]
+    finally:
+        f.teardown()
