@@ -17,17 +17,17 @@ _LAZY = {
     "fn": ("kubetorch_amd.client.fn", "fn"),
     "cls": ("kubetorch_amd.client.cls", "cls"),
     "app": ("kubetorch_amd.client.app", "app"),
-    "Compute": ("kubetorch_amd.compute.compute", "Compute"),
-    "Image": ("kubetorch_amd.compute.image", "Image"),
-    "Volume": ("kubetorch_amd.compute.volume", "Volume"),
-    "Secret": ("kubetorch_amd.compute.secret", "Secret"),
-    "Endpoint": ("kubetorch_amd.compute.endpoint", "Endpoint"),
-    "AutoscalingConfig": ("kubetorch_amd.compute.autoscaling", "AutoscalingConfig"),
-    "images": ("kubetorch_amd.compute.images", None),
-    "compute": ("kubetorch_amd.compute.decorators", "compute"),
-    "distribute": ("kubetorch_amd.compute.decorators", "distribute"),
-    "autoscale": ("kubetorch_amd.compute.decorators", "autoscale"),
-    "async_": ("kubetorch_amd.compute.decorators", "async_"),
+    "Compute": ("kubetorch_amd.resources.compute", "Compute"),
+    "Image": ("kubetorch_amd.resources.image", "Image"),
+    "Volume": ("kubetorch_amd.resources.volume", "Volume"),
+    "Secret": ("kubetorch_amd.resources.secret", "Secret"),
+    "Endpoint": ("kubetorch_amd.resources.endpoint", "Endpoint"),
+    "AutoscalingConfig": ("kubetorch_amd.resources.autoscaling", "AutoscalingConfig"),
+    "images": ("kubetorch_amd.resources.images", None),
+    "compute": ("kubetorch_amd.resources.decorators", "compute"),
+    "distribute": ("kubetorch_amd.resources.decorators", "distribute"),
+    "autoscale": ("kubetorch_amd.resources.decorators", "autoscale"),
+    "async_": ("kubetorch_amd.resources.decorators", "async_"),
     "put": ("kubetorch_amd.data_store.commands", "put"),
     "get": ("kubetorch_amd.data_store.commands", "get"),
     "ls": ("kubetorch_amd.data_store.commands", "ls"),

@@ -1,0 +1,323 @@
+"""`kt` CLI (reference parity: python_client/kubetorch/cli.py — the same
+command surface, MI355X control plane underneath)."""
+import importlib
+import importlib.util
+import json
+import os
+import sys
+
+import typer
+from rich.console import Console
+from rich.table import Table
+
+app = typer.Typer(help="kubetorch_amd: MI355X-native serverless ML dispatch")
+console = Console()
+
+secrets_app = typer.Typer(help="manage secrets")
+volumes_app = typer.Typer(help="manage volumes")
+server_app = typer.Typer(help="run framework services")
+app.add_typer(secrets_app, name="secrets")
+app.add_typer(volumes_app, name="volumes")
+app.add_typer(server_app, name="server")
+
+
+def _load_module_file(path):
+    path = os.path.abspath(path)
+    sys.path.insert(0, os.path.dirname(path))
+    spec = importlib.util.spec_from_file_location(
+        os.path.splitext(os.path.basename(path))[0], path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    return mod
+
+
+@app.command()
+def check():
+    """Cluster/controller doctor: connectivity, driver, GPU visibility."""
+    from kubetorch_amd.globals import controller_client
+
+    try:
+        cc = controller_client()
+        import httpx
+
+        h = httpx.get(cc.base_url + "/health", timeout=10).json()
+        console.print(f"[green]controller ok[/green] at {cc.base_url} "
+                      f"(driver: {h.get('driver')})")
+    except Exception as e:
+        console.print(f"[red]controller unreachable:[/red] {e}")
+        raise typer.Exit(1)
+    try:
+        import torch
+
+        n = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        console.print(f"local GPUs visible: {n}")
+    except Exception:
+        console.print("torch not importable locally")
+    from kubetorch_amd.ops import hip_available
+
+    console.print(f"gfx950 HIP extension built: {hip_available()}")
+
+
+@app.command()
+def config(key: str = typer.Argument(None), value: str = typer.Argument(None),
+           persist: bool = typer.Option(False, "--persist")):
+    """Show or set client config."""
+    from kubetorch_amd.config import config as cfg
+
+    if key is None:
+        console.print_json(json.dumps(cfg.as_dict()))
+    elif value is None:
+        console.print(str(cfg.get(key)))
+    else:
+        cfg.set(key, value, persist=persist)
+        console.print(f"set {key}={value}")
+
+
+@app.command()
+def deploy(target: str,
+           workers: int = typer.Option(None, help="override workers")):
+    """Deploy decorated modules from a file: `kt deploy train.py[:fn]`."""
+    path, _, symbol = target.partition(":")
+    mod = _load_module_file(path)
+    from kubetorch_amd.resources.decorators import PartialModule
+
+    partials = (
+        {symbol: getattr(mod, symbol)} if symbol
+        else {k: v for k, v in vars(mod).items()
+              if isinstance(v, PartialModule)}
+    )
+    if not partials:
+        console.print("[red]no @kt.compute-decorated callables found[/red]")
+        raise typer.Exit(1)
+    for name, pm in partials.items():
+        if not isinstance(pm, PartialModule):
+            console.print(f"[red]{name} is not decorated[/red]")
+            raise typer.Exit(1)
+        m = pm.build_module()
+        if workers and m.compute.distributed_config:
+            m.compute.distributed_config["workers"] = workers
+            m.compute.replicas = workers
+        m.to(m.compute)
+        console.print(f"[green]deployed[/green] {m.name}")
+
+
+@app.command()
+def call(service: str, method: str = typer.Argument(None),
+         args_json: str = typer.Option("[]", "--args"),
+         kwargs_json: str = typer.Option("{}", "--kwargs")):
+    """Call a deployed service: `kt call my-fn --args '[1,2]'`."""
+    from kubetorch_amd.client.http_client import HTTPClient
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client, service_url
+
+    w = controller_client().get_workload(service, cfg.namespace)
+    if not w:
+        console.print(f"[red]no workload {service!r}[/red]")
+        raise typer.Exit(1)
+    pods = w.get("pods") or []
+    name = w["metadata"].get("callable_name", service)
+    client = HTTPClient(service_url(service, cfg.namespace, pods), name)
+    result = client.call(args=json.loads(args_json),
+                         kwargs=json.loads(kwargs_json), method=method,
+                         serialization="json")
+    console.print_json(json.dumps({"result": result}))
+
+
+@app.command("list")
+def list_cmd(namespace: str = typer.Option(None, "-n")):
+    """List workloads."""
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+
+    ns = namespace or cfg.namespace
+    ws = controller_client().list_workloads(ns).get("workloads", [])
+    t = Table("name", "namespace", "kind", "launch_id", "module")
+    for w in ws:
+        t.add_row(w["name"], w["namespace"],
+                  w.get("service_config", {}).get("kind", "?"),
+                  str(w.get("launch_id")),
+                  w.get("metadata", {}).get("callable_name", ""))
+    console.print(t)
+
+
+@app.command()
+def describe(service: str, namespace: str = typer.Option(None, "-n")):
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+
+    w = controller_client().get_workload(service, namespace or cfg.namespace)
+    if not w:
+        console.print("[red]not found[/red]")
+        raise typer.Exit(1)
+    console.print_json(json.dumps(w, default=str))
+
+
+@app.command()
+def teardown(service: str = typer.Argument(None),
+             prefix: str = typer.Option(None, "-p"),
+             namespace: str = typer.Option(None, "-n")):
+    """Tear down a workload (or all with a name prefix)."""
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+
+    ns = namespace or cfg.namespace
+    cc = controller_client()
+    if prefix:
+        names = [w["name"] for w in cc.list_workloads(ns).get("workloads", [])
+                 if w["name"].startswith(prefix)]
+    elif service:
+        names = [service]
+    else:
+        console.print("[red]pass a service or -p prefix[/red]")
+        raise typer.Exit(1)
+    for n in names:
+        cc.delete_workload(n, ns)
+        console.print(f"deleted {n}")
+
+
+@app.command()
+def logs(service: str, namespace: str = typer.Option(None, "-n"),
+         limit: int = typer.Option(100)):
+    from kubetorch_amd.client.http_client import HTTPClient
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client, service_url
+
+    ns = namespace or cfg.namespace
+    w = controller_client().get_workload(service, ns)
+    if not w:
+        raise typer.Exit(1)
+    client = HTTPClient(service_url(service, ns, w.get("pods")), service)
+    for e in client.logs(limit=limit):
+        console.print(f"[dim]{e['source']}[/dim] {e['line']}")
+
+
+@app.command()
+def run(command: str, name: str = typer.Option("app"),
+        cpus: str = typer.Option(None), gpus: int = typer.Option(0)):
+    """Deploy an arbitrary command as an app: `kt run 'python serve.py'`."""
+    import kubetorch_amd as kt
+
+    a = kt.app(command, name=name)
+    a.to(kt.Compute(cpus=cpus, gpus=gpus))
+    console.print(f"[green]running[/green] {a.name}")
+
+
+@app.command()
+def apply(manifest_path: str, namespace: str = typer.Option(None, "-n")):
+    """Apply a raw manifest through the controller."""
+    import yaml
+
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+
+    with open(manifest_path) as f:
+        manifest = yaml.safe_load(f)
+    name = manifest["metadata"]["name"]
+    resp = controller_client().deploy(
+        name=name, namespace=namespace or cfg.namespace, manifest=manifest)
+    console.print_json(json.dumps(resp))
+
+
+@app.command()
+def put(key: str, src: str):
+    from kubetorch_amd.data_store import commands as ds
+
+    console.print_json(json.dumps(ds.put(key, src=src)))
+
+
+@app.command()
+def get(key: str, dest: str = typer.Argument(None)):
+    from kubetorch_amd.data_store import commands as ds
+
+    console.print(str(ds.get(key, dest=dest)))
+
+
+@app.command()
+def ls(prefix: str = typer.Argument("")):
+    from kubetorch_amd.data_store import commands as ds
+
+    t = Table("key", "size")
+    for e in ds.ls(prefix):
+        t.add_row(e["key"], str(e["size"]))
+    console.print(t)
+
+
+@app.command()
+def rm(key: str):
+    from kubetorch_amd.data_store import commands as ds
+
+    ds.rm(key)
+    console.print(f"removed {key}")
+
+
+@app.command()
+def workload(service: str, namespace: str = typer.Option(None, "-n")):
+    describe(service, namespace)
+
+
+@app.command()
+def debug(service: str, port: int = typer.Option(None),
+          namespace: str = typer.Option(None, "-n")):
+    """Attach to a remote breakpoint() (pdb over WebSocket)."""
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+    from kubetorch_amd.serving.pdb_ws import attach
+
+    ns = namespace or cfg.namespace
+    w = controller_client().get_workload(service, ns)
+    if not w:
+        raise typer.Exit(1)
+    pods = w.get("pods") or []
+    host = pods[0].split(":")[0] if pods else "127.0.0.1"
+    attach(host, port or 4444)
+
+
+@secrets_app.command("create")
+def secrets_create(name: str, provider: str = typer.Option(None),
+                   values_json: str = typer.Option("{}", "--values")):
+    from kubetorch_amd.resources.secret import Secret
+
+    s = Secret(name, values=json.loads(values_json), provider=provider)
+    console.print_json(json.dumps(s.to_manifest("default"), default=str))
+
+
+@volumes_app.command("create")
+def volumes_create(name: str, size: str = typer.Option("10Gi")):
+    from kubetorch_amd.resources.volume import Volume
+
+    console.print_json(json.dumps(Volume(name, size=size).to_pvc_manifest("default")))
+
+
+@server_app.command("start")
+def server_start(port: int = typer.Option(None),
+                 kind: str = typer.Option("worker",
+                                          help="worker|controller|store|gpu-data")):
+    """Run a framework service in the foreground (pod/BYO entrypoint)."""
+    if kind == "worker":
+        from kubetorch_amd.serving import http_server
+
+        sys.argv = ["http_server"] + (["--port", str(port)] if port else [])
+        http_server.main()
+    elif kind == "controller":
+        from kubetorch_amd.controller import app as controller
+
+        sys.argv = ["controller"] + (["--port", str(port)] if port else [])
+        controller.main()
+    elif kind == "store":
+        from kubetorch_amd.data_store import server as store
+
+        sys.argv = ["store"] + (["--port", str(port)] if port else [])
+        store.main()
+    elif kind == "gpu-data":
+        from kubetorch_amd.data_store import pod_data_server
+
+        sys.argv = ["pod_data_server"]
+        pod_data_server.main()
+
+
+def main():
+    app()
+
+
+if __name__ == "__main__":
+    main()

@@ -4,7 +4,7 @@ optional health path gates readiness."""
 import os
 
 from kubetorch_amd.client.module import Module, sanitize_name
-from kubetorch_amd.compute.compute import Compute
+from kubetorch_amd.resources.compute import Compute
 from kubetorch_amd.config import config
 
 

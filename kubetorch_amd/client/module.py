@@ -16,7 +16,7 @@ import uuid
 
 from kubetorch_amd import constants as C
 from kubetorch_amd.client.http_client import HTTPClient, shared_client
-from kubetorch_amd.compute.compute import Compute
+from kubetorch_amd.resources.compute import Compute
 from kubetorch_amd.config import config
 from kubetorch_amd.exceptions import LaunchError
 from kubetorch_amd.globals import controller_client

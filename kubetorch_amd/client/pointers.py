@@ -29,7 +29,7 @@ def locate_working_dir(start_path):
 def extract_pointers(obj):
     """-> dict(project_root, file_path, rel_path, name). Works for functions
     and classes defined in real files (notebook cells are dumped first)."""
-    name = obj.__qualname__.split(".")[0]
+    name = obj.__name__
     try:
         file_path = os.path.abspath(inspect.getfile(obj))
     except TypeError as e:

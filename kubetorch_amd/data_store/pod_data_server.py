@@ -356,7 +356,7 @@ class PodDataClient:
         return self.request({"cmd": "ping"})
 
 
-def ensure_server(sock_path=SOCK_PATH, tcp_port=None, timeout=20.0):
+def ensure_server(sock_path=SOCK_PATH, tcp_port=None, timeout=120.0):
     """Start the per-node daemon if not running (file-lock singleton)."""
     import subprocess
     import sys

@@ -7,8 +7,8 @@ mechanics live in provisioning/manifests.py here.)"""
 import copy
 
 from kubetorch_amd import constants as C
-from kubetorch_amd.compute.autoscaling import AutoscalingConfig
-from kubetorch_amd.compute.images import DEFAULT_PYTORCH_ROCM
+from kubetorch_amd.resources.autoscaling import AutoscalingConfig
+from kubetorch_amd.resources.images import DEFAULT_PYTORCH_ROCM
 from kubetorch_amd.config import config
 from kubetorch_amd.provisioning import manifests as M
 

@@ -25,7 +25,7 @@ class PartialModule:
 
         from kubetorch_amd.client.cls import cls as cls_factory
         from kubetorch_amd.client.fn import fn as fn_factory
-        from kubetorch_amd.compute.compute import Compute
+        from kubetorch_amd.resources.compute import Compute
 
         compute = Compute(**self.compute_kwargs)
         if self.distribute_args:

@@ -1,7 +1,7 @@
 """Stock images — ROCm-first (reference ships nvcr.io pytorch images;
 here the default worker is PyTorch-ROCm gfx950). Reference parity:
 resources/images/images.py."""
-from kubetorch_amd.compute.image import Image
+from kubetorch_amd.resources.image import Image
 
 DEFAULT_PYTORCH_ROCM = "rocm/pytorch:latest"
 

@@ -25,6 +25,8 @@ def _worker_main(idx, req_q, resp_q, base_env, eager_load, log_q=None):
     log queue, optionally eager-load the callable, then serve requests on a
     thread pool."""
     os.environ.update({k: str(v) for k, v in base_env.items()})
+    os.environ.setdefault(
+        "PYTHONBREAKPOINT", "kubetorch_amd.serving.pdb_ws.deep_breakpoint")
     import contextvars
     import io
     import sys

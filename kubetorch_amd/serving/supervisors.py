@@ -327,13 +327,56 @@ class SPMDSupervisor(DistributedSupervisor):
         return out
 
 
+class RaySupervisor(ExecutionSupervisor):
+    """Head-only Ray supervisor: starts `ray start --head` (or the command
+    in KUBERAY_GEN_RAY_START_CMD) as a side process; calls run in a single
+    worker that connects to the local GCS. Membership is Ray's business.
+    (Reference parity: serving/ray_supervisor.py.)"""
+
+    distributed = True
+
+    def __init__(self, num_proc=1, **kw):
+        import shutil
+        import subprocess
+
+        self._ray_proc = None
+        cmd = os.environ.get("KUBERAY_GEN_RAY_START_CMD")
+        if cmd is None and shutil.which("ray"):
+            cmd = "ray start --head --block --dashboard-host=0.0.0.0"
+        if cmd:
+            self._ray_proc = subprocess.Popen(["bash", "-lc", cmd])
+        super().__init__(num_proc=1)
+
+    def cleanup(self):
+        if self._ray_proc is not None and self._ray_proc.poll() is None:
+            self._ray_proc.terminate()
+        super().cleanup()
+
+
+class MonarchSupervisor(ExecutionSupervisor):
+    """Monarch actor meshes are not available in the MI355X image (no
+    monarch wheel for ROCm yet); fail loudly with guidance rather than
+    silently degrading. (Reference: serving/monarch_supervisor.py.)"""
+
+    def __init__(self, **kw):
+        raise NotImplementedError(
+            "Monarch is not available on this image; use "
+            "distribute('pytorch'|'spmd'|'ray') instead."
+        )
+
+
 def supervisor_factory(distribution_type=None, **kw):
     """"local"/None -> ExecutionSupervisor; "pytorch"/"jax"/"tensorflow"/
     "spmd" -> SPMDSupervisor. (Reference: serving/supervisor_factory.py.)"""
     if distribution_type in (None, "", "local"):
         kw.pop("num_workers", None)
         kw.pop("framework", None)
+        kw.pop("quorum_timeout", None)
         return ExecutionSupervisor(num_proc=kw.pop("num_proc", 1) or 1,
                                    **{k: v for k, v in kw.items()
                                       if k in ("eager_load",)})
+    if distribution_type == "ray":
+        return RaySupervisor()
+    if distribution_type == "monarch":
+        return MonarchSupervisor()
     return SPMDSupervisor(framework=distribution_type, **kw)

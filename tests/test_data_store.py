@@ -200,7 +200,7 @@ def _gpu_putter(sock, q):
     t = torch.arange(1024, dtype=torch.bfloat16, device="cuda") * 0.5
     gpu_store.put("gpu_w", t)
     q.put("published")
-    time.sleep(30)  # keep the owning process (and its HBM) alive
+    time.sleep(600)  # keep the owning process (and its HBM) alive
 
 
 @pytest.mark.gpu
@@ -222,7 +222,7 @@ def test_gpu_ipc_put_get(tmp_path):
     p = ctx.Process(target=_gpu_putter, args=(sock, q), daemon=True)
     p.start()
     try:
-        assert q.get(timeout=120) == "published"
+        assert q.get(timeout=420) == "published"
         from kubetorch_amd.data_store import gpu_store
 
         dest = torch.zeros(1024, dtype=torch.bfloat16, device="cuda")
