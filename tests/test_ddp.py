@@ -14,7 +14,14 @@ import torch.multiprocessing as mp
 from kubetorch_amd.models import Llama, llama_tiny
 from kubetorch_amd.parallel import FlatDDP
 
-PORT = 29511
+def _free_port():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
 
 
 def _make(seed=0):
@@ -37,9 +44,9 @@ def _data(cfg):
     return x, y
 
 
-def _worker(rank, world, q):
+def _worker(rank, world, q, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(PORT)
+    os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
         model, cfg = _make()
@@ -62,7 +69,8 @@ def _worker(rank, world, q):
 def test_flatddp_two_ranks_match_single_process():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_worker, args=(r, 2, q, port)) for r in range(2)]
     for p in procs:
         p.start()
     results = {}
