@@ -58,7 +58,63 @@ class Hub:
 
 
 HUB = Hub()
-app = FastAPI()
+
+
+def _parse_ttl(raw):
+    if not raw:
+        return None
+    raw = str(raw).strip().lower()
+    mult = {"s": 1, "m": 60, "h": 3600, "d": 86400}.get(raw[-1])
+    try:
+        return int(float(raw[:-1]) * mult) if mult else int(float(raw))
+    except ValueError:
+        return None
+
+
+async def _ttl_reaper(interval=30.0):
+    """Tear down workloads whose pods report no activity for longer than
+    their inactivity TTL (reference: controller TTL reaper fed by the
+    kt_last_activity_timestamp metric)."""
+    import httpx
+
+    while True:
+        await asyncio.sleep(interval)
+        for (ns, name), w in list(HUB.workloads.items()):
+            ttl = _parse_ttl(
+                w.get("manifest", {}).get("metadata", {})
+                .get("annotations", {}).get(C.INACTIVITY_TTL_ANNOTATION))
+            if not ttl:
+                continue
+            last = w.get("updated", 0)
+            try:
+                pods = HUB.driver.pods(name, ns)
+                if pods:
+                    async with httpx.AsyncClient(timeout=5) as client:
+                        r = await client.get(f"http://{pods[0]}/metrics")
+                    for line in r.text.splitlines():
+                        if line.startswith("kt_last_activity_timestamp"):
+                            last = max(last, float(line.split()[-1]))
+            except Exception:
+                continue
+            if time.time() - last > ttl:
+                HUB.workloads.pop((ns, name), None)
+                try:
+                    await asyncio.to_thread(HUB.driver.delete, name, ns)
+                except Exception:
+                    pass
+
+
+from contextlib import asynccontextmanager
+
+
+@asynccontextmanager
+async def _lifespan(app):
+    task = asyncio.create_task(_ttl_reaper())
+    yield
+    task.cancel()
+
+
+app = FastAPI(lifespan=_lifespan)
 
 
 @app.get("/health")

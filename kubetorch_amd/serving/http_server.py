@@ -176,6 +176,9 @@ async def _controller_ws_loop(url):
 async def lifespan(app: FastAPI):
     log_capture.install()
     STATE["started_at"] = time.time()
+    from kubetorch_amd.serving import gpu_metrics
+
+    gpu_metrics.register()
     pusher = metrics.MetricsPusher()
     pusher.start()
     stop_ev = threading.Event()

@@ -1,0 +1,89 @@
+"""Checkpoint spill/restore sized for 288 GB HBM per MI355X.
+
+GPU state dicts spill through pinned host buffers with async copies
+(hipMemcpyAsync under torch's non_blocking copy on a dedicated stream) so
+the D2H transfer runs at full PCIe/host bandwidth and overlaps across
+tensors, then persist to a local path and/or the cluster data store
+(`kt.put` with lifespan="cluster" semantics). Reference substrate:
+SURVEY.md §5 checkpoint/resume — kt.put/get of state dicts."""
+import os
+
+import torch
+
+
+def spill_state_dict(sd, pin=True):
+    """GPU -> CPU state dict via pinned staging + async copies."""
+    out = {}
+    if not torch.cuda.is_available():
+        return {k: (t.detach().cpu() if isinstance(t, torch.Tensor) else t)
+                for k, t in sd.items()}
+    stream = torch.cuda.Stream()
+    stream.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(stream):
+        for k, t in sd.items():
+            if isinstance(t, torch.Tensor) and t.is_cuda:
+                host = torch.empty(t.shape, dtype=t.dtype, device="cpu",
+                                   pin_memory=pin)
+                host.copy_(t.detach(), non_blocking=True)
+                out[k] = host
+            elif isinstance(t, torch.Tensor):
+                out[k] = t.detach().clone()
+            else:
+                out[k] = t
+    stream.synchronize()
+    return out
+
+
+def save_checkpoint(state_dict, path, store_key=None):
+    """Spill + torch.save; optionally also push to the cluster data store."""
+    host_sd = spill_state_dict(state_dict)
+    os.makedirs(os.path.dirname(os.path.abspath(path)) or ".", exist_ok=True)
+    torch.save(host_sd, path)
+    if store_key:
+        from kubetorch_amd.data_store import commands as ds
+
+        ds.put(store_key, src=path)
+    return path
+
+
+def load_checkpoint(path_or_key, map_location="cpu"):
+    """Load from a local path, falling back to the data store key."""
+    if not os.path.exists(path_or_key):
+        from kubetorch_amd.data_store import commands as ds
+
+        local = ds.get(path_or_key, dest=None)
+        if isinstance(local, str) and os.path.exists(local):
+            path_or_key = local
+    return torch.load(path_or_key, map_location=map_location,
+                      weights_only=False)
+
+
+def save_engine_checkpoint(model, engine, path, store_key=None):
+    """Model + FlatDDP optimizer state in one file (rank 0 only helper)."""
+    sd = {"model": spill_state_dict(model.state_dict()),
+          "engine": {
+              "step": engine.step_count,
+              "m": [spill_state_dict({"t": b.m})["t"] for b in engine.buckets],
+              "v": [spill_state_dict({"t": b.v})["t"] for b in engine.buckets],
+              "flat_param": [spill_state_dict({"t": b.flat_param})["t"]
+                             for b in engine.buckets],
+          }}
+    os.makedirs(os.path.dirname(os.path.abspath(path)) or ".", exist_ok=True)
+    torch.save(sd, path)
+    if store_key:
+        from kubetorch_amd.data_store import commands as ds
+
+        ds.put(store_key, src=path)
+    return path
+
+
+def load_engine_checkpoint(model, engine, path_or_key):
+    sd = load_checkpoint(path_or_key)
+    model.load_state_dict(sd["model"])
+    eng = sd["engine"]
+    engine.step_count = eng["step"]
+    for b, m, v, fp in zip(engine.buckets, eng["m"], eng["v"],
+                           eng["flat_param"]):
+        b.m.copy_(m)
+        b.v.copy_(v)
+        b.flat_param.copy_(fp)
