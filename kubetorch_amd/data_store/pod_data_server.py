@@ -62,12 +62,14 @@ def export_tensor(t):
     meta = {"shape": tuple(t.shape), "dtype": str(t.dtype).split(".")[-1],
             "device": "cuda" if t.is_cuda else "cpu"}
     if t.is_cuda:
-        t = t.contiguous()
-        storage = t.untyped_storage()
-        handle = storage._share_cuda_()
-        return {"meta": meta, "ipc": handle,
-                "storage_offset": t.storage_offset(),
-                "device_index": t.device.index}
+        # torch's canonical IPC reduction (hipIpcMemHandle + ref-counter +
+        # event under the hood); raw _new_shared_cuda without the ref-count
+        # machinery segfaults the consumer on ROCm.
+        from torch.multiprocessing.reductions import reduce_tensor
+
+        func, args = reduce_tensor(t.contiguous().detach())
+        assert func.__name__ == "rebuild_cuda_tensor", func
+        return {"meta": meta, "cuda_reduced": args}
     return {"meta": meta, "bytes": t.contiguous().cpu().numpy().tobytes()}
 
 
@@ -77,12 +79,11 @@ def import_tensor(payload):
 
     meta = payload["meta"]
     dtype = getattr(torch, meta["dtype"])
-    if "ipc" in payload:
-        storage = torch.UntypedStorage._new_shared_cuda(*payload["ipc"])
-        t = torch.tensor([], dtype=dtype,
-                         device=f"cuda:{payload['device_index']}")
-        t.set_(storage, payload.get("storage_offset", 0), meta["shape"])
-        return t
+    if "cuda_reduced" in payload:
+        from torch.multiprocessing.reductions import rebuild_cuda_tensor
+
+        torch.cuda.init()
+        return rebuild_cuda_tensor(*payload["cuda_reduced"])
     t = torch.frombuffer(bytearray(payload["bytes"]), dtype=dtype)
     return t.reshape(meta["shape"]).clone()
 
@@ -187,7 +188,7 @@ class PodDataServer:
             if entry is None:
                 return {"ok": False, "error": f"key {msg['key']!r} not found"}
             src = entry["tensor"]
-            if "dest" in msg and "ipc" in msg["dest"]:
+            if "dest" in msg and "cuda_reduced" in msg["dest"]:
                 # hipIpc-mapped dest: device-to-device copy, caller sees it
                 dest = import_tensor(msg["dest"])
                 dest.copy_(src)
