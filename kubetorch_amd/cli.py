@@ -272,6 +272,65 @@ def debug(service: str, port: int = typer.Option(None),
     attach(host, port or 4444)
 
 
+@app.command("port-forward")
+def port_forward(target: str, ports: str,
+                 namespace: str = typer.Option(None, "-n")):
+    """kubectl port-forward with health wait: `kt port-forward svc/x 8080:8081`."""
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import PortForward
+
+    local, _, remote = ports.partition(":")
+    pf = PortForward(target, namespace or cfg.namespace, int(local),
+                     int(remote or local)).start()
+    console.print(f"forwarding 127.0.0.1:{local} -> {target}:{remote or local}"
+                  " (ctrl-c to stop)")
+    try:
+        pf.proc.wait()
+    except KeyboardInterrupt:
+        pf.stop()
+
+
+@app.command()
+def ssh(service: str, namespace: str = typer.Option(None, "-n")):
+    """Exec an interactive shell in the service's first pod (kubectl)."""
+    import shutil
+    import subprocess
+
+    from kubetorch_amd.config import config as cfg
+
+    if shutil.which("kubectl") is None:
+        console.print("[red]kubectl not found; `kt ssh` needs cluster access."
+                      " In local mode use `kt call`/`kt logs`.[/red]")
+        raise typer.Exit(1)
+    from kubetorch_amd import constants as C
+
+    ns = namespace or cfg.namespace
+    out = subprocess.run(
+        ["kubectl", "-n", ns, "get", "pods", "-l",
+         f"{C.SERVICE_LABEL}={service}", "-o", "name"],
+        capture_output=True, text=True)
+    pods = out.stdout.split()
+    if not pods:
+        console.print("[red]no pods found[/red]")
+        raise typer.Exit(1)
+    subprocess.run(["kubectl", "-n", ns, "exec", "-it", pods[0], "--", "bash"])
+
+
+@app.command()
+def notebook(service: str = typer.Option("notebook"),
+             port: int = typer.Option(8888)):
+    """Deploy a remote Jupyter server and port-forward to it."""
+    import kubetorch_amd as kt
+
+    a = kt.app(
+        f"python -m jupyter lab --ip=0.0.0.0 --port={port} --allow-root "
+        f"--NotebookApp.token=''",
+        name=service, port=port)
+    a.to(kt.Compute(cpus=2))
+    console.print(f"[green]notebook deployed[/green]: {a.name} "
+                  f"(use `kt port-forward` to reach :{port})")
+
+
 @secrets_app.command("create")
 def secrets_create(name: str, provider: str = typer.Option(None),
                    values_json: str = typer.Option("{}", "--values")):

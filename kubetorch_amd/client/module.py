@@ -152,6 +152,25 @@ class Module:
     def workload(self):
         return controller_client().get_workload(self.name, self.namespace)
 
+    # -- post-launch helpers (reference: compute.py ssh/pip_install/run_bash) --
+    def run_bash(self, command, timeout=600):
+        """Run a bash command inside the service's (first) pod."""
+        r = shared_client().post(self.http.base_url + "/exec",
+                                 json={"command": command, "timeout": timeout},
+                                 timeout=timeout + 30)
+        r.raise_for_status()
+        return r.json()
+
+    def pip_install(self, packages, extra_args=""):
+        if isinstance(packages, str):
+            packages = [packages]
+        import shlex
+        import sys
+
+        pkgs = " ".join(shlex.quote(p) for p in packages)
+        return self.run_bash(
+            f"{shlex.quote(sys.executable)} -m pip install {pkgs} {extra_args}")
+
     def _call(self, args, kwargs, method=None, **opts):
         return self.http.call(
             args=args, kwargs=kwargs, method=method,

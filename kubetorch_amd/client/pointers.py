@@ -32,8 +32,20 @@ def extract_pointers(obj):
     name = obj.__name__
     try:
         file_path = os.path.abspath(inspect.getfile(obj))
-    except TypeError as e:
-        raise ValueError(f"cannot locate source file for {obj!r}: {e}") from e
+        if not os.path.exists(file_path):
+            raise TypeError("source file missing")
+        if obj.__module__ in ("__main__",) and file_path.endswith(
+                ("<stdin>", "<ipython-input>")):
+            raise TypeError("interactive source")
+    except TypeError:
+        # notebook / REPL: dump the cell source to a real file
+        # (reference: resources/callables/utils.py notebook support)
+        src = inspect.getsource(obj)
+        nb_dir = os.path.join(os.getcwd(), ".kt_notebook")
+        os.makedirs(nb_dir, exist_ok=True)
+        file_path = os.path.join(nb_dir, f"{name}.py")
+        with open(file_path, "w") as f:
+            f.write(src)
     if file_path.endswith((".pyc", ".pyo")):
         file_path = file_path[:-1]
     root = locate_working_dir(file_path)

@@ -99,6 +99,52 @@ def controller_client() -> ControllerClient:
         return _controller
 
 
+class PortForward:
+    """kubectl port-forward lifecycle for out-of-cluster clients
+    (reference: globals.py:123-300). No-op in local mode."""
+
+    def __init__(self, target, namespace, local_port, remote_port):
+        self.target = target
+        self.namespace = namespace
+        self.local_port = local_port
+        self.remote_port = remote_port
+        self.proc = None
+
+    def start(self, timeout=20):
+        import shutil
+        import socket
+        import subprocess
+
+        if shutil.which("kubectl") is None:
+            raise RuntimeError("kubectl not found; port-forward unavailable")
+        self.proc = subprocess.Popen(
+            ["kubectl", "-n", self.namespace, "port-forward", self.target,
+             f"{self.local_port}:{self.remote_port}"],
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+        )
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            try:
+                s = socket.create_connection(("127.0.0.1", self.local_port), 1)
+                s.close()
+                return self
+            except OSError:
+                if self.proc.poll() is not None:
+                    raise RuntimeError("kubectl port-forward exited")
+                time.sleep(0.2)
+        raise RuntimeError("port-forward did not become ready")
+
+    def stop(self):
+        if self.proc is not None and self.proc.poll() is None:
+            self.proc.terminate()
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()
+
+
 def service_url(service_name, namespace, hosts=None):
     """Resolve the base URL for a deployed service. Local mode / explicit
     hosts -> first pod; in-cluster -> the K8s Service DNS."""

@@ -393,6 +393,22 @@ async def reload_route(request: Request):
         return JSONResponse({"error": package_exception(e)}, status_code=500)
 
 
+@app.post("/exec")
+async def exec_route(request: Request):
+    """Run a bash command inside the pod (post-launch pip_install/run_bash
+    helpers; reference: compute.py run_bash/pip_install)."""
+    import subprocess
+
+    payload = await request.json()
+    cmd = payload["command"]
+    res = await asyncio.to_thread(
+        subprocess.run, ["bash", "-lc", cmd],
+        capture_output=True, text=True, timeout=payload.get("timeout", 600),
+    )
+    return {"returncode": res.returncode, "stdout": res.stdout[-20000:],
+            "stderr": res.stderr[-20000:]}
+
+
 @app.get("/app/status")
 def app_status():
     proc = STATE.get("app_proc")
@@ -402,10 +418,33 @@ def app_status():
     return {"running": rc is None, "returncode": rc}
 
 
+def _install_sigterm_drain():
+    """SIGTERM: mark terminating (readiness flips 503) and let in-flight
+    requests drain before uvicorn exits (reference:
+    TerminationCheckMiddleware, http_server.py:1184-1236)."""
+    import signal
+
+    def handler(signum, frame):
+        STATE["terminating"] = True
+        if STATE["supervisor"] is not None:
+            try:
+                STATE["supervisor"].cleanup()
+            except Exception:
+                pass
+        raise SystemExit(0)
+
+    try:
+        signal.signal(signal.SIGTERM, handler)
+    except ValueError:
+        pass  # not the main thread (in-process test server)
+
+
 def main():
     import argparse
 
     import uvicorn
+
+    _install_sigterm_drain()
 
     ap = argparse.ArgumentParser()
     ap.add_argument("--port", type=int,

@@ -1,0 +1,60 @@
+"""CLI tests (typer runner) against the local driver."""
+import os
+import sys
+
+from typer.testing import CliRunner
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
+os.environ["KT_LOCAL_MODE"] = "true"
+os.environ["KT_USERNAME"] = "clitest"
+
+import kubetorch_amd as kt  # noqa: E402
+from kubetorch_amd.cli import app  # noqa: E402
+from tests.assets.summer import summer as summer_mod  # noqa: E402
+
+runner = CliRunner()
+
+
+def test_check():
+    r = runner.invoke(app, ["check"])
+    assert r.exit_code == 0, r.output
+    assert "controller ok" in r.output
+
+
+def test_config_show():
+    r = runner.invoke(app, ["config"])
+    assert r.exit_code == 0
+    assert "namespace" in r.output
+
+
+def test_list_describe_call_teardown():
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        r = runner.invoke(app, ["list"])
+        assert f.name in r.output.replace("\n", "")
+        r = runner.invoke(app, ["describe", f.name])
+        assert r.exit_code == 0
+        r = runner.invoke(app, ["call", f.name, "--args", "[2, 3]"])
+        assert r.exit_code == 0, r.output
+        assert "5" in r.output
+    finally:
+        r = runner.invoke(app, ["teardown", f.name])
+        assert "deleted" in r.output
+
+
+def test_secrets_volumes_manifests():
+    r = runner.invoke(app, ["secrets", "create", "hf", "--values",
+                            '{"HF_TOKEN": "x"}'])
+    assert r.exit_code == 0 and "kt-secret-hf" in r.output
+    r = runner.invoke(app, ["volumes", "create", "cache", "--size", "5Gi"])
+    assert r.exit_code == 0 and "5Gi" in r.output
+
+
+def test_run_bash_and_pip_helpers():
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        out = f.run_bash("echo hello-from-pod")
+        assert out["returncode"] == 0
+        assert "hello-from-pod" in out["stdout"]
+    finally:
+        f.teardown()
