@@ -150,13 +150,29 @@ class Llama(nn.Module):
                 nn.init.ones_(m.weight)
 
     def forward(self, tokens):
+        """Pre-norm transformer with the residual adds fused into the norm
+        kernels: the residual stream is carried as (res, delta) where delta
+        is the last branch output not yet added; ops.add_rmsnorm performs
+        `res += delta` and the next norm in one pass."""
         S = tokens.shape[1]
-        x = self.embed(tokens)
         cos = self.rope_cos[:S]
         sin = self.rope_sin[:S]
+        res = self.embed(tokens)
+        delta = None
+        eps = self.cfg.norm_eps
         for layer in self.layers:
-            x = layer(x, cos, sin)
-        return self.lm_head(self.norm(x))
+            if delta is None:
+                n1 = ops.rmsnorm(res, layer.attn_norm.weight, eps)
+            else:
+                n1, res = ops.add_rmsnorm(delta, res, layer.attn_norm.weight, eps)
+            a = layer.attn(n1, cos, sin)
+            n2, res = ops.add_rmsnorm(a, res, layer.mlp_norm.weight, eps)
+            delta = layer.mlp(n2)
+        if delta is None:
+            h = ops.rmsnorm(res, self.norm.weight, eps)
+        else:
+            h, _ = ops.add_rmsnorm(delta, res, self.norm.weight, eps)
+        return self.lm_head(h)
 
     def loss(self, tokens, targets):
         """Forward + fused CE (logits buffer is consumed by the fused op)."""

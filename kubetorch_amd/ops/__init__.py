@@ -60,11 +60,26 @@ def _rmsnorm_ref_fwd(x, w, eps):
     return y, invrms.squeeze(-1).reshape(-1)
 
 
+def _rmsnorm_ref_bwd(dy, ds, s, w, invrms):
+    H = s.shape[-1]
+    sf = s.float().reshape(-1, H)
+    dyf = dy.float().reshape(-1, H)
+    wf = w.float()
+    ir = invrms.reshape(-1, 1)
+    S = (dyf * wf * sf).sum(-1, keepdim=True)
+    g = ir * dyf * wf - sf * (ir ** 3) * S / H
+    if ds is not None:
+        g = g + ds.float().reshape(-1, H)
+    dw = (dyf * sf * ir).sum(0).to(w.dtype)
+    return g.to(s.dtype).reshape(s.shape), dw
+
+
 class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, eps):
         if x.is_cuda:
-            y, invrms = _ext().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+            y, invrms = _ext().rmsnorm_fwd(x.contiguous(), None,
+                                           w.contiguous(), eps)
         else:
             y, invrms = _rmsnorm_ref_fwd(x, w, eps)
         ctx.save_for_backward(x, w, invrms)
@@ -74,23 +89,52 @@ class _RMSNorm(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, invrms = ctx.saved_tensors
         if x.is_cuda:
-            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), x.contiguous(),
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), None, x.contiguous(),
                                         w.contiguous(), invrms)
         else:
-            H = x.shape[-1]
-            xf = x.float().reshape(-1, H)
-            dyf = dy.float().reshape(-1, H)
-            wf = w.float()
-            ir = invrms.reshape(-1, 1)
-            S = (dyf * wf * xf).sum(-1, keepdim=True)
-            dx = (ir * dyf * wf - xf * (ir ** 3) * S / H).to(x.dtype).reshape(x.shape)
-            dw = (dyf * xf * ir).sum(0).to(w.dtype)
+            dx, dw = _rmsnorm_ref_bwd(dy, None, x, w, invrms)
         return dx, dw, None
 
 
 def rmsnorm(x, w, eps=1e-5):
     """y = x * rsqrt(mean(x^2, -1) + eps) * w (bf16 in/out, fp32 accum)."""
     return _RMSNorm.apply(x, w, eps)
+
+
+class _AddRMSNorm(torch.autograd.Function):
+    """Fused pre-norm residual add: (y, s) = (rmsnorm(x + res) * w, x + res).
+    Backward folds the residual fan-in add: dx = dres = ds + d(norm)·dy."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        ctx.set_materialize_grads(False)  # unused s output -> ds is None
+        if x.is_cuda:
+            y, invrms, s = _ext().rmsnorm_fwd(x.contiguous(), res.contiguous(),
+                                              w.contiguous(), eps)
+        else:
+            s = (x.float() + res.float()).to(x.dtype)
+            y, invrms = _rmsnorm_ref_fwd(s, w, eps)
+        ctx.save_for_backward(s, w, invrms)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, w, invrms = ctx.saved_tensors
+        if dy is None:
+            dy = torch.zeros_like(s)
+        if ds is not None:
+            ds = ds.contiguous()
+        if s.is_cuda:
+            g, dw = _ext().rmsnorm_bwd(dy.contiguous(), ds, s, w.contiguous(),
+                                       invrms)
+        else:
+            g, dw = _rmsnorm_ref_bwd(dy, ds, s, w, invrms)
+        return g, g, dw, None
+
+
+def add_rmsnorm(x, res, w, eps=1e-5):
+    """Fused residual add + RMSNorm: returns (normed, x + res)."""
+    return _AddRMSNorm.apply(x, res, w, eps)
 
 
 # ---------------------------------------------------------------------------

@@ -10,11 +10,13 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 extern "C" {
-void kt_rmsnorm_fwd(const void* x, const void* w, void* y, void* invrms,
-                    int N, int H, float eps, hipStream_t stream);
-void kt_rmsnorm_bwd(const void* dy, const void* x, const void* w,
-                    const void* invrms, void* dx, void* dw_partial, void* dw,
-                    int P, int N, int H, hipStream_t stream);
+void kt_rmsnorm_fwd(const void* x, const void* res, const void* w, void* y,
+                    void* s_out, void* invrms, int N, int H, float eps,
+                    hipStream_t stream);
+void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
+                    const void* w, const void* invrms, void* dx,
+                    void* dw_partial, void* dw, int P, int N, int H,
+                    hipStream_t stream);
 void kt_rope(const void* x, void* o, const void* cost, const void* sint,
              long total_quads, int S, int Hh, int D, float sign,
              hipStream_t stream);
@@ -41,8 +43,9 @@ hipStream_t cur_stream(const at::Tensor& t) {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(t.device().index()).stream();
 }
 
-std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w,
-                                    double eps) {
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x,
+                                    const std::optional<at::Tensor>& res,
+                                    const at::Tensor& w, double eps) {
   CHECK_BF16_CONTIG(x);
   CHECK_BF16_CONTIG(w);
   const int H = (int)x.size(-1);
@@ -52,13 +55,25 @@ std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w,
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto y = at::empty_like(x);
   auto invrms = at::empty({N}, x.options().dtype(at::kFloat));
-  kt_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), invrms.data_ptr(),
-                 (int)N, H, (float)eps, cur_stream(x));
+  at::Tensor s;
+  const void* res_ptr = nullptr;
+  void* s_ptr = nullptr;
+  if (res.has_value()) {
+    CHECK_BF16_CONTIG(res.value());
+    TORCH_CHECK(res->sizes() == x.sizes(), "residual shape mismatch");
+    s = at::empty_like(x);
+    res_ptr = res->data_ptr();
+    s_ptr = s.data_ptr();
+  }
+  kt_rmsnorm_fwd(x.data_ptr(), res_ptr, w.data_ptr(), y.data_ptr(), s_ptr,
+                 invrms.data_ptr(), (int)N, H, (float)eps, cur_stream(x));
+  if (res.has_value()) return {y, invrms, s};
   return {y, invrms};
 }
 
-std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
-                                    const at::Tensor& w,
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
+                                    const std::optional<at::Tensor>& ds,
+                                    const at::Tensor& x, const at::Tensor& w,
                                     const at::Tensor& invrms) {
   CHECK_BF16_CONTIG(dy);
   CHECK_BF16_CONTIG(x);
@@ -70,9 +85,14 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dw = at::empty_like(w);
   const int P = N < 512 ? (int)N : 512;
   auto dw_partial = at::empty({P, H}, x.options().dtype(at::kFloat));
-  kt_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(), invrms.data_ptr(),
-                 dx.data_ptr(), dw_partial.data_ptr(), dw.data_ptr(), P,
-                 (int)N, H, cur_stream(x));
+  const void* ds_ptr = nullptr;
+  if (ds.has_value()) {
+    CHECK_BF16_CONTIG(ds.value());
+    ds_ptr = ds->data_ptr();
+  }
+  kt_rmsnorm_bwd(dy.data_ptr(), ds_ptr, x.data_ptr(), w.data_ptr(),
+                 invrms.data_ptr(), dx.data_ptr(), dw_partial.data_ptr(),
+                 dw.data_ptr(), P, (int)N, H, cur_stream(x));
   return {dx, dw};
 }
 

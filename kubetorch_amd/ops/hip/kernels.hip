@@ -51,39 +51,57 @@ __device__ __forceinline__ float block_reduce_sum(float v) {
 }
 
 // ---------------------------------------------------------------------------
-// RMSNorm forward: y = x * rsqrt(mean(x^2) + eps) * w ; saves invrms for bwd.
-// One block per row (grid-stride over rows). x re-read in pass 2 (row is hot
-// in L2; register-caching with runtime-indexed arrays would spill to scratch).
+// (Fused-residual) RMSNorm forward.
+//   s = x (+ res, when res != null); y = s * rsqrt(mean(s^2)+eps) * w
+// When res is given, the summed residual stream s is written to s_out —
+// fusing the transformer's pre-norm residual add into the norm kernel
+// (saves a full elementwise pass per norm). One block per row; s re-read
+// in pass 2 (hot in L2; register caching would spill — see guide rule 20).
 // ---------------------------------------------------------------------------
 __global__ void rmsnorm_fwd_kernel(const u16* __restrict__ x,
+                                   const u16* __restrict__ res,
                                    const u16* __restrict__ w,
                                    u16* __restrict__ y,
+                                   u16* __restrict__ s_out,
                                    float* __restrict__ invrms,
                                    int N, int H, float eps) {
   const int nvec = H >> 3;
   for (int row = blockIdx.x; row < N; row += gridDim.x) {
     const u16* xr = x + (size_t)row * H;
+    const u16* rr = res ? res + (size_t)row * H : nullptr;
     u16* yr = y + (size_t)row * H;
+    u16* sr = s_out ? s_out + (size_t)row * H : nullptr;
     float ss = 0.f;
     for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
       vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+      vec8u sv;
+      if (rr) {
+        vec8u rv = *reinterpret_cast<const vec8u*>(rr + vI * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) sv[j] = f2bf(bf2f(xv[j]) + bf2f(rv[j]));
+        if (sr) *reinterpret_cast<vec8u*>(sr + vI * 8) = sv;
+      } else {
+        sv = xv;
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float f = bf2f(xv[j]);
+        float f = bf2f(sv[j]);
         ss += f * f;
       }
     }
     ss = block_reduce_sum(ss);
     const float ir = rsqrtf(ss / (float)H + eps);
     if (threadIdx.x == 0 && invrms) invrms[row] = ir;
+    const u16* src = (rr && sr) ? sr : xr;  // pass 2 reads the summed stream
     for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
-      vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+      vec8u xv = *reinterpret_cast<const vec8u*>(src + vI * 8);
       vec8u wv = *reinterpret_cast<const vec8u*>(w + vI * 8);
       vec8u ov;
 #pragma unroll
       for (int j = 0; j < 8; ++j) ov[j] = f2bf(bf2f(xv[j]) * ir * bf2f(wv[j]));
       *reinterpret_cast<vec8u*>(yr + vI * 8) = ov;
     }
+    __syncthreads();  // sr written this row must not race the next row
   }
 }
 
@@ -94,7 +112,11 @@ __global__ void rmsnorm_fwd_kernel(const u16* __restrict__ x,
 //          to a [gridDim.x, H] fp32 partial buffer; reduced by colsum below)
 // LDS budget: H fp32 <= 160KB -> H <= 40960 (Llama H=4096 -> 16 KB). Fine.
 // ---------------------------------------------------------------------------
+// ds (nullable): upstream gradient of the summed residual stream s (fused
+// path) — added into dx, so dx = ds + d(norm)/ds·dy, which is the gradient
+// for BOTH inputs of the fused add (they are identical).
 __global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
+                                   const u16* __restrict__ ds,
                                    const u16* __restrict__ x,
                                    const u16* __restrict__ w,
                                    const float* __restrict__ invrms,
@@ -123,16 +145,21 @@ __global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
     }
     S = block_reduce_sum(S);
     const float k = ir * ir * ir * S / (float)H;
+    const u16* dsr = ds ? ds + (size_t)row * H : nullptr;
     for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
       vec8u dyv = *reinterpret_cast<const vec8u*>(dyr + vI * 8);
       vec8u wv = *reinterpret_cast<const vec8u*>(w + vI * 8);
       vec8u xv = *reinterpret_cast<const vec8u*>(xr + vI * 8);
+      vec8u dsv;
+      if (dsr) dsv = *reinterpret_cast<const vec8u*>(dsr + vI * 8);
       vec8u ov;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float dyf = bf2f(dyv[j]);
         float xf = bf2f(xv[j]);
-        ov[j] = f2bf(ir * dyf * bf2f(wv[j]) - xf * k);
+        float g = ir * dyf * bf2f(wv[j]) - xf * k;
+        if (dsr) g += bf2f(dsv[j]);
+        ov[j] = f2bf(g);
         dwacc[vI * 8 + j] += dyf * xf * ir;  // thread owns slot: no conflict
       }
       *reinterpret_cast<vec8u*>(dxr + vI * 8) = ov;
@@ -419,22 +446,25 @@ static inline int grid_for(long work, int block, int cap = 2048) {
 
 extern "C" {
 
-void kt_rmsnorm_fwd(const void* x, const void* w, void* y, void* invrms,
-                    int N, int H, float eps, hipStream_t stream) {
+void kt_rmsnorm_fwd(const void* x, const void* res, const void* w, void* y,
+                    void* s_out, void* invrms, int N, int H, float eps,
+                    hipStream_t stream) {
   int grid = N < 2048 ? N : 2048;
   hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const u16*)x, (const u16*)w, (u16*)y, (float*)invrms, N,
-                     H, eps);
+                     (const u16*)x, (const u16*)res, (const u16*)w, (u16*)y,
+                     (u16*)s_out, (float*)invrms, N, H, eps);
 }
 
-void kt_rmsnorm_bwd(const void* dy, const void* x, const void* w,
-                    const void* invrms, void* dx, void* dw_partial,
-                    void* dw, int P, int N, int H, hipStream_t stream) {
+void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
+                    const void* w, const void* invrms, void* dx,
+                    void* dw_partial, void* dw, int P, int N, int H,
+                    hipStream_t stream) {
   int grid = N < P ? N : P;
   size_t lds = (size_t)H * sizeof(float);
   hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), lds, stream,
-                     (const u16*)dy, (const u16*)x, (const u16*)w,
-                     (const float*)invrms, (u16*)dx, (float*)dw_partial, N, H);
+                     (const u16*)dy, (const u16*)ds, (const u16*)x,
+                     (const u16*)w, (const float*)invrms, (u16*)dx,
+                     (float*)dw_partial, N, H);
   int cgrid = (H + 3) / 4;
   hipLaunchKernelGGL(colsum_bf16_kernel, dim3(cgrid), dim3(256), 0, stream,
                      (const float*)dw_partial, (u16*)dw, grid, H);

@@ -95,6 +95,55 @@ class TestRMSNorm:
         _assert_close(y, ref_rmsnorm(x, w, 1e-5), msg="rmsnorm 4096")
 
 
+class TestAddRMSNorm:
+    def _check(self, device, use_ds=True):
+        torch.manual_seed(0)
+        N, H = 33, 256
+        x = torch.randn(N, H, dtype=BF16, device=device)
+        r = torch.randn(N, H, dtype=BF16, device=device)
+        w = torch.randn(H, dtype=BF16, device=device)
+
+        xi = x.clone().requires_grad_(True)
+        ri = r.clone().requires_grad_(True)
+        wi = w.clone().requires_grad_(True)
+        y, s = ops.add_rmsnorm(xi, ri, wi, 1e-5)
+        gout_y = torch.randn_like(y)
+        if use_ds:
+            gout_s = torch.randn_like(s)
+            torch.autograd.backward([y, s], [gout_y, gout_s])
+        else:
+            y.backward(gout_y)
+
+        xr = x.float().requires_grad_(True)
+        rr = r.float().requires_grad_(True)
+        wr = w.float().requires_grad_(True)
+        sr = xr + rr
+        yr = ref_rmsnorm(sr, wr, 1e-5)
+        if use_ds:
+            torch.autograd.backward([yr, sr], [gout_y.float(), gout_s.float()])
+        else:
+            yr.backward(gout_y.float())
+        _assert_close(y, yr, msg="add_rmsnorm y")
+        _assert_close(s, sr, msg="add_rmsnorm s")
+        _assert_close(xi.grad, xr.grad, msg="add_rmsnorm dx")
+        _assert_close(ri.grad, rr.grad, msg="add_rmsnorm dres")
+        _assert_close(wi.grad, wr.grad, rtol=3e-2, atol=3e-1, msg="add_rmsnorm dw")
+
+    def test_cpu(self):
+        self._check("cpu")
+
+    def test_cpu_no_ds(self):
+        self._check("cpu", use_ds=False)
+
+    @pytest.mark.gpu
+    def test_gpu(self):
+        self._check("cuda")
+
+    @pytest.mark.gpu
+    def test_gpu_no_ds(self):
+        self._check("cuda", use_ds=False)
+
+
 class TestRope:
     def _check(self, device):
         torch.manual_seed(0)
