@@ -267,6 +267,56 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
 
 
 # ---------------------------------------------------------------------------
+# Flash attention: custom gfx950 forward + torch (AITER asm) backward
+# ---------------------------------------------------------------------------
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        o, lse = _ext().attn_fwd(q.contiguous(), k.contiguous(),
+                                 v.contiguous(), scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        q, k, v, o, lse = ctx.saved_tensors
+        B, Hq, S, D = q.shape
+        Hkv = k.shape[1]
+        g = Hq // Hkv
+        if g > 1:  # expand KV for the dense backward, then reduce over groups
+            k_exp = k.repeat_interleave(g, dim=1)
+            v_exp = v.repeat_interleave(g, dim=1)
+        else:
+            k_exp, v_exp = k, v
+        seed = torch.zeros((), dtype=torch.long, device=q.device)
+        offset = torch.zeros((), dtype=torch.long, device=q.device)
+        dq, dk, dv, _ = torch.ops.aten._scaled_dot_product_efficient_attention_backward(
+            grad_out.contiguous(), q, k_exp, v_exp, None, o, lse, seed, offset,
+            0.0, [True, True, True, False], True, scale=ctx.scale,
+        )
+        if g > 1:
+            dk = dk.view(B, Hkv, g, S, D).sum(2)
+            dv = dv.view(B, Hkv, g, S, D).sum(2)
+        return dq, dk, dv, None
+
+
+def flash_attention(q, k, v, scale=None):
+    """Causal GQA attention, bf16, D=128, S % 64 == 0: custom gfx950 MFMA
+    forward (ops/hip/attention.hip) + torch's AITER asm backward.
+    Layout: [B, H, S, D]."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    return _FlashAttention.apply(q, k, v, scale)
+
+
+def flash_attention_supported(q, k, v, is_causal):
+    return (is_causal and q.is_cuda and q.dtype == torch.bfloat16
+            and q.shape[-1] == 128 and q.shape[2] % 64 == 0
+            and hip_available())
+
+
+# ---------------------------------------------------------------------------
 # Fused AdamW on flat bf16 buckets (used by kubetorch_amd.parallel)
 # ---------------------------------------------------------------------------
 def adamw_(p, g, m, v, lr, beta1, beta2, eps, wd, step, grad_scale=1.0):

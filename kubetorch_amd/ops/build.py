@@ -16,7 +16,7 @@ BUILD_DIR = os.path.join(OPS_DIR, "_build")
 SO_PATH = os.path.join(OPS_DIR, "_hip_ops.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-SOURCES = ["kernels.hip", "bindings.cpp"]
+SOURCES = ["kernels.hip", "attention.hip", "bindings.cpp"]
 
 
 def _torch_paths():

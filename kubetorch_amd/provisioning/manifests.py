@@ -187,6 +187,37 @@ def build_raycluster_manifest(service_name, namespace, image, workers=0,
     }
 
 
+def build_training_job_manifest(kind, service_name, namespace, image,
+                                workers=1, username=None, queue=None,
+                                **pod_kw):
+    """TFJob / MXJob / XGBoostJob (Kubeflow training operators); PyTorchJob
+    has its own builder below. (Reference: SUPPORTED_TRAINING_JOBS.)"""
+    kind_map = {"tfjob": ("TFJob", "tfReplicaSpecs"),
+                "mxjob": ("MXJob", "mxReplicaSpecs"),
+                "xgboostjob": ("XGBoostJob", "xgbReplicaSpecs")}
+    if kind not in kind_map:
+        raise ValueError(f"unsupported training job kind {kind!r}")
+    k8s_kind, spec_key = kind_map[kind]
+    pod = build_pod_spec(service_name, image, **pod_kw)
+    labels = _labels(service_name, username)
+    if queue:
+        labels[C.KUEUE_QUEUE_LABEL] = queue
+    return {
+        "apiVersion": "kubeflow.org/v1",
+        "kind": k8s_kind,
+        "metadata": {"name": service_name, "namespace": namespace,
+                     "labels": labels},
+        "spec": {
+            "runPolicy": {"suspend": bool(queue)},
+            spec_key: {
+                "Worker": {"replicas": workers,
+                           "template": {"metadata": {"labels": dict(labels)},
+                                        "spec": pod}},
+            },
+        },
+    }
+
+
 def build_pytorchjob_manifest(service_name, namespace, image, workers=1,
                               num_proc=8, username=None, queue=None, **pod_kw):
     pod = build_pod_spec(service_name, image, **pod_kw)
