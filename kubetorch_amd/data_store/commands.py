@@ -44,7 +44,49 @@ def _local_path(key):
     return p
 
 
-def put(key, src, window=None, timeout=C.HTTP_TIMEOUT):
+_EXCLUDE = ("__pycache__", ".git", ".pytest_cache")
+
+
+def _local_manifest(src):
+    out = {}
+    for root, dirs, files in os.walk(src):
+        dirs[:] = [d for d in dirs if d not in _EXCLUDE]
+        for f in files:
+            if f.endswith(".pyc"):
+                continue
+            full = os.path.join(root, f)
+            st = os.stat(full)
+            out[os.path.relpath(full, src)] = [st.st_size, round(st.st_mtime, 3)]
+    return out
+
+
+def put_delta(key, src, timeout=C.HTTP_TIMEOUT):
+    """Incremental dir upload: diff the store's manifest against local
+    size+mtime, upload only changed files, delete removed ones (the
+    rsync-style hot loop; reference: RsyncClient). Falls back to a full
+    tar upload on the first sync."""
+    url = _store_url()
+    src = os.path.expanduser(str(src))
+    local = _local_manifest(src)
+    r = httpx.get(f"{url}/manifest/{key}", timeout=timeout)
+    remote = r.json().get("files", {}) if r.status_code == 200 else {}
+    if not remote:
+        return put(key, src, timeout=timeout, _delta=False)
+    changed = [p for p, meta in local.items() if remote.get(p) != meta]
+    removed = [p for p in remote if p not in local]
+    for rel in changed:
+        with open(os.path.join(src, rel), "rb") as f:
+            httpx.put(f"{url}/files/{key}/{rel}", content=f.read(),
+                      headers={"X-KT-Mtime": str(local[rel][1])},
+                      timeout=timeout).raise_for_status()
+    if removed:
+        httpx.request("DELETE", f"{url}/manifest/{key}",
+                      json={"paths": removed}, timeout=timeout)
+    return {"key": key, "locale": "store", "changed": len(changed),
+            "removed": len(removed)}
+
+
+def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True):
     """Store a file/dir (by path) or publish GPU tensors under `key`."""
     if _is_gpu_data(src):
         from kubetorch_amd.data_store import gpu_store
@@ -54,6 +96,8 @@ def put(key, src, window=None, timeout=C.HTTP_TIMEOUT):
     if not os.path.exists(src):
         raise FileNotFoundError(src)
     url = _store_url()
+    if url is not None and _delta and os.path.isdir(src):
+        return put_delta(key, src, timeout=timeout)
     if url is None:
         dest = _local_path(key)
         if os.path.isdir(src):

@@ -59,6 +59,12 @@ async def put_file(key: str, request: Request):
     else:
         with open(path, "wb") as f:
             f.write(body)
+        mtime = request.headers.get("X-KT-Mtime")
+        if mtime:  # preserve client mtime so delta-sync manifests compare
+            try:
+                os.utime(path, (float(mtime), float(mtime)))
+            except ValueError:
+                pass
     return {"ok": True, "bytes": len(body)}
 
 
@@ -80,6 +86,35 @@ def get_file(key: str):
                 yield chunk
 
     return StreamingResponse(stream(), media_type="application/octet-stream")
+
+
+@app.get("/manifest/{key:path}")
+def manifest(key: str):
+    """File manifest {relpath: [size, mtime]} under a key — the client diffs
+    against it and uploads only changed files (delta sync, the rsync-style
+    hot loop)."""
+    base = _path_for(key)
+    out = {}
+    if os.path.isdir(base):
+        for root, _dirs, files in os.walk(base):
+            for f in files:
+                full = os.path.join(root, f)
+                st = os.stat(full)
+                out[os.path.relpath(full, base)] = [st.st_size,
+                                                    round(st.st_mtime, 3)]
+    return {"files": out}
+
+
+@app.delete("/manifest/{key:path}")
+async def delete_files(key: str, request: Request):
+    """Delete a list of relative paths under key (delta-sync removals)."""
+    body = await request.json()
+    base = _path_for(key)
+    for rel in body.get("paths", []):
+        p = os.path.abspath(os.path.join(base, rel))
+        if p.startswith(base) and os.path.exists(p):
+            os.remove(p)
+    return {"ok": True}
 
 
 @app.get("/ls")

@@ -84,6 +84,23 @@ def test_file_store_http_server(tmp_store):
         ds.get("ns/run1", dest=str(dest))
         assert (dest / "model.txt").read_text() == "weights"
         assert any(e["key"].endswith("model.txt") for e in ds.ls("ns"))
+        # delta sync: change one file -> only it is re-uploaded
+        import time as _t
+
+        (src / "model.txt").write_text("weights-v2")
+        (src / "extra.txt").write_text("new")
+        _t.sleep(0.01)
+        out = ds.put("ns/run1", src=str(src))
+        assert out.get("changed") == 2, out
+        out2 = ds.put("ns/run1", src=str(src))
+        assert out2.get("changed") == 0, out2
+        os.remove(src / "extra.txt")
+        out3 = ds.put("ns/run1", src=str(src))
+        assert out3.get("removed") == 1, out3
+        dest2 = tmp_store / "down2"
+        ds.get("ns/run1", dest=str(dest2))
+        assert (dest2 / "model.txt").read_text() == "weights-v2"
+        assert not (dest2 / "extra.txt").exists()
         # metadata endpoints
         httpx.post(url + "/meta/ns/k1", json={"host": "1.2.3.4"}).raise_for_status()
         assert httpx.get(url + "/meta/ns/k1").json()["host"] == "1.2.3.4"
