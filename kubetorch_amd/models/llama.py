@@ -68,6 +68,8 @@ class Attention(nn.Module):
         self.wo = nn.Linear(cfg.n_heads * hd, d, bias=False)
 
     def forward(self, x, cos, sin):
+        import os
+
         B, S, _ = x.shape
         hd = self.cfg.head_dim
         qkv = self.wqkv(x)
@@ -78,6 +80,12 @@ class Attention(nn.Module):
         k = ops.rope(k.view(B, S, self.n_kv, hd), cos, sin)
         v = v.view(B, S, self.n_kv, hd)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        if (os.environ.get("KT_ATTN") == "custom"
+                and ops.flash_attention_supported(q, k, v, True)):
+            # experimental: custom gfx950 MFMA forward + AITER backward
+            o = ops.flash_attention(q.contiguous(), k.contiguous(),
+                                    v.contiguous())
+            return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         if self.n_kv != self.n_heads:
             try:
                 o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)

@@ -81,7 +81,13 @@ def build(verbose=True, force=False):
     objs = []
     for src in SOURCES:
         obj = os.path.join(BUILD_DIR, os.path.splitext(src)[0] + ".o")
-        cmd = [hipcc, "-c", os.path.join(HIP_DIR, src), "-o", obj] + flags
+        src_flags = flags
+        if src == "attention.hip":
+            # rocWMMA needs the __half conversions torch's flags disable;
+            # this TU has no torch headers, so drop the half guards.
+            src_flags = [f for f in flags
+                         if not f.startswith("-D__HIP_NO_HALF")]
+        cmd = [hipcc, "-c", os.path.join(HIP_DIR, src), "-o", obj] + src_flags
         if verbose:
             print("[hipcc]", " ".join(cmd))
         subprocess.run(cmd, check=True)

@@ -185,7 +185,7 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "q must be [B,H,S,128]");
   const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
   const int Hkv = (int)k.size(1);
-  TORCH_CHECK(S % 64 == 0, "S must be a multiple of 64");
+  TORCH_CHECK(S % 128 == 0 && S >= 256, "S must be a multiple of 128, >= 256");
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
   auto o = at::empty_like(q);

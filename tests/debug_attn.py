@@ -1,5 +1,5 @@
-"""GPU debug for the custom attention fwd: numerics vs torch SDPA (fwd+bwd),
-LSE check vs aten efficient attention, and perf timing. Run on a GPU box:
+"""GPU debug for the custom attention fwd: numerics vs torch (fwd+bwd via
+the aten efficient-attention kernels), LSE check, perf timing.
 PYTHONPATH=. python tests/debug_attn.py"""
 import time
 
@@ -10,71 +10,71 @@ from kubetorch_amd import ops
 
 torch.manual_seed(0)
 dev = "cuda"
-
-
-def ref_sdpa(q, k, v):
-    from torch.nn.attention import SDPBackend, sdpa_kernel
-
-    with sdpa_kernel(SDPBackend.EFFICIENT_ATTENTION):
-        Hq, Hkv = q.shape[1], k.shape[1]
-        if Hq != Hkv:
-            return F.scaled_dot_product_attention(q, k, v, is_causal=True,
-                                                  enable_gqa=True)
-        return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+aten_fwd = torch.ops.aten._scaled_dot_product_efficient_attention
+aten_bwd = torch.ops.aten._scaled_dot_product_efficient_attention_backward
 
 
 def check(B, Hq, Hkv, S, tag):
+    g = Hq // Hkv
     q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
     k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
     v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    k_exp = k.repeat_interleave(g, 1).contiguous()
+    v_exp = v.repeat_interleave(g, 1).contiguous()
     scale = 128 ** -0.5
 
     o, lse = ops._ext().attn_fwd(q, k, v, scale)
-    # fp32 reference on a slice of heads (memory)
-    with torch.no_grad():
-        ref = F.scaled_dot_product_attention(
-            q.float(), k.repeat_interleave(Hq // Hkv, 1).float(),
-            v.repeat_interleave(Hq // Hkv, 1).float(), is_causal=True)
-    d = (o.float() - ref).abs()
-    print(f"[{tag}] fwd max_abs={d.max().item():.4e} "
-          f"mean={d.mean().item():.2e} nan={o.isnan().any().item()}")
-
-    # lse vs manual
-    with torch.no_grad():
-        scores = torch.einsum(
-            "bhsd,bhtd->bhst", q.float(),
-            k.repeat_interleave(Hq // Hkv, 1).float()) * scale
-        mask = torch.ones(S, S, device=dev, dtype=torch.bool).tril()
-        scores = scores.masked_fill(~mask, float("-inf"))
-        lse_ref = torch.logsumexp(scores, dim=-1)
+    out_ref, lse_ref, seed, offset = aten_fwd(q, k_exp, v_exp, None, True,
+                                              0.0, True, scale=scale)
+    d = (o.float() - out_ref.float()).abs()
     dl = (lse - lse_ref).abs()
-    print(f"[{tag}] lse max_abs={dl.max().item():.4e}")
-    del scores, lse_ref
+    print(f"[{tag}] fwd max={d.max().item():.4e} lse max={dl.max().item():.4e} "
+          f"lse_ref sample={lse_ref.flatten()[:2].tolist()} "
+          f"mine={lse.flatten()[:2].tolist()}")
+    print(f"[{tag}] seed={seed} offset={offset} lse shape ref={tuple(lse_ref.shape)}")
 
-    # full autograd path vs torch SDPA bf16
+    gout = torch.randn_like(o)
+    # bwd with aten's own fwd outputs (sanity)
+    dq0, dk0, dv0, _ = aten_bwd(gout, q, k_exp, v_exp, None, out_ref,
+                                lse_ref, seed, offset, 0.0,
+                                [True, True, True, False], True, scale=scale)
+    # bwd with MY fwd outputs
+    dq1, dk1, dv1, _ = aten_bwd(gout, q, k_exp, v_exp, None, o,
+                                lse, seed, offset, 0.0,
+                                [True, True, True, False], True, scale=scale)
+    for name, a, b in (("dq", dq1, dq0), ("dk", dk1, dk0), ("dv", dv1, dv0)):
+        dd = (a.float() - b.float()).abs()
+        print(f"[{tag}] {name} vs aten-own max={dd.max().item():.4e} "
+              f"nan_mine={a.isnan().any().item()} nan_aten={b.isnan().any().item()}")
+
+    # full wrapper path
     qa = q.clone().requires_grad_(True)
     ka = k.clone().requires_grad_(True)
     va = v.clone().requires_grad_(True)
     out = ops.flash_attention(qa, ka, va)
-    gout = torch.randn_like(out)
     out.backward(gout)
-
-    qb = q.clone().requires_grad_(True)
-    kb = k.clone().requires_grad_(True)
-    vb = v.clone().requires_grad_(True)
-    outb = ref_sdpa(qb, kb, vb)
-    outb.backward(gout)
-    for name, a, b in (("dq", qa.grad, qb.grad), ("dk", ka.grad, kb.grad),
-                       ("dv", va.grad, vb.grad)):
-        dd = (a.float() - b.float()).abs()
-        rel = dd.max() / b.float().abs().max().clamp(min=1e-6)
-        print(f"[{tag}] {name} max_abs={dd.max().item():.4e} rel={rel.item():.3e}")
+    # reference grads in fp32 autograd
+    qr = q.float().requires_grad_(True)
+    kr = k.float().requires_grad_(True)
+    vr = v.float().requires_grad_(True)
+    outr = F.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(g, 1), vr.repeat_interleave(g, 1),
+        is_causal=True)
+    outr.backward(gout.float())
+    for name, a, b in (("dq", qa.grad, qr.grad), ("dk", ka.grad, kr.grad),
+                       ("dv", va.grad, vr.grad)):
+        dd = (a.float() - b).abs()
+        rel = dd.max() / b.abs().max().clamp(min=1e-6)
+        print(f"[{tag}] wrapper {name} max={dd.max().item():.4e} rel={rel.item():.3e}")
 
 
 def perf(B, Hq, Hkv, S, iters=20):
+    g = Hq // Hkv
     q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
     k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
     v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    k_exp = k.repeat_interleave(g, 1).contiguous()
+    v_exp = v.repeat_interleave(g, 1).contiguous()
     scale = 128 ** -0.5
 
     def timeit(fn):
@@ -88,17 +88,17 @@ def perf(B, Hq, Hkv, S, iters=20):
         return (time.perf_counter() - t0) / iters
 
     t_mine = timeit(lambda: ops._ext().attn_fwd(q, k, v, scale))
-    with torch.no_grad():
-        t_torch = timeit(lambda: ref_sdpa(q, k, v))
+    t_torch = timeit(lambda: aten_fwd(q, k_exp, v_exp, None, True, 0.0, True,
+                                      scale=scale))
     flops = 4 * B * Hq * S * S * 128 * 0.5
     print(f"perf B{B} H{Hq}/{Hkv} S{S}: mine {t_mine*1e3:.2f} ms "
-          f"({flops/t_mine/1e12:.0f} TF)  torch {t_torch*1e3:.2f} ms "
+          f"({flops/t_mine/1e12:.0f} TF)  aten-fwd {t_torch*1e3:.2f} ms "
           f"({flops/t_torch/1e12:.0f} TF)")
 
 
 if __name__ == "__main__":
-    check(1, 2, 2, 128, "tiny MHA")
-    check(1, 4, 2, 256, "small GQA")
+    check(1, 2, 2, 256, "tiny MHA")
+    check(1, 4, 2, 384, "small GQA")
     check(2, 8, 2, 1024, "mid GQA")
     perf(4, 32, 8, 4096)
     perf(1, 32, 8, 2048)

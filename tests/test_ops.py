@@ -264,6 +264,51 @@ class TestAdamW:
         self._check("cuda")
 
 
+class TestFlashAttention:
+    """Custom gfx950 attention forward + AITER backward (experimental,
+    opt-in via KT_ATTN=custom). Verifies output, LSE and all grads against
+    fp32 SDPA."""
+
+    @pytest.mark.gpu
+    def test_fwd_bwd_gqa(self):
+        torch.manual_seed(0)
+        B, Hq, Hkv, S, D = 2, 8, 2, 512, 128
+        q = torch.randn(B, Hq, S, D, dtype=BF16, device="cuda",
+                        requires_grad=True)
+        k = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda",
+                        requires_grad=True)
+        v = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda",
+                        requires_grad=True)
+        out = ops.flash_attention(q, k, v)
+        gout = torch.randn_like(out)
+        out.backward(gout)
+
+        g = Hq // Hkv
+        qr = q.detach().float().requires_grad_(True)
+        kr = k.detach().float().requires_grad_(True)
+        vr = v.detach().float().requires_grad_(True)
+        outr = F.scaled_dot_product_attention(
+            qr, kr.repeat_interleave(g, 1), vr.repeat_interleave(g, 1),
+            is_causal=True)
+        outr.backward(gout.float())
+        _assert_close(out, outr, msg="flash fwd")
+        _assert_close(q.grad, qr.grad, msg="flash dq")
+        _assert_close(k.grad, kr.grad, msg="flash dk")
+        _assert_close(v.grad, vr.grad, msg="flash dv")
+
+    @pytest.mark.gpu
+    def test_lse_matches_aten(self):
+        torch.manual_seed(1)
+        q = torch.randn(1, 4, 256, 128, dtype=BF16, device="cuda")
+        k = torch.randn(1, 4, 256, 128, dtype=BF16, device="cuda")
+        v = torch.randn(1, 4, 256, 128, dtype=BF16, device="cuda")
+        o, lse = ops._ext().attn_fwd(q, k, v, 128 ** -0.5)
+        out_ref, lse_ref, _, _ = torch.ops.aten._scaled_dot_product_efficient_attention(
+            q, k, v, None, True, 0.0, True, scale=128 ** -0.5)
+        torch.testing.assert_close(lse, lse_ref, rtol=1e-5, atol=1e-5)
+        _assert_close(o, out_ref, msg="flash vs aten out")
+
+
 @pytest.mark.gpu
 def test_hip_extension_required_on_gpu():
     """On a GPU box the HIP extension must be present and be the code path
