@@ -110,10 +110,23 @@ class HTTPClient:
         if stream_logs:
             streamer = LogStreamer(self.base_url, rid).start()
         try:
-            r = shared_client().post(
-                url, json=body, params=params, headers=headers,
-                timeout=timeout or C.HTTP_TIMEOUT,
-            )
+            last = None
+            for attempt in range(4):
+                try:
+                    r = shared_client().post(
+                        url, json=body, params=params, headers=headers,
+                        timeout=timeout or C.HTTP_TIMEOUT,
+                    )
+                    break
+                except httpx.ConnectError as e:
+                    # connection never established -> safe to retry (pod may
+                    # still be binding its port after a reload/launch)
+                    last = e
+                    import time
+
+                    time.sleep(0.5 * (attempt + 1))
+            else:
+                raise last
         finally:
             if streamer:
                 streamer.stop()
