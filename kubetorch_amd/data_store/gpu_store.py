@@ -13,7 +13,12 @@ import os
 import httpx
 
 from kubetorch_amd import constants as C
-from kubetorch_amd.data_store.pod_data_server import PodDataClient, _my_ip
+from kubetorch_amd.data_store.pod_data_server import (
+    PodDataClient,
+    _my_ip,
+    export_tensor,
+    import_tensor,
+)
 
 _client = None
 
@@ -74,9 +79,12 @@ def _flatten(sd):
 # -- public api ----------------------------------------------------------------
 def put(key, src, window=None):
     """Publish a tensor or state dict. Tensors stay in the owner's memory
-    (zero-copy hipIpc registration); only metadata goes to the store."""
+    (zero-copy hipIpc registration); only metadata goes to the store.
+    With window.world_size > 2, serves a W-party broadcast group instead."""
     import torch
 
+    if window is not None and getattr(window, "world_size", 2) > 2:
+        return put_broadcast(key, src, window)
     cli = _pd_client()
     if isinstance(src, torch.Tensor):
         cli.register(key, src)
@@ -110,9 +118,12 @@ def put(key, src, window=None):
 
 def get(key, dest, window=None):
     """Receive into pre-allocated dest tensor / state dict (shapes must
-    match the published metadata)."""
+    match the published metadata). With window.world_size > 2, joins the
+    W-party broadcast group."""
     import torch
 
+    if window is not None and getattr(window, "world_size", 2) > 2:
+        return get_broadcast(key, dest, window)
     cli = _pd_client()
     meta = get_meta(key)
     source = meta["host"]
@@ -149,6 +160,105 @@ def get(key, dest, window=None):
         return dest
     for sub in sd:
         fetch(f"{key}/{sub}", sd[sub])
+    return dest
+
+
+# -- W-party broadcast (one RCCL/gloo group, fan-out to many receivers) ------
+def _join_group(group_id, world_size, master=None, port=None):
+    url = os.environ.get("KT_STORE_URL")
+    body = {"group_id": group_id, "world_size": world_size}
+    if master:
+        body["master"] = master
+        body["port"] = port
+    if url:
+        r = httpx.post(f"{url}/broadcast/join", json=body, timeout=10)
+        r.raise_for_status()
+        return r.json()
+    # local fallback: file-based coordination with an exclusive lock
+    import fcntl
+
+    path = os.path.join(_meta_root(),
+                        "bcast_" + group_id.replace("/", "_") + ".json")
+    with open(path + ".lock", "w") as lockf:
+        fcntl.flock(lockf, fcntl.LOCK_EX)
+        rec = {"world_size": world_size, "members": 0, "master": None,
+               "port": None}
+        if os.path.exists(path):
+            with open(path) as f:
+                rec = json.load(f)
+        if master:
+            rec.update(master=master, port=port, world_size=world_size,
+                       members=0)
+            rank = 0
+        else:
+            rec["members"] += 1
+            rank = rec["members"]
+        with open(path, "w") as f:
+            json.dump(rec, f)
+        return {"rank": rank, "world_size": rec["world_size"],
+                "master": rec["master"], "port": rec["port"]}
+
+
+def put_broadcast(key, src, window):
+    """Serve `src` (tensor or state dict) to window.world_size-1 receivers
+    through ONE collective group (RCCL on GPU / gloo on CPU). Blocks in the
+    daemon until all receivers join (window.timeout)."""
+    import torch
+
+    cli = _pd_client()
+    sd = {key: src} if isinstance(src, torch.Tensor) else \
+        {f"{key}/{s}": t for s, t in _flatten(src).items()}
+    for k2, t in sd.items():
+        cli.register(k2, t)
+    r = cli.request({"cmd": "serve_bcast", "keys": sorted(sd),
+                     "world_size": window.world_size})
+    if not r["ok"]:
+        raise RuntimeError(r["error"])
+    meta = {
+        "kind": "tensor" if isinstance(src, torch.Tensor) else "state_dict",
+        "host": _my_host(cli),
+        "broadcast": {"group_id": key, "port": r["port"],
+                      "world_size": window.world_size},
+    }
+    if not isinstance(src, torch.Tensor):
+        meta["entries"] = {s: {"shape": tuple(t.shape),
+                               "dtype": str(t.dtype).split(".")[-1],
+                               "numel": t.numel()}
+                          for s, t in _flatten(src).items()}
+    publish_meta(key, meta)
+    _join_group(key, window.world_size, master=r["ip"], port=r["port"])
+    return {"key": key, "broadcast": True, "world_size": window.world_size}
+
+
+def get_broadcast(key, dest, window, poll=0.2):
+    """Join the broadcast group for `key` and receive into dest."""
+    import time
+
+    import torch
+
+    cli = _pd_client()
+    deadline = time.time() + (window.timeout if window else 300)
+    info = None
+    while time.time() < deadline:
+        info = _join_group(key, window.world_size)
+        if info.get("master"):
+            break
+        time.sleep(poll)
+    if not info or not info.get("master"):
+        raise TimeoutError(f"broadcast source for {key!r} never registered")
+    dests = ([dest] if isinstance(dest, torch.Tensor)
+             else [t for _, t in sorted(_flatten(dest).items())])
+    resp = cli.request({
+        "cmd": "join_bcast",
+        "dests": [export_tensor(d) for d in dests],
+        "rank": info["rank"], "world_size": info["world_size"],
+        "port": info["port"], "master_ip": info["master"].split(":")[0],
+    })
+    if not resp["ok"]:
+        raise RuntimeError(resp["error"])
+    if resp.get("payloads"):
+        for d, p in zip(dests, resp["payloads"]):
+            d.copy_(import_tensor(p))
     return dest
 
 

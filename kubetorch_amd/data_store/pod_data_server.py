@@ -126,7 +126,10 @@ class PodDataServer:
     def serve_forever(self):
         self.start()
         while not self._stop.wait(1.0):
-            pass
+            # exit when our socket file is removed (e.g. the owning test/
+            # pod dir was cleaned up) — prevents orphaned daemons piling up
+            if not os.path.exists(self.sock_path):
+                return
 
     def _accept_loop(self, sock):
         while not self._stop.is_set():
@@ -293,7 +296,22 @@ class PodDataClient:
             self._conn.connect(self.sock_path)
         return self._conn
 
+    # commands that block in the daemon (collective joins) get a dedicated
+    # connection so they don't starve other callers of the shared one.
+    _BLOCKING = ("join_bcast", "serve_bcast")
+
     def request(self, msg, retries=2):
+        if msg.get("cmd") in self._BLOCKING:
+            conn = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            conn.connect(self.sock_path)
+            try:
+                send_msg(conn, msg)
+                resp = recv_msg(conn)
+                if resp is None:
+                    raise ConnectionError("server closed connection")
+                return resp
+            finally:
+                conn.close()
         with self._lock:
             for attempt in range(retries + 1):
                 try:

@@ -172,6 +172,52 @@ def del_meta(key: str):
     return {"ok": True}
 
 
+_bcast = {}
+_bcast_lock = threading.Lock()
+
+
+@app.post("/broadcast/join")
+async def broadcast_join(request: Request):
+    """Coordinate a W-party broadcast group: the source registers with
+    rank 0 (+ rendezvous host/port); receivers join and get assigned the
+    next rank. (Reference parity: metadata server join_broadcast.)"""
+    body = await request.json()
+    gid = body["group_id"]
+    with _bcast_lock:
+        rec = _bcast.get(gid)
+        if rec is None:
+            rec = _bcast[gid] = {"world_size": body.get("world_size", 2),
+                                 "members": 0, "master": None, "port": None,
+                                 "created": time.time()}
+        if body.get("master"):
+            rec["master"] = body["master"]
+            rec["port"] = body.get("port")
+            rec["world_size"] = body.get("world_size", rec["world_size"])
+            rec["members"] = 0  # fresh generation
+            rank = 0
+        else:
+            rec["members"] += 1
+            rank = rec["members"]
+        return {"rank": rank, "world_size": rec["world_size"],
+                "master": rec["master"], "port": rec["port"]}
+
+
+@app.get("/broadcast/status")
+def broadcast_status(group_id: str):
+    with _bcast_lock:
+        rec = _bcast.get(group_id)
+    if rec is None:
+        return JSONResponse({"error": "not found"}, status_code=404)
+    return rec
+
+
+@app.delete("/broadcast/{group_id}")
+def broadcast_complete(group_id: str):
+    with _bcast_lock:
+        _bcast.pop(group_id, None)
+    return {"ok": True}
+
+
 @app.post("/logs/push")
 async def logs_push(request: Request):
     body = await request.json()

@@ -249,3 +249,60 @@ def test_gpu_ipc_put_get(tmp_path):
     finally:
         p.terminate()
         p.join(10)
+
+
+def _bcast_receiver(sock, store_root, rank_tag, q):
+    os.environ["KT_GPU_DATA_SOCK"] = sock
+    os.environ["KT_STORE_ROOT"] = store_root
+    os.environ.pop("KT_STORE_URL", None)
+    import kubetorch_amd.data_store.pod_data_server as pds
+
+    pds.SOCK_PATH = sock
+    pds.LOCK_PATH = sock + ".lock"
+    import kubetorch_amd.data_store.gpu_store as gs
+
+    gs._client = None
+    import torch as t
+
+    from kubetorch_amd.data_store.types import BroadcastWindow
+
+    dest = {"w": t.zeros(64), "b": t.zeros(8)}
+    gs.get("bc/sd", dest, window=BroadcastWindow(world_size=3, timeout=60))
+    q.put((rank_tag, dest["w"].sum().item(), dest["b"].sum().item()))
+
+
+def test_three_party_broadcast_cpu(tmp_path, monkeypatch):
+    """1 source + 2 receivers through ONE gloo broadcast group (the RCCL
+    path on GPU nodes; BroadcastWindow coordination via the metadata
+    layer)."""
+    sock = str(tmp_path / "bc.sock")
+    store_root = str(tmp_path / "store")
+    monkeypatch.setenv("KT_GPU_DATA_SOCK", sock)
+    monkeypatch.setenv("KT_STORE_ROOT", store_root)
+    monkeypatch.delenv("KT_STORE_URL", raising=False)
+    import kubetorch_amd.data_store.pod_data_server as pds
+
+    monkeypatch.setattr(pds, "SOCK_PATH", sock)
+    monkeypatch.setattr(pds, "LOCK_PATH", sock + ".lock")
+    import kubetorch_amd.data_store.gpu_store as gs
+
+    gs._client = None
+    import torch as t
+
+    from kubetorch_amd.data_store.types import BroadcastWindow
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    recs = [ctx.Process(target=_bcast_receiver,
+                        args=(sock, store_root, i, q), daemon=True)
+            for i in range(2)]
+    sd = {"w": t.ones(64) * 2, "b": t.ones(8) * 3}
+    gs.put("bc/sd", sd, window=BroadcastWindow(world_size=3, timeout=60))
+    for p in recs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(2)]
+    for _tag, wsum, bsum in results:
+        assert wsum == 128.0 and bsum == 24.0, results
+    for p in recs:
+        p.join(30)
+    gs._client = None
