@@ -16,8 +16,8 @@ class Cls(Module):
         @functools.wraps(lambda: None)
         def remote_method(*args, **kwargs):
             opts = {}
-            for key in ("workers", "restart_procs", "stream_logs", "timeout",
-                        "serialization"):
+            for key in ("workers", "restart_procs", "stream_logs",
+                        "stream_metrics", "timeout", "serialization"):
                 if f"kt_{key}" in kwargs:
                     opts[key] = kwargs.pop(f"kt_{key}")
             return self._call(args, kwargs, method=item, **opts)

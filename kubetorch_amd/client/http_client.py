@@ -71,6 +71,45 @@ class LogStreamer:
             self._poll_once()
 
 
+class MetricsStreamer:
+    """Polls the pod's /metrics during a call and prints hardware lines
+    (GPU util/VRAM/power from the amd-smi collector). Reference parity:
+    per-call metric streaming, http_client.py:758-954."""
+
+    WATCH = ("kt_gpu_utilization_percent", "kt_gpu_vram_used_bytes",
+             "kt_gpu_power_watts", "kt_active_requests")
+
+    def __init__(self, base_url, interval=3.0, printer=print):
+        self.base_url = base_url
+        self.interval = interval
+        self.printer = printer
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._run, daemon=True)
+
+    def start(self):
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+
+    def _run(self):
+        while not self._stop.wait(self.interval):
+            try:
+                r = shared_client().get(self.base_url + "/metrics", timeout=5)
+                vals = {}
+                for line in r.text.splitlines():
+                    for w in self.WATCH:
+                        if line.startswith(w):
+                            vals[line.split()[0]] = float(line.split()[-1])
+                if vals:
+                    pretty = " ".join(f"{k.split('kt_')[-1]}={v:g}"
+                                      for k, v in sorted(vals.items()))
+                    self.printer(f"[remote metrics] {pretty}")
+            except Exception:
+                pass
+
+
 class HTTPClient:
     def __init__(self, base_url, name):
         self.base_url = base_url.rstrip("/")
@@ -88,8 +127,8 @@ class HTTPClient:
             return False
 
     def call(self, args=(), kwargs=None, method=None, serialization="pickle",
-             stream_logs=False, timeout=None, workers=None,
-             restart_procs=False, request_id=None):
+             stream_logs=False, stream_metrics=False, timeout=None,
+             workers=None, restart_procs=False, request_id=None):
         rid = request_id or uuid.uuid4().hex
         url = f"{self.base_url}/call/{self.name}"
         if method:
@@ -107,8 +146,11 @@ class HTTPClient:
         else:
             body = {"args": list(args), "kwargs": kwargs or {}}
         streamer = None
+        mstreamer = None
         if stream_logs:
             streamer = LogStreamer(self.base_url, rid).start()
+        if stream_metrics:
+            mstreamer = MetricsStreamer(self.base_url).start()
         try:
             last = None
             for attempt in range(4):
@@ -130,6 +172,8 @@ class HTTPClient:
         finally:
             if streamer:
                 streamer.stop()
+            if mstreamer:
+                mstreamer.stop()
         try:
             data = r.json()
         except json.JSONDecodeError:

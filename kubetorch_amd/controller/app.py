@@ -26,7 +26,7 @@ import time
 import uuid
 
 from fastapi import FastAPI, Request
-from fastapi.responses import JSONResponse, StreamingResponse
+from fastapi.responses import JSONResponse, Response, StreamingResponse
 
 from kubetorch_amd import constants as C
 from kubetorch_amd.controller.drivers import K8sDriver, LocalDriver
@@ -260,6 +260,41 @@ async def reload_ack(request: Request):
         if not rec["pending"]:
             rec["event"].set()
     return {"ok": True}
+
+
+@app.api_route("/controller/k8s/{path:path}",
+               methods=["GET", "POST", "PUT", "DELETE", "PATCH"])
+async def k8s_proxy(path: str, request: Request):
+    """K8s API passthrough for out-of-cluster clients (reference: nginx
+    /api|/apis routes + controller proxy). Uses the controller pod's RBAC
+    via kubectl --raw; unavailable on the local driver."""
+    import subprocess
+
+    from kubetorch_amd.controller.drivers import K8sDriver
+
+    if not K8sDriver().available():
+        return JSONResponse({"error": "kubectl unavailable (local driver)"},
+                            status_code=501)
+    verb_map = {"GET": "get", "POST": "create", "PUT": "replace",
+                "DELETE": "delete", "PATCH": "patch"}
+    verb = verb_map[request.method]
+    args = ["kubectl", verb, "--raw", "/" + path]
+    body = await request.body()
+    kw = {}
+    if body:
+        args += ["-f", "-"]
+        kw["input"] = body
+    res = await asyncio.to_thread(
+        subprocess.run, args, capture_output=True, **kw)
+    if res.returncode != 0:
+        return JSONResponse({"error": res.stderr.decode()[-2000:]},
+                            status_code=502)
+    import json as _json
+
+    try:
+        return _json.loads(res.stdout)
+    except ValueError:
+        return Response(res.stdout)
 
 
 @app.get("/controller/debug/connections")

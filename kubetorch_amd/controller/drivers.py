@@ -104,9 +104,16 @@ class LocalDriver:
                 env[C.ENV_INIT_ARGS] = md["init_args"]
             if md.get("distributed_config"):
                 env[C.ENV_DISTRIBUTED_CONFIG] = json.dumps(md["distributed_config"])
+            if md.get("module_type") == "app" and md.get("command"):
+                # app mode: the user command IS the pod main process
+                env["KT_APP_PORT"] = str(md.get("port") or port)
+                cmd = ["bash", "-lc", md["command"]]
+            else:
+                cmd = [sys.executable, "-m",
+                       "kubetorch_amd.serving.http_server",
+                       "--port", str(port), "--host", "127.0.0.1"]
             proc = subprocess.Popen(
-                [sys.executable, "-m", "kubetorch_amd.serving.http_server",
-                 "--port", str(port), "--host", "127.0.0.1"],
+                cmd,
                 env=env,
                 stdout=subprocess.DEVNULL,
                 stderr=subprocess.DEVNULL,
