@@ -110,7 +110,14 @@ class TestDistributed:
         ddp = kt.fn(summer_mod.gloo_allreduce).to(
             kt.Compute(cpus=1).distribute("pytorch", workers=2, num_proc=1))
         try:
-            results = ddp(3, kt_restart_procs=True, kt_timeout=180)
+            try:
+                results = ddp(3, kt_restart_procs=True, kt_timeout=120)
+            except Exception:
+                # rendezvous port clash under heavy parallel test load:
+                # full re-provision picks a fresh master port
+                ddp.teardown()
+                ddp.to()
+                results = ddp(3, kt_restart_procs=True, kt_timeout=120)
             assert len(results) == 2
             assert all(r["sum"] == 6.0 for r in results)
             assert sorted(r["rank"] for r in results) == [0, 1]
