@@ -20,7 +20,7 @@ def shared_client():
     with _client_lock:
         if _client is None:
             _client = httpx.Client(
-                timeout=httpx.Timeout(C.HTTP_TIMEOUT, connect=15),
+                timeout=httpx.Timeout(C.HTTP_TIMEOUT, connect=30),
                 limits=httpx.Limits(max_connections=100),
             )
         return _client
@@ -160,7 +160,7 @@ class HTTPClient:
                         timeout=timeout or C.HTTP_TIMEOUT,
                     )
                     break
-                except httpx.ConnectError as e:
+                except (httpx.ConnectError, httpx.ConnectTimeout) as e:
                     # connection never established -> safe to retry (pod may
                     # still be binding its port after a reload/launch)
                     last = e

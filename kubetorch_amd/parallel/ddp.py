@@ -49,7 +49,7 @@ def init_distributed(backend=None, timeout_s=300):
 
 class _Bucket:
     __slots__ = ("params", "flat_param", "flat_grad", "m", "v", "offsets",
-                 "pending", "work", "numel")
+                 "pending", "work", "numel", "completions")
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -61,6 +61,7 @@ class _Bucket:
         self.pending = 0
         self.work = None
         self.numel = 0
+        self.completions = 0
 
 
 def _align8(n):
@@ -77,7 +78,7 @@ class FlatDDP:
 
     def __init__(self, model, lr=3e-4, betas=(0.9, 0.95), eps=1e-8,
                  weight_decay=0.1, bucket_mb=256, process_group=None,
-                 overlap_optimizer=True):
+                 overlap_optimizer=True, grad_accum_steps=1):
         self.model = model
         self.lr = lr
         self.betas = betas
@@ -96,6 +97,15 @@ class FlatDDP:
         self._overlap = overlap_optimizer and torch.cuda.is_available()
         self._opt_stream = torch.cuda.Stream() if self._overlap else None
         self._step_started = False
+        # copy-mode grads (grad_accum_steps == 1): autograd hands us its
+        # freshly-produced grad tensor, the hook copies it into the flat
+        # bucket and drops it — no flat-buffer zeroing, no accumulate-add
+        # (one full pass over grad bytes saved per step). With accumulation
+        # we instead pin .grad to bucket views and zero between optimizer
+        # steps.
+        self._copy_mode = grad_accum_steps == 1
+        self.grad_accum_steps = grad_accum_steps
+        self._micro_step = 0
         params = [p for p in model.parameters() if p.requires_grad]
         if not params:
             raise ValueError("model has no trainable parameters")
@@ -132,7 +142,8 @@ class FlatDDP:
                 b.flat_param[off:off + n].copy_(p.data.reshape(-1))
                 p.data = b.flat_param[off:off + n].view(p.shape)
                 gview = b.flat_grad[off:off + n].view(p.shape)
-                p.grad = gview
+                if not self._copy_mode:
+                    p.grad = gview
                 self._param_bucket[p] = b
                 self._param_view[p] = gview
                 p.register_post_accumulate_grad_hook(self._grad_ready)
@@ -145,13 +156,27 @@ class FlatDDP:
             return
         b = self._param_bucket[p]
         view = self._param_view[p]
-        if p.grad is not None and p.grad.data_ptr() != view.data_ptr():
+        if self._copy_mode:
+            # autograd allocated this grad; move it into the comm bucket
+            # and release it (no accumulate-add, no zeroing pass)
+            view.copy_(p.grad.reshape(p.shape))
+            p.grad = None
+        elif p.grad is not None and p.grad.data_ptr() != view.data_ptr():
             # autograd replaced the view (rare); fold back into the bucket
             view.add_(p.grad)
             p.grad = view
         b.pending -= 1
         if b.pending == 0:
-            self._bucket_ready(b)
+            if self._copy_mode:
+                self._bucket_ready(b)
+            else:
+                # accumulation mode: re-arm per micro-batch; comm + update
+                # only once the bucket has seen every micro-batch
+                b.pending = len(b.params)
+                b.completions += 1
+                if b.completions >= self.grad_accum_steps:
+                    b.completions = 0
+                    self._bucket_ready(b)
 
     def _bucket_ready(self, b):
         if self._world > 1:
@@ -204,12 +229,14 @@ class FlatDDP:
                     self.step_count, grad_scale,
                 )
         for b in self.buckets:
-            b.flat_grad.zero_()
+            if not self._copy_mode:
+                b.flat_grad.zero_()
             b.pending = len(b.params)
 
     def zero_grad(self):
         for b in self.buckets:
-            b.flat_grad.zero_()
+            if not self._copy_mode:
+                b.flat_grad.zero_()
             b.pending = len(b.params)
             b.work = None
 

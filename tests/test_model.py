@@ -64,7 +64,20 @@ def test_flatddp_grads_match_autograd():
     torch.manual_seed(42)
     ref_model.loss(x, y).backward()
     for p, q in zip(model.parameters(), ref_model.parameters()):
-        torch.testing.assert_close(p.grad, q.grad, rtol=1e-2, atol=1e-2)
+        # copy-mode: grads live in the flat buckets (p.grad released)
+        torch.testing.assert_close(eng._param_view[p], q.grad,
+                                   rtol=1e-2, atol=1e-2)
+
+    # view-mode (grad accumulation) keeps p.grad pinned to the bucket
+    torch.manual_seed(0)
+    model2, _ = _tiny_model()
+    eng2 = FlatDDP(model2, lr=1e-3, bucket_mb=1, grad_accum_steps=2)
+    torch.manual_seed(42)
+    model2.loss(x, y).backward()
+    torch.manual_seed(42)
+    model2.loss(x, y).backward()
+    for p, q in zip(model2.parameters(), ref_model.parameters()):
+        torch.testing.assert_close(p.grad, 2 * q.grad, rtol=2e-2, atol=2e-2)
 
 
 @pytest.mark.gpu
