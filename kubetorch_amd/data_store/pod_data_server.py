@@ -402,7 +402,8 @@ def ensure_server(sock_path=SOCK_PATH, tcp_port=None, timeout=120.0):
             env = dict(os.environ)
             env["KT_GPU_DATA_SOCK"] = sock_path
             cmd = [sys.executable, "-m",
-                   "kubetorch_amd.data_store.pod_data_server"]
+                   "kubetorch_amd.data_store.pod_data_server",
+                   "--sock", sock_path]
             if tcp_port:
                 cmd += ["--tcp-port", str(tcp_port)]
             subprocess.Popen(cmd, env=env, stdout=subprocess.DEVNULL,

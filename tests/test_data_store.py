@@ -306,3 +306,34 @@ def test_three_party_broadcast_cpu(tmp_path, monkeypatch):
     for p in recs:
         p.join(30)
     gs._client = None
+
+
+def test_daemon_failure_auto_restart(tmp_path, monkeypatch):
+    """Kill the pod-data-server; the client's next request restarts it
+    (reference parity: test_gpu_store.py failure injection + auto-restart)."""
+    sock = str(tmp_path / "fi.sock")
+    monkeypatch.setenv("KT_GPU_DATA_SOCK", sock)
+    monkeypatch.setenv("KT_STORE_ROOT", str(tmp_path / "store"))
+    import kubetorch_amd.data_store.pod_data_server as pds
+
+    monkeypatch.setattr(pds, "SOCK_PATH", sock)
+    monkeypatch.setattr(pds, "LOCK_PATH", sock + ".lock")
+    cli = pds.PodDataClient(sock)
+    assert cli.ping()["ok"]
+
+    # find and kill the daemon by its exact pid (match on our socket path)
+    import subprocess
+
+    out = subprocess.run(["pgrep", "-af", "pod_data_server"],
+                         capture_output=True, text=True).stdout
+    pids = [line.split()[0] for line in out.splitlines() if sock in line]
+    assert pids, out
+    for pid in pids:
+        subprocess.run(["kill", "-9", pid], check=True)
+    time.sleep(1.0)
+
+    # client retries + auto-restart -> service recovers (registry is fresh)
+    t = torch.randn(8)
+    cli.register("recovered", t)
+    r = cli.request({"cmd": "list"})
+    assert "recovered" in r["keys"]
