@@ -345,7 +345,26 @@ class RaySupervisor(ExecutionSupervisor):
             cmd = "ray start --head --block --dashboard-host=0.0.0.0"
         if cmd:
             self._ray_proc = subprocess.Popen(["bash", "-lc", cmd])
+            self._wait_gcs()
         super().__init__(num_proc=1)
+
+    def _wait_gcs(self, port=6379, timeout=120):
+        """Poll the Ray GCS port until live (reference: ray_supervisor.py
+        GCS liveness check)."""
+        import socket
+        import time
+
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            if self._ray_proc.poll() is not None:
+                raise RuntimeError("ray head exited during startup")
+            try:
+                s = socket.create_connection(("127.0.0.1", port), 1)
+                s.close()
+                return
+            except OSError:
+                time.sleep(0.5)
+        raise RuntimeError(f"Ray GCS not live on :{port} after {timeout}s")
 
     def cleanup(self):
         if self._ray_proc is not None and self._ray_proc.poll() is None:

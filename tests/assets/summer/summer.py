@@ -53,3 +53,16 @@ def slow_echo(x, delay=10):
 
     time.sleep(delay)
     return x
+
+
+def deploy_child_and_call():
+    """In-cluster client: a deployed fn that deploys + calls ANOTHER fn
+    (pod-from-pod; reference: test_pod_from_pod.py)."""
+    import kubetorch_amd as kt
+    from tests_child_helper import double  # resolved via synced workdir
+
+    child = kt.fn(double, name="child-of-pod").to(kt.Compute(cpus=1))
+    try:
+        return child(21)
+    finally:
+        child.teardown()

@@ -123,3 +123,13 @@ class TestDistributed:
             assert sorted(r["rank"] for r in results) == [0, 1]
         finally:
             ddp.teardown()
+
+
+def test_pod_from_pod():
+    """A worker pod acts as a client: deploys and calls a child service
+    through the same controller (in-cluster client path)."""
+    parent = kt.fn(summer_mod.deploy_child_and_call).to(kt.Compute(cpus=1))
+    try:
+        assert parent(kt_timeout=180) == 42
+    finally:
+        parent.teardown()
