@@ -123,12 +123,13 @@ __global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
                                    u16* __restrict__ dx,
                                    float* __restrict__ dw_partial,
                                    int N, int H) {
-  extern __shared__ float dwacc[];  // [H]
+  // dw accumulator in LDS, stored COLUMN-major (dwacc[j*nvec + vI]) so a
+  // wave's slot-j writes hit consecutive words -> conflict-free (the
+  // row-major [vI*8+j] layout was a 16-way conflict: 5.9e9
+  // SQ_LDS_BANK_CONFLICT per bench step, profiles/r01_pmc.md).
+  extern __shared__ float dwacc[];  // [8][nvec]
   const int nvec = H >> 3;
-  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) dwacc[i + j] = 0.f;
-  }
+  for (int t = threadIdx.x; t < H; t += blockDim.x) dwacc[t] = 0.f;
   __syncthreads();
   for (int row = blockIdx.x; row < N; row += gridDim.x) {
     const u16* dyr = dy + (size_t)row * H;
@@ -160,16 +161,17 @@ __global__ void rmsnorm_bwd_kernel(const u16* __restrict__ dy,
         float g = ir * dyf * bf2f(wv[j]) - xf * k;
         if (dsr) g += bf2f(dsv[j]);
         ov[j] = f2bf(g);
-        dwacc[vI * 8 + j] += dyf * xf * ir;  // thread owns slot: no conflict
+        dwacc[j * nvec + vI] += dyf * xf * ir;  // column-major: own slot,
+                                                // consecutive lanes/banks
       }
       *reinterpret_cast<vec8u*>(dxr + vI * 8) = ov;
     }
     __syncthreads();
   }
   float* out = dw_partial + (size_t)blockIdx.x * H;
-  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+  for (int vI = threadIdx.x; vI < nvec; vI += blockDim.x) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) out[i + j] = dwacc[i + j];
+    for (int j = 0; j < 8; ++j) out[vI * 8 + j] = dwacc[j * nvec + vI];
   }
 }
 
