@@ -14,6 +14,28 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
+    config.addinivalue_line(
+        "markers",
+        "flaky_retry: rerun once on failure (multi-process e2e tests are "
+        "sensitive to CI host load spikes)")
+
+
+def pytest_runtest_protocol(item, nextitem):
+    """Retry-once for tests marked flaky_retry: the local-driver e2e tests
+    spawn pods + worker subprocesses and can trip timeouts when the shared
+    CI host stalls; a failure there is re-run once before being reported."""
+    if "flaky_retry" not in item.keywords:
+        return None
+    from _pytest.runner import runtestprotocol
+
+    reports = runtestprotocol(item, nextitem=nextitem, log=False)
+    if any(r.failed for r in reports):
+        reports_retry = runtestprotocol(item, nextitem=nextitem, log=False)
+        if not any(r.failed for r in reports_retry):
+            reports = reports_retry
+    for r in reports:
+        item.ihook.pytest_runtest_logreport(report=r)
+    return True
 
 
 def pytest_collection_modifyitems(config, items):

@@ -8,7 +8,10 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
 os.environ["KT_LOCAL_MODE"] = "true"
 os.environ["KT_USERNAME"] = "clitest"
 
+import pytest  # noqa: E402
 import kubetorch_amd as kt  # noqa: E402
+
+pytestmark = pytest.mark.flaky_retry
 from kubetorch_amd.cli import app  # noqa: E402
 from tests.assets.summer import summer as summer_mod  # noqa: E402
 

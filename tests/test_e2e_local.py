@@ -14,6 +14,8 @@ os.environ["KT_LOCAL_MODE"] = "true"
 os.environ["KT_USERNAME"] = "citest"
 
 import kubetorch_amd as kt  # noqa: E402
+
+pytestmark = pytest.mark.flaky_retry
 from tests.assets.summer import summer as summer_mod  # noqa: E402
 
 

@@ -13,6 +13,8 @@ os.environ["KT_USERNAME"] = "gputest"
 
 import kubetorch_amd as kt  # noqa: E402
 
+pytestmark = pytest.mark.flaky_retry
+
 
 @pytest.mark.gpu
 @pytest.mark.timeout(600)

@@ -11,6 +11,8 @@ os.environ["KT_LOCAL_MODE"] = "true"
 os.environ["KT_USERNAME"] = "rltest"
 
 import kubetorch_amd as kt  # noqa: E402
+
+pytestmark = pytest.mark.flaky_retry
 from tests.assets.rl_services import rl_services  # noqa: E402
 
 
