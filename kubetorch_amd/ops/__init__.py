@@ -50,6 +50,27 @@ def hip_available():
         return False
 
 
+_SPILL = None
+
+
+def _spill_ext():
+    """The native checkpoint spill engine (_hip_spill.so)."""
+    global _SPILL
+    if _SPILL is not None:
+        return _SPILL
+    so = os.path.join(os.path.dirname(__file__), "_hip_spill.so")
+    if not os.path.exists(so):
+        raise RuntimeError(
+            f"spill extension not built: {so} missing "
+            "(python -m kubetorch_amd.ops.build)")
+    spec = importlib.util.spec_from_file_location(
+        "kubetorch_amd.ops._hip_spill", so)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    _SPILL = mod
+    return _SPILL
+
+
 # ---------------------------------------------------------------------------
 # RMSNorm
 # ---------------------------------------------------------------------------

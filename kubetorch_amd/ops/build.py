@@ -13,10 +13,14 @@ import sysconfig
 OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(OPS_DIR, "hip")
 BUILD_DIR = os.path.join(OPS_DIR, "_build")
-SO_PATH = os.path.join(OPS_DIR, "_hip_ops.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-SOURCES = ["kernels.hip", "attention.hip", "bindings.cpp"]
+# module name -> sources (each builds an in-tree .so)
+MODULES = {
+    "_hip_ops": ["kernels.hip", "attention.hip", "bindings.cpp"],
+    "_hip_spill": ["spill.cpp"],
+}
+SO_PATH = os.path.join(OPS_DIR, "_hip_ops.so")  # primary (back-compat)
 
 
 def _torch_paths():
@@ -45,7 +49,6 @@ def _common_flags(includes):
         "-D__HIP_NO_HALF_OPERATORS__=1",
         "-D__HIP_NO_HALF_CONVERSIONS__=1",
         "-DTORCH_API_INCLUDE_EXTENSION_H",
-        "-DTORCH_EXTENSION_NAME=_hip_ops",
     ]
     import torch
 
@@ -57,29 +60,30 @@ def _common_flags(includes):
     return flags
 
 
-def _needs_rebuild():
-    if not os.path.exists(SO_PATH):
+def _needs_rebuild(so_path, sources):
+    if not os.path.exists(so_path):
         return True
-    so_mtime = os.path.getmtime(SO_PATH)
-    for src in SOURCES + ["build.py"]:
-        p = os.path.join(HIP_DIR, src) if src != "build.py" else os.path.join(OPS_DIR, src)
+    so_mtime = os.path.getmtime(so_path)
+    for src in sources + ["build.py"]:
+        p = (os.path.join(HIP_DIR, src) if src != "build.py"
+             else os.path.join(OPS_DIR, src))
         if os.path.getmtime(p) > so_mtime:
             return True
     return False
 
 
-def build(verbose=True, force=False):
-    """Compile kernels.hip + bindings.cpp into _hip_ops.so (idempotent)."""
-    if not force and not _needs_rebuild():
+def _build_module(name, sources, verbose, force):
+    so_path = os.path.join(OPS_DIR, name + ".so")
+    if not force and not _needs_rebuild(so_path, sources):
         if verbose:
-            print(f"[kubetorch_amd.ops.build] up to date: {SO_PATH}")
-        return SO_PATH
+            print(f"[kubetorch_amd.ops.build] up to date: {so_path}")
+        return so_path
     os.makedirs(BUILD_DIR, exist_ok=True)
     includes, libdir = _torch_paths()
-    flags = _common_flags(includes)
+    flags = _common_flags(includes) + [f"-DTORCH_EXTENSION_NAME={name}"]
     hipcc = os.environ.get("HIPCC", "hipcc")
     objs = []
-    for src in SOURCES:
+    for src in sources:
         obj = os.path.join(BUILD_DIR, os.path.splitext(src)[0] + ".o")
         src_flags = flags
         if src == "attention.hip":
@@ -93,7 +97,7 @@ def build(verbose=True, force=False):
         subprocess.run(cmd, check=True)
         objs.append(obj)
     link = (
-        [hipcc, "-shared", "-fPIC", "-o", SO_PATH]
+        [hipcc, "-shared", "-fPIC", "-o", so_path]
         + objs
         + [
             f"-L{libdir}",
@@ -110,8 +114,18 @@ def build(verbose=True, force=False):
         print("[hipcc link]", " ".join(link))
     subprocess.run(link, check=True)
     if verbose:
-        print(f"[kubetorch_amd.ops.build] built {SO_PATH}")
-    return SO_PATH
+        print(f"[kubetorch_amd.ops.build] built {so_path}")
+    return so_path
+
+
+def build(verbose=True, force=False):
+    """Compile every HIP extension module in-tree (idempotent)."""
+    out = None
+    for name, sources in MODULES.items():
+        path = _build_module(name, sources, verbose, force)
+        if name == "_hip_ops":
+            out = path
+    return out
 
 
 if __name__ == "__main__":

@@ -58,6 +58,51 @@ def load_checkpoint(path_or_key, map_location="cpu"):
                       weights_only=False)
 
 
+def save_engine_checkpoint_fast(engine, path, store_key=None):
+    """Native spill of a FlatDDP engine (flat params + fp32 m/v) through the
+    C++ pinned-ring engine (ops/hip/spill.cpp): raw bucket bytes + a JSON
+    manifest. Orders of magnitude less Python overhead than torch.save for
+    the ~100 GB optimizer+param state of an 8B model. Returns GB/s."""
+    import json
+
+    from kubetorch_amd import ops
+
+    tensors = []
+    manifest = {"step": engine.step_count, "buckets": []}
+    for b in engine.buckets:
+        for name, t in (("flat_param", b.flat_param), ("m", b.m), ("v", b.v)):
+            tensors.append(t)
+            manifest["buckets"].append(
+                {"name": name, "numel": t.numel(),
+                 "dtype": str(t.dtype).split(".")[-1]})
+    os.makedirs(os.path.dirname(os.path.abspath(path)) or ".", exist_ok=True)
+    gbps = ops._spill_ext().spill_to_file(tensors, path)
+    with open(path + ".json", "w") as f:
+        json.dump(manifest, f)
+    if store_key:
+        from kubetorch_amd.data_store import commands as ds
+
+        ds.put(store_key, src=path)
+        ds.put(store_key + ".json", src=path + ".json")
+    return gbps
+
+
+def load_engine_checkpoint_fast(engine, path):
+    """Restore a FlatDDP engine spilled by save_engine_checkpoint_fast."""
+    import json
+
+    from kubetorch_amd import ops
+
+    with open(path + ".json") as f:
+        manifest = json.load(f)
+    engine.step_count = manifest["step"]
+    tensors = []
+    for b in engine.buckets:
+        tensors += [b.flat_param, b.m, b.v]
+    assert len(tensors) == len(manifest["buckets"]), "bucket layout mismatch"
+    return ops._spill_ext().restore_from_file(path, tensors)
+
+
 def save_engine_checkpoint(model, engine, path, store_key=None):
     """Model + FlatDDP optimizer state in one file (rank 0 only helper)."""
     sd = {"model": spill_state_dict(model.state_dict()),
