@@ -376,6 +376,7 @@ async def spmd_subcall(request: Request):
         result = await asyncio.to_thread(
             sup.call, serialized_body=payload["body"],
             method=payload.get("method"), distributed_subcall=True,
+            subtree=payload.get("subtree"),
         )
         return {"result": base64.b64encode(pickle.dumps(result)).decode()}
     except BaseException as e:  # noqa: BLE001

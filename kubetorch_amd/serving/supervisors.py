@@ -23,8 +23,8 @@ from kubetorch_amd.exceptions import (
 from kubetorch_amd.serving import discovery
 from kubetorch_amd.serving.process_pool import ProcessPool
 
-TREE_THRESHOLD = 100   # switch to tree fan-out at >=100 workers
-TREE_FANOUT = 50
+TREE_THRESHOLD = int(os.environ.get("KT_TREE_THRESHOLD", "100"))
+TREE_FANOUT = int(os.environ.get("KT_TREE_FANOUT", "50"))
 
 
 # ---------------------------------------------------------------------------
@@ -228,12 +228,32 @@ class SPMDSupervisor(DistributedSupervisor):
             self.pool.restart()
 
         if distributed_subcall:
-            # we are a non-coordinator worker: run local ranks only
+            # non-coordinator worker: run local ranks; in tree mode also
+            # relay to our subtree (fanout bounded per node)
             hosts = json.loads(os.environ.get("KT_SPMD_HOSTS", "[]")) or \
                 self._ordered_hosts()
             me = _self_host()
             node_rank = hosts.index(me) if me in hosts else 0
-            return self._run_local_ranks(body, method, hosts, node_rank, timeout)
+            subtree = _.get("subtree") if isinstance(_.get("subtree"), list) \
+                else None
+            results = self._run_local_ranks(body, method, hosts, node_rank,
+                                            timeout)
+            if subtree and len(subtree) > 1:
+                from kubetorch_amd.serving.remote_pool import (
+                    call_worker_subcall,
+                )
+
+                rest = subtree[1:]
+                with ThreadPoolExecutor(max_workers=min(16, len(rest))) as ex:
+                    futs = [
+                        ex.submit(call_worker_subcall, h, body, method, hosts,
+                                  None, timeout)
+                        for h in rest
+                    ]
+                    for f in futs:
+                        r = f.result()
+                        results.extend(r if isinstance(r, list) else [r])
+            return results
 
         hosts = self._ordered_hosts()
         if workers == "any":

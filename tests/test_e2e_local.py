@@ -135,3 +135,23 @@ def test_pod_from_pod():
         assert parent(kt_timeout=180) == 42
     finally:
         parent.teardown()
+
+
+def test_tree_fanout():
+    """Tree topology: coordinator fans to children which relay to their
+    subtrees (KT_TREE_THRESHOLD lowered so 4 workers use the tree path)."""
+    os.environ["KT_TREE_THRESHOLD"] = "2"
+    os.environ["KT_TREE_FANOUT"] = "2"
+    try:
+        f = kt.fn(summer_mod.rank_env).to(
+            kt.Compute(cpus=1).distribute("pytorch", workers=4, num_proc=1))
+        try:
+            results = f(kt_timeout=240)
+            assert len(results) == 4, results
+            assert sorted(r["rank"] for r in results) == [0, 1, 2, 3]
+            assert all(r["world_size"] == 4 for r in results)
+        finally:
+            f.teardown()
+    finally:
+        os.environ.pop("KT_TREE_THRESHOLD", None)
+        os.environ.pop("KT_TREE_FANOUT", None)
