@@ -1,0 +1,117 @@
+"""Data-plane micro-benchmarks on a GPU box (not a pytest file):
+  * warm remote-call RTT through a deployed service (BASELINE metric 2)
+  * same-node GPU tensor transfer through the pod-data-server (hipIpc
+    device-to-device copy between two processes)
+  * packed vs unpacked state-dict publish/pull
+  * native spill engine bandwidth at multi-GB scale
+Run: PYTHONPATH=. python tests/bench_dataplane.py
+"""
+import json
+import multiprocessing as mp
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
+os.environ.setdefault("KT_LOCAL_MODE", "true")
+os.environ.setdefault("KT_USERNAME", "dpbench")
+
+import torch  # noqa: E402
+
+RESULTS = {}
+
+
+def bench_rtt():
+    import kubetorch_amd as kt
+    from tests.assets.summer import summer as sm
+
+    f = kt.fn(sm.summer).to(kt.Compute(cpus=1))
+    f.stream_logs = False
+    try:
+        for _ in range(20):
+            f(1, 2)
+        n = 200
+        t0 = time.perf_counter()
+        for i in range(n):
+            f(i, i)
+        RESULTS["warm_rtt_ms"] = round((time.perf_counter() - t0) / n * 1e3, 3)
+    finally:
+        f.teardown()
+
+
+def _publisher(q, size_mb):
+    from kubetorch_amd.data_store import gpu_store
+
+    n = size_mb * 1024 * 1024 // 2
+    t = torch.arange(n, dtype=torch.bfloat16, device="cuda")
+    gpu_store.put("dp/big", t)
+    q.put("ok")
+    time.sleep(300)
+
+
+def bench_ipc_transfer(size_mb=1024):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_publisher, args=(q, size_mb), daemon=True)
+    p.start()
+    try:
+        assert q.get(timeout=300) == "ok"
+        from kubetorch_amd.data_store import gpu_store
+
+        n = size_mb * 1024 * 1024 // 2
+        dest = torch.zeros(n, dtype=torch.bfloat16, device="cuda")
+        gpu_store.get("dp/big", dest)  # warm (maps IPC handle)
+        t0 = time.perf_counter()
+        iters = 5
+        for _ in range(iters):
+            gpu_store.get("dp/big", dest)
+        dt = (time.perf_counter() - t0) / iters
+        RESULTS["ipc_d2d_transfer_GBps"] = round(size_mb / 1024 / dt, 2)
+    finally:
+        p.terminate()
+        p.join(10)
+
+
+def bench_state_dict(pack):
+    from kubetorch_amd.data_store import gpu_store
+    from kubetorch_amd.data_store.types import BroadcastWindow
+
+    sd = {f"layer{i}": torch.randn(1024, 4096, dtype=torch.bfloat16,
+                                   device="cuda") for i in range(32)}
+    total_mb = sum(t.numel() * 2 for t in sd.values()) / 2**20
+    key = "dp/sd_packed" if pack else "dp/sd"
+    w = BroadcastWindow(pack=pack) if pack else None
+    t0 = time.perf_counter()
+    gpu_store.put(key, sd, window=w)
+    put_s = time.perf_counter() - t0
+    dest = {k: torch.zeros_like(v) for k, v in sd.items()}
+    t0 = time.perf_counter()
+    gpu_store.get(key, dest)
+    get_s = time.perf_counter() - t0
+    tag = "packed" if pack else "unpacked"
+    RESULTS[f"state_dict_{tag}_put_GBps"] = round(total_mb / 1024 / put_s, 2)
+    RESULTS[f"state_dict_{tag}_get_GBps"] = round(total_mb / 1024 / get_s, 2)
+
+
+def bench_spill(size_gb=4):
+    from kubetorch_amd import ops
+
+    n = size_gb * (1024**3) // 2
+    t = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+    path = "/tmp/spill.bin"
+    out = ops._spill_ext().spill_to_file([t], path)
+    t2 = torch.zeros_like(t)
+    inp = ops._spill_ext().restore_from_file(path, [t2])
+    assert torch.equal(t, t2)
+    RESULTS["spill_d2h_disk_GBps"] = round(out, 2)
+    RESULTS["restore_disk_h2d_GBps"] = round(inp, 2)
+    os.remove(path)
+
+
+if __name__ == "__main__":
+    bench_rtt()
+    bench_ipc_transfer()
+    bench_state_dict(pack=False)
+    bench_state_dict(pack=True)
+    bench_spill()
+    print(json.dumps(RESULTS, indent=1))
