@@ -230,22 +230,19 @@ app = FastAPI(lifespan=lifespan)
 
 
 @app.middleware("http")
-def metrics_middleware(request: Request, call_next):
-    async def inner():
-        t0 = time.time()
-        path = request.url.path.split("/")[1] or "root"
-        metrics.ACTIVE_REQUESTS.inc()
-        try:
-            resp = await call_next(request)
-            metrics.HTTP_REQUESTS.labels(path=path, status=resp.status_code).inc()
-            return resp
-        finally:
-            metrics.ACTIVE_REQUESTS.dec()
-            metrics.HTTP_DURATION.labels(path=path).observe(time.time() - t0)
-            if path in ("call", "spmd"):
-                metrics.touch_activity()
-
-    return inner()
+async def metrics_middleware(request: Request, call_next):
+    t0 = time.time()
+    path = request.url.path.split("/")[1] or "root"
+    metrics.ACTIVE_REQUESTS.inc()
+    try:
+        resp = await call_next(request)
+        metrics.HTTP_REQUESTS.labels(path=path, status=resp.status_code).inc()
+        return resp
+    finally:
+        metrics.ACTIVE_REQUESTS.dec()
+        metrics.HTTP_DURATION.labels(path=path).observe(time.time() - t0)
+        if path in ("call", "spmd"):
+            metrics.touch_activity()
 
 
 @app.get("/health")

@@ -280,10 +280,10 @@ class SPMDSupervisor(DistributedSupervisor):
             self.monitor.start(hosts)
         aborted = []
         self.monitor.subscribe(aborted.append)
+        executor = ThreadPoolExecutor(max_workers=min(64, len(hosts) + 4))
         try:
             remote_hosts = hosts[1:]
             futs = []
-            executor = ThreadPoolExecutor(max_workers=min(64, len(hosts) + 4))
             # remote fan-out (flat; tree for very large worlds)
             from kubetorch_amd.serving.remote_pool import call_worker_subcall
 
