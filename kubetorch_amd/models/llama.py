@@ -80,11 +80,14 @@ class Attention(nn.Module):
         k = ops.rope(k.view(B, S, self.n_kv, hd), cos, sin)
         v = v.view(B, S, self.n_kv, hd)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))
-        if (os.environ.get("KT_ATTN") == "custom"
+        attn_impl = os.environ.get("KT_ATTN", "ck")
+        if (attn_impl in ("ck", "custom")
                 and ops.flash_attention_supported(q, k, v, True)):
-            # experimental: custom gfx950 MFMA forward + AITER backward
-            o = ops.flash_attention(q.contiguous(), k.contiguous(),
-                                    v.contiguous())
+            # DEFAULT: CK-tile FMHA fwd + AITER asm bwd (KT_ATTN=torch for
+            # the SDPA path, KT_ATTN=custom for the in-tree rocWMMA kernel)
+            o = ops.flash_attention(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                impl="wmma" if attn_impl == "custom" else "ck")
             return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         if self.n_kv != self.n_heads:
             try:

@@ -292,9 +292,9 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
 # ---------------------------------------------------------------------------
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
-        o, lse = _ext().attn_fwd(q.contiguous(), k.contiguous(),
-                                 v.contiguous(), scale)
+    def forward(ctx, q, k, v, scale, impl):
+        fwd = _ext().attn_fwd_ck if impl == "ck" else _ext().attn_fwd
+        o, lse = fwd(q.contiguous(), k.contiguous(), v.contiguous(), scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
         return o
@@ -319,7 +319,7 @@ class _FlashAttention(torch.autograd.Function):
         if g > 1:
             dk = dk.view(B, Hkv, g, S, D).sum(2)
             dv = dv.view(B, Hkv, g, S, D).sum(2)
-        return dq, dk, dv, None
+        return dq, dk, dv, None, None
 
 
 def flash_attention(q, k, v, scale=None):

@@ -33,6 +33,9 @@ void kt_adamw(void* p, const void* g, void* m, void* v, long n, float lr,
 void kt_attn_fwd(const void* q, const void* k, const void* v, void* o,
                  void* lse, int B, int Hq, int Hkv, int S, float scale,
                  hipStream_t stream);
+void kt_attn_fwd_ck(const void* q, const void* k, const void* v, void* o,
+                    void* lse, int B, int Hq, int Hkv, int S, float scale,
+                    hipStream_t stream);
 }
 
 namespace {
@@ -195,11 +198,31 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
+std::vector<at::Tensor> attn_fwd_ck(const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& v, double scale) {
+  // CK-tile FMHA fwd: q [B,Hq,S,128], k/v [B,Hkv,S,128], causal, LSE out.
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_CONTIG(v);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "q must be [B,H,S,128]");
+  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
+  const int Hkv = (int)k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  kt_attn_fwd_ck(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                 lse.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
+  return {o, lse};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd", &attn_fwd,
           "Causal GQA attention fwd (bf16, D=128) -> (o, lse)");
+  mod.def("attn_fwd_ck", &attn_fwd_ck,
+          "CK-tile FMHA fwd (bf16, D=128, causal, GQA) -> (o, lse)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16)");
   mod.def("rope", &rope, "RoPE rotate-half (bf16), sign=+1 fwd / -1 bwd");

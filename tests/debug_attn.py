@@ -23,7 +23,7 @@ def check(B, Hq, Hkv, S, tag):
     v_exp = v.repeat_interleave(g, 1).contiguous()
     scale = 128 ** -0.5
 
-    o, lse = ops._ext().attn_fwd(q, k, v, scale)
+    o, lse = ops._ext().attn_fwd_ck(q, k, v, scale)
     out_ref, lse_ref, seed, offset = aten_fwd(q, k_exp, v_exp, None, True,
                                               0.0, True, scale=scale)
     d = (o.float() - out_ref.float()).abs()
@@ -87,12 +87,14 @@ def perf(B, Hq, Hkv, S, iters=20):
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters
 
-    t_mine = timeit(lambda: ops._ext().attn_fwd(q, k, v, scale))
+    t_mine = timeit(lambda: ops._ext().attn_fwd_ck(q, k, v, scale))
+    t_wmma = timeit(lambda: ops._ext().attn_fwd(q, k, v, scale))
     t_torch = timeit(lambda: aten_fwd(q, k_exp, v_exp, None, True, 0.0, True,
                                       scale=scale))
     flops = 4 * B * Hq * S * S * 128 * 0.5
-    print(f"perf B{B} H{Hq}/{Hkv} S{S}: mine {t_mine*1e3:.2f} ms "
-          f"({flops/t_mine/1e12:.0f} TF)  aten-fwd {t_torch*1e3:.2f} ms "
+    print(f"perf B{B} H{Hq}/{Hkv} S{S}: ck {t_mine*1e3:.2f} ms "
+          f"({flops/t_mine/1e12:.0f} TF)  wmma {t_wmma*1e3:.2f} ms "
+          f"({flops/t_wmma/1e12:.0f} TF)  aten {t_torch*1e3:.2f} ms "
           f"({flops/t_torch/1e12:.0f} TF)")
 
 
