@@ -322,19 +322,21 @@ class _FlashAttention(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
-def flash_attention(q, k, v, scale=None):
-    """Causal GQA attention, bf16, D=128, S % 64 == 0: custom gfx950 MFMA
-    forward (ops/hip/attention.hip) + torch's AITER asm backward.
-    Layout: [B, H, S, D]."""
+def flash_attention(q, k, v, scale=None, impl="ck"):
+    """Causal GQA attention fwd, bf16, D=128, layout [B, H, S, D], paired
+    with torch's AITER asm backward. impl="ck" (default) runs the CK-tile
+    FMHA instantiation (1.56x AOTriton on the Llama shape — profiles/);
+    impl="wmma" runs the in-tree rocWMMA kernel (ops/hip/attention.hip,
+    requires S % 128 == 0)."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
-    return _FlashAttention.apply(q, k, v, scale)
+    return _FlashAttention.apply(q, k, v, scale, impl)
 
 
 def flash_attention_supported(q, k, v, is_causal):
     return (is_causal and q.is_cuda and q.dtype == torch.bfloat16
-            and q.shape[-1] == 128 and q.shape[2] % 64 == 0
-            and hip_available())
+            and q.shape[-1] == 128 and q.shape[2] % 128 == 0
+            and q.shape[2] >= 256 and hip_available())
 
 
 # ---------------------------------------------------------------------------
