@@ -293,7 +293,8 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, impl):
-        fwd = _ext().attn_fwd_ck if impl == "ck" else _ext().attn_fwd
+        fwd = (_ext().attn_fwd_ck_tr if impl == "ck"
+               else _ext().attn_fwd)  # "ck" = tr-load CK-tile variant
         o, lse = fwd(q.contiguous(), k.contiguous(), v.contiguous(), scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale

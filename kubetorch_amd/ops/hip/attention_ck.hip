@@ -58,7 +58,45 @@ using FmhaEpilogue = ck_tile::Default2DEpilogue<
 
 using Kernel = ck_tile::FmhaFwdKernel<FmhaPipeline, FmhaEpilogue>;
 
+using FmhaPipelineProblemTr = ck_tile::BlockFmhaPipelineProblem<
+    qkv_t, qkv_t, qkv_t, float, float, qkv_t, uint8_t, float, qkv_t, float,
+    qkv_t, FmhaShape, false, ck_tile::StandardAttention, FmhaMask,
+    true /* kUseTrLoad: gfx950 ds_read_tr path */, FmhaTraits>;
+using FmhaPipelineTr = ck_tile::BlockFmhaPipelineQRKSVSAsync<FmhaPipelineProblemTr>;
+using KernelTr = ck_tile::FmhaFwdKernel<FmhaPipelineTr, FmhaEpilogue>;
+
+template <typename K>
+void run_fmha(const void* q, const void* k, const void* v, void* o,
+              void* lse, int B, int Hq, int Hkv, int S, float scale,
+              hipStream_t stream) {
+  const ck_tile::index_t D = 128;
+  auto kargs = K::MakeKargs(
+      q, k, v, nullptr, nullptr, lse, o, S, S, D, D, Hq, Hq / Hkv, scale,
+      1.0f, 1.0f, 0.0f, D, D, D, 0, 0, D,
+      (ck_tile::index_t)S * D, (ck_tile::index_t)S * D,
+      (ck_tile::index_t)S * D, 0, 0, S, (ck_tile::index_t)S * D,
+      (ck_tile::index_t)Hq * S * D, (ck_tile::index_t)Hkv * S * D,
+      (ck_tile::index_t)Hkv * S * D, 0, 0, (ck_tile::index_t)Hq * S,
+      (ck_tile::index_t)Hq * S * D, -1, 0,
+      (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
+      0.0f, false, std::make_tuple<uint64_t, uint64_t>(0, 0));
+  dim3 grid = K::GridSize(B, Hq, S, D, false);
+  ck_tile::stream_config cfg{};
+  cfg.stream_id_ = stream;
+  cfg.cold_niters_ = 0;
+  cfg.nrepeat_ = 1;
+  ck_tile::launch_kernel(cfg, ck_tile::make_kernel<K::kBlockPerCu>(
+                                  K{}, grid, K::kBlockSize, 0, kargs));
+}
+
 }  // namespace
+
+extern "C" void kt_attn_fwd_ck_tr(const void* q, const void* k,
+                                  const void* v, void* o, void* lse, int B,
+                                  int Hq, int Hkv, int S, float scale,
+                                  hipStream_t stream) {
+  run_fmha<KernelTr>(q, k, v, o, lse, B, Hq, Hkv, S, scale, stream);
+}
 
 extern "C" void kt_attn_fwd_ck(const void* q, const void* k, const void* v,
                                void* o, void* lse, int B, int Hq, int Hkv,

@@ -36,6 +36,9 @@ void kt_attn_fwd(const void* q, const void* k, const void* v, void* o,
 void kt_attn_fwd_ck(const void* q, const void* k, const void* v, void* o,
                     void* lse, int B, int Hq, int Hkv, int S, float scale,
                     hipStream_t stream);
+void kt_attn_fwd_ck_tr(const void* q, const void* k, const void* v, void* o,
+                       void* lse, int B, int Hq, int Hkv, int S, float scale,
+                       hipStream_t stream);
 }
 
 namespace {
@@ -198,8 +201,10 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
-std::vector<at::Tensor> attn_fwd_ck(const at::Tensor& q, const at::Tensor& k,
-                                    const at::Tensor& v, double scale) {
+std::vector<at::Tensor> attn_fwd_ck_impl(const at::Tensor& q,
+                                         const at::Tensor& k,
+                                         const at::Tensor& v, double scale,
+                                         bool trload) {
   // CK-tile FMHA fwd: q [B,Hq,S,128], k/v [B,Hkv,S,128], causal, LSE out.
   CHECK_BF16_CONTIG(q);
   CHECK_BF16_CONTIG(k);
@@ -211,9 +216,21 @@ std::vector<at::Tensor> attn_fwd_ck(const at::Tensor& q, const at::Tensor& k,
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
   auto o = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
-  kt_attn_fwd_ck(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                 lse.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
+  auto fn = trload ? kt_attn_fwd_ck_tr : kt_attn_fwd_ck;
+  fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+     lse.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
   return {o, lse};
+}
+
+std::vector<at::Tensor> attn_fwd_ck(const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& v, double scale) {
+  return attn_fwd_ck_impl(q, k, v, scale, false);
+}
+
+std::vector<at::Tensor> attn_fwd_ck_tr(const at::Tensor& q,
+                                       const at::Tensor& k,
+                                       const at::Tensor& v, double scale) {
+  return attn_fwd_ck_impl(q, k, v, scale, true);
 }
 
 }  // namespace
@@ -223,6 +240,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "Causal GQA attention fwd (bf16, D=128) -> (o, lse)");
   mod.def("attn_fwd_ck", &attn_fwd_ck,
           "CK-tile FMHA fwd (bf16, D=128, causal, GQA) -> (o, lse)");
+  mod.def("attn_fwd_ck_tr", &attn_fwd_ck_tr,
+          "CK-tile FMHA fwd, gfx950 tr-load variant -> (o, lse)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16)");
   mod.def("rope", &rope, "RoPE rotate-half (bf16), sign=+1 fwd / -1 bwd");

@@ -24,10 +24,14 @@ def check(B, Hq, Hkv, S, tag):
     scale = 128 ** -0.5
 
     o, lse = ops._ext().attn_fwd_ck(q, k, v, scale)
+    o_tr, lse_tr = ops._ext().attn_fwd_ck_tr(q, k, v, scale)
     out_ref, lse_ref, seed, offset = aten_fwd(q, k_exp, v_exp, None, True,
                                               0.0, True, scale=scale)
     d = (o.float() - out_ref.float()).abs()
     dl = (lse - lse_ref).abs()
+    dtr = (o_tr.float() - out_ref.float()).abs()
+    print(f"[{tag}] tr fwd max={dtr.max().item():.4e} "
+          f"tr lse max={(lse_tr - lse_ref).abs().max().item():.4e}")
     print(f"[{tag}] fwd max={d.max().item():.4e} lse max={dl.max().item():.4e} "
           f"lse_ref sample={lse_ref.flatten()[:2].tolist()} "
           f"mine={lse.flatten()[:2].tolist()}")
@@ -88,14 +92,14 @@ def perf(B, Hq, Hkv, S, iters=20):
         return (time.perf_counter() - t0) / iters
 
     t_mine = timeit(lambda: ops._ext().attn_fwd_ck(q, k, v, scale))
+    t_tr = timeit(lambda: ops._ext().attn_fwd_ck_tr(q, k, v, scale))
     t_wmma = timeit(lambda: ops._ext().attn_fwd(q, k, v, scale))
     t_torch = timeit(lambda: aten_fwd(q, k_exp, v_exp, None, True, 0.0, True,
                                       scale=scale))
     flops = 4 * B * Hq * S * S * 128 * 0.5
-    print(f"perf B{B} H{Hq}/{Hkv} S{S}: ck {t_mine*1e3:.2f} ms "
-          f"({flops/t_mine/1e12:.0f} TF)  wmma {t_wmma*1e3:.2f} ms "
-          f"({flops/t_wmma/1e12:.0f} TF)  aten {t_torch*1e3:.2f} ms "
-          f"({flops/t_torch/1e12:.0f} TF)")
+    print(f"perf B{B} H{Hq}/{Hkv} S{S}: ck {flops/t_mine/1e12:.0f} TF  "
+          f"ck_tr {flops/t_tr/1e12:.0f} TF  wmma {flops/t_wmma/1e12:.0f} TF  "
+          f"aten {flops/t_torch/1e12:.0f} TF")
 
 
 if __name__ == "__main__":
