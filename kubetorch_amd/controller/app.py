@@ -71,12 +71,16 @@ def _parse_ttl(raw):
         return None
 
 
-async def _ttl_reaper(interval=30.0):
+async def _ttl_reaper(interval=None):
     """Tear down workloads whose pods report no activity for longer than
     their inactivity TTL (reference: controller TTL reaper fed by the
     kt_last_activity_timestamp metric)."""
     import httpx
 
+    if interval is None:
+        interval = float(os.environ.get("KT_TTL_REAPER_INTERVAL", "30"))
+
+    grace = float(os.environ.get("KT_TTL_GRACE", "60"))
     while True:
         await asyncio.sleep(interval)
         for (ns, name), w in list(HUB.workloads.items()):
@@ -85,6 +89,9 @@ async def _ttl_reaper(interval=30.0):
                 .get("annotations", {}).get(C.INACTIVITY_TTL_ANNOTATION))
             if not ttl:
                 continue
+            if time.time() - w.get("created", 0) < ttl + grace:
+                continue  # launch grace: pods report no activity until the
+                          # first call lands
             last = w.get("updated", 0)
             try:
                 pods = HUB.driver.pods(name, ns)

@@ -155,3 +155,33 @@ def test_tree_fanout():
     finally:
         os.environ.pop("KT_TREE_THRESHOLD", None)
         os.environ.pop("KT_TREE_FANOUT", None)
+
+
+def test_inactivity_ttl_reaper():
+    """A workload with inactivity_ttl is torn down by the controller reaper
+    once its pods report no activity for longer than the TTL."""
+    os.environ["KT_TTL_REAPER_INTERVAL"] = "1"
+    os.environ["KT_TTL_GRACE"] = "0"
+    from kubetorch_amd.controller import local as ctl_local
+
+    # restart the in-process controller so the fast reaper interval applies
+    ctl_local.shutdown_local_controller()
+    import kubetorch_amd.globals as g
+
+    g._controller = None
+    try:
+        f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1, inactivity_ttl="5s"))
+        name, ns = f.name, f.namespace
+        assert f(1, 1) == 2
+        from kubetorch_amd.globals import controller_client
+
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            if controller_client().get_workload(name, ns) is None:
+                break
+            time.sleep(1)
+        assert controller_client().get_workload(name, ns) is None, \
+            "workload not reaped after TTL"
+    finally:
+        os.environ.pop("KT_TTL_REAPER_INTERVAL", None)
+        os.environ.pop("KT_TTL_GRACE", None)
