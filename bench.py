@@ -43,6 +43,9 @@ def main():
     from kubetorch_amd.parallel import FlatDDP, init_distributed
 
     rank, world, local_rank = init_distributed()
+    if world != args.gpus:
+        log(f"[bench] note: WORLD_SIZE={world} != --gpus {args.gpus}; "
+            f"reporting actual world size {world}")
     if torch.cuda.is_available():
         dev = torch.device("cuda", local_rank % torch.cuda.device_count())
     else:
