@@ -84,10 +84,12 @@ class Attention(nn.Module):
         if (attn_impl in ("ck", "custom")
                 and ops.flash_attention_supported(q, k, v, True)):
             # DEFAULT: CK-tile FMHA fwd + AITER asm bwd (KT_ATTN=torch for
-            # the SDPA path, KT_ATTN=custom for the in-tree rocWMMA kernel)
+            # the SDPA path, KT_ATTN=custom for the in-tree rocWMMA kernel).
+            # The CK path is stride-aware: the permuted [B,S,H,D] views go
+            # in directly and O comes back in the same layout, so the whole
+            # attention block runs without a single transpose copy.
             o = ops.flash_attention(
-                q.contiguous(), k.contiguous(), v.contiguous(),
-                impl="wmma" if attn_impl == "custom" else "ck")
+                q, k, v, impl="wmma" if attn_impl == "custom" else "ck")
             return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         if self.n_kv != self.n_heads:
             try:

@@ -293,9 +293,12 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, impl):
-        fwd = (_ext().attn_fwd_ck_tr if impl == "ck"
-               else _ext().attn_fwd)  # "ck" = tr-load CK-tile variant
-        o, lse = fwd(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+        if impl == "ck":
+            # CK path is stride-aware: permuted [B,S,H,D] views go in as-is
+            o, lse = _ext().attn_fwd_ck_tr(q, k, v, scale)
+        else:
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+            o, lse = _ext().attn_fwd(q, k, v, scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
         return o

@@ -7,6 +7,7 @@
 // pair it with torch's AITER backward in ops.flash_attention.
 // (Upstream dispatch `fmha_fwd()` is codegen at CK build time and is not
 // shipped in /opt/rocm, so we instantiate the kernel template directly.)
+#include <cstring>
 #include <ck_tile/core.hpp>
 #include <ck_tile/host/kernel_launch.hpp>
 #include <ck_tile/ops/epilogue.hpp>
@@ -65,19 +66,29 @@ using FmhaPipelineProblemTr = ck_tile::BlockFmhaPipelineProblem<
 using FmhaPipelineTr = ck_tile::BlockFmhaPipelineQRKSVSAsync<FmhaPipelineProblemTr>;
 using KernelTr = ck_tile::FmhaFwdKernel<FmhaPipelineTr, FmhaEpilogue>;
 
+struct FmhaStrides {
+  // element strides along (seqlen, head, batch) per tensor
+  long q_s, q_h, q_b;
+  long k_s, k_h, k_b;
+  long v_s, v_h, v_b;
+  long o_s, o_h, o_b;
+};
+
 template <typename K>
 void run_fmha(const void* q, const void* k, const void* v, void* o,
               void* lse, int B, int Hq, int Hkv, int S, float scale,
-              hipStream_t stream) {
+              const FmhaStrides& st, hipStream_t stream) {
   const ck_tile::index_t D = 128;
   auto kargs = K::MakeKargs(
       q, k, v, nullptr, nullptr, lse, o, S, S, D, D, Hq, Hq / Hkv, scale,
-      1.0f, 1.0f, 0.0f, D, D, D, 0, 0, D,
-      (ck_tile::index_t)S * D, (ck_tile::index_t)S * D,
-      (ck_tile::index_t)S * D, 0, 0, S, (ck_tile::index_t)S * D,
-      (ck_tile::index_t)Hq * S * D, (ck_tile::index_t)Hkv * S * D,
-      (ck_tile::index_t)Hkv * S * D, 0, 0, (ck_tile::index_t)Hq * S,
-      (ck_tile::index_t)Hq * S * D, -1, 0,
+      1.0f, 1.0f, 0.0f,
+      (ck_tile::index_t)st.q_s, (ck_tile::index_t)st.k_s,
+      (ck_tile::index_t)st.v_s, 0, 0, (ck_tile::index_t)st.o_s,
+      (ck_tile::index_t)st.q_h, (ck_tile::index_t)st.k_h,
+      (ck_tile::index_t)st.v_h, 0, 0, S, (ck_tile::index_t)st.o_h,
+      (ck_tile::index_t)st.q_b, (ck_tile::index_t)st.k_b,
+      (ck_tile::index_t)st.v_b, 0, 0, (ck_tile::index_t)Hq * S,
+      (ck_tile::index_t)st.o_b, -1, 0,
       (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
       0.0f, false, std::make_tuple<uint64_t, uint64_t>(0, 0));
   dim3 grid = K::GridSize(B, Hq, S, D, false);
@@ -94,47 +105,17 @@ void run_fmha(const void* q, const void* k, const void* v, void* o,
 extern "C" void kt_attn_fwd_ck_tr(const void* q, const void* k,
                                   const void* v, void* o, void* lse, int B,
                                   int Hq, int Hkv, int S, float scale,
-                                  hipStream_t stream) {
-  run_fmha<KernelTr>(q, k, v, o, lse, B, Hq, Hkv, S, scale, stream);
+                                  const long* strides, hipStream_t stream) {
+  FmhaStrides st;
+  memcpy(&st, strides, sizeof(st));
+  run_fmha<KernelTr>(q, k, v, o, lse, B, Hq, Hkv, S, scale, st, stream);
 }
 
 extern "C" void kt_attn_fwd_ck(const void* q, const void* k, const void* v,
                                void* o, void* lse, int B, int Hq, int Hkv,
-                               int S, float scale, hipStream_t stream) {
-  // layouts: q/o [B, Hq, S, 128]; k/v [B, Hkv, S, 128]; lse [B, Hq, S] fp32
-  const ck_tile::index_t D = 128;
-  auto kargs = Kernel::MakeKargs(
-      q, k, v,
-      nullptr /* bias */, nullptr /* rand_val */, lse, o,
-      S /* seqlen_q */, S /* seqlen_k */, D /* hdim_q */, D /* hdim_v */,
-      Hq /* num_head_q */, Hq / Hkv /* nhead_ratio_qk */,
-      scale /* scale_s */, 1.0f /* scale_p */, 1.0f /* scale_o */,
-      0.0f /* logits_soft_cap */,
-      D /* stride_q */, D /* stride_k */, D /* stride_v */,
-      0 /* stride_bias */, 0 /* stride_randval */, D /* stride_o */,
-      (ck_tile::index_t)S * D /* nhead_stride_q */,
-      (ck_tile::index_t)S * D /* nhead_stride_k */,
-      (ck_tile::index_t)S * D /* nhead_stride_v */,
-      0 /* nhead_stride_bias */, 0 /* nhead_stride_randval */,
-      S /* nhead_stride_lse */,
-      (ck_tile::index_t)S * D /* nhead_stride_o */,
-      (ck_tile::index_t)Hq * S * D /* batch_stride_q */,
-      (ck_tile::index_t)Hkv * S * D /* batch_stride_k */,
-      (ck_tile::index_t)Hkv * S * D /* batch_stride_v */,
-      0 /* batch_stride_bias */, 0 /* batch_stride_randval */,
-      (ck_tile::index_t)Hq * S /* batch_stride_lse */,
-      (ck_tile::index_t)Hq * S * D /* batch_stride_o */,
-      -1 /* window_size_left */, 0 /* window_size_right */,
-      (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
-      0.0f /* p_drop */, false /* s_randval */,
-      std::make_tuple<uint64_t, uint64_t>(0, 0));
-
-  dim3 grid = Kernel::GridSize(B, Hq, S, D, false);
-  ck_tile::stream_config cfg{};
-  cfg.stream_id_ = stream;
-  cfg.cold_niters_ = 0;
-  cfg.nrepeat_ = 1;
-  ck_tile::launch_kernel(
-      cfg, ck_tile::make_kernel<Kernel::kBlockPerCu>(
-               Kernel{}, grid, Kernel::kBlockSize, 0, kargs));
+                               int S, float scale, const long* strides,
+                               hipStream_t stream) {
+  FmhaStrides st;
+  memcpy(&st, strides, sizeof(st));
+  run_fmha<Kernel>(q, k, v, o, lse, B, Hq, Hkv, S, scale, st, stream);
 }

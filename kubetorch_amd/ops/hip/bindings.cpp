@@ -35,10 +35,10 @@ void kt_attn_fwd(const void* q, const void* k, const void* v, void* o,
                  hipStream_t stream);
 void kt_attn_fwd_ck(const void* q, const void* k, const void* v, void* o,
                     void* lse, int B, int Hq, int Hkv, int S, float scale,
-                    hipStream_t stream);
+                    const long* strides, hipStream_t stream);
 void kt_attn_fwd_ck_tr(const void* q, const void* k, const void* v, void* o,
                        void* lse, int B, int Hq, int Hkv, int S, float scale,
-                       hipStream_t stream);
+                       const long* strides, hipStream_t stream);
 }
 
 namespace {
@@ -206,19 +206,33 @@ std::vector<at::Tensor> attn_fwd_ck_impl(const at::Tensor& q,
                                          const at::Tensor& v, double scale,
                                          bool trload) {
   // CK-tile FMHA fwd: q [B,Hq,S,128], k/v [B,Hkv,S,128], causal, LSE out.
-  CHECK_BF16_CONTIG(q);
-  CHECK_BF16_CONTIG(k);
-  CHECK_BF16_CONTIG(v);
+  // Strided inputs are supported (last dim must be contiguous) — [B,S,H,D]
+  // permuted views go straight in, no transpose copies.
+#define CHECK_BF16_LASTC(t)                                             \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                     \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16");  \
+  TORCH_CHECK((t).stride(3) == 1, #t " last dim must be contiguous")
+  CHECK_BF16_LASTC(q);
+  CHECK_BF16_LASTC(k);
+  CHECK_BF16_LASTC(v);
+#undef CHECK_BF16_LASTC
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "q must be [B,H,S,128]");
   const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
   const int Hkv = (int)k.size(1);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
-  auto o = at::empty_like(q);
+  // write O in the same (possibly permuted) layout as q
+  auto o = at::empty_strided(q.sizes(), q.strides(), q.options());
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  const long strides[12] = {
+      q.stride(2), q.stride(1), q.stride(0),
+      k.stride(2), k.stride(1), k.stride(0),
+      v.stride(2), v.stride(1), v.stride(0),
+      o.stride(2), o.stride(1), o.stride(0),
+  };
   auto fn = trload ? kt_attn_fwd_ck_tr : kt_attn_fwd_ck;
   fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-     lse.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
+     lse.data_ptr(), B, Hq, Hkv, S, (float)scale, strides, cur_stream(q));
   return {o, lse};
 }
 

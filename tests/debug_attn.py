@@ -14,11 +14,20 @@ aten_fwd = torch.ops.aten._scaled_dot_product_efficient_attention
 aten_bwd = torch.ops.aten._scaled_dot_product_efficient_attention_backward
 
 
-def check(B, Hq, Hkv, S, tag):
+def check(B, Hq, Hkv, S, tag, strided=False):
     g = Hq // Hkv
-    q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
-    k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
-    v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    if strided:
+        # [B,S,H,D] storage, permuted views (the model layout, no copies)
+        q = torch.randn(B, S, Hq, 128, dtype=torch.bfloat16,
+                        device=dev).permute(0, 2, 1, 3)
+        k = torch.randn(B, S, Hkv, 128, dtype=torch.bfloat16,
+                        device=dev).permute(0, 2, 1, 3)
+        v = torch.randn(B, S, Hkv, 128, dtype=torch.bfloat16,
+                        device=dev).permute(0, 2, 1, 3)
+    else:
+        q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+        k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+        v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
     k_exp = k.repeat_interleave(g, 1).contiguous()
     v_exp = v.repeat_interleave(g, 1).contiguous()
     scale = 128 ** -0.5
@@ -106,6 +115,7 @@ if __name__ == "__main__":
     check(1, 2, 2, 256, "tiny MHA")
     check(1, 4, 2, 384, "small GQA")
     check(2, 8, 2, 1024, "mid GQA")
+    check(2, 8, 2, 1024, "strided GQA", strided=True)
     perf(4, 32, 8, 4096)
     perf(1, 32, 8, 2048)
     print("done")
