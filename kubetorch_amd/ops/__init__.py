@@ -190,7 +190,8 @@ class _Rope(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         if x.is_cuda:
             B, S, Hh, D = x.shape
-            return _ext().rope(x.contiguous().view(B * S, Hh, D), cos, sin, S, 1.0).view(x.shape)
+            xv = x.reshape(B * S, Hh, D)  # strided views pass through
+            return _ext().rope(xv, cos, sin, S, 1.0).view(x.shape)
         return _rope_ref(x, cos, sin, 1.0)
 
     @staticmethod
@@ -198,7 +199,8 @@ class _Rope(torch.autograd.Function):
         cos, sin = ctx.saved_tensors
         if dy.is_cuda:
             B, S, Hh, D = dy.shape
-            dx = _ext().rope(dy.contiguous().view(B * S, Hh, D), cos, sin, S, -1.0).view(dy.shape)
+            dyv = dy.reshape(B * S, Hh, D)
+            dx = _ext().rope(dyv, cos, sin, S, -1.0).view(dy.shape)
         else:
             dx = _rope_ref(dy, cos, sin, -1.0)
         return dx, None, None

@@ -18,8 +18,8 @@ void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
                     void* dw_partial, void* dw, int P, int N, int H,
                     hipStream_t stream);
 void kt_rope(const void* x, void* o, const void* cost, const void* sint,
-             long total_quads, int S, int Hh, int D, float sign,
-             hipStream_t stream);
+             long total_quads, int S, int Hh, int D, long src_t_stride,
+             long src_h_stride, float sign, hipStream_t stream);
 void kt_swiglu_fwd(const void* gu, void* out, long N, int I,
                    hipStream_t stream);
 void kt_swiglu_bwd(const void* dout, const void* gu, void* dgu, long N, int I,
@@ -107,19 +107,22 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
 
 at::Tensor rope(const at::Tensor& x, const at::Tensor& cost,
                 const at::Tensor& sint, int64_t S, double sign) {
-  // x: [T, Hh, D] with T = B*S; cos/sin: [S, D/2] fp32
-  CHECK_BF16_CONTIG(x);
+  // x: [T, Hh, D] with T = B*S (strided views OK if the head dim is
+  // contiguous — e.g. qkv-split slices / transposed grads); output packed.
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.dim() == 3, "rope expects [T, Hh, D]");
+  TORCH_CHECK(x.stride(2) == 1, "head dim must be contiguous");
   TORCH_CHECK(cost.scalar_type() == at::kFloat && cost.is_contiguous());
   const int D = (int)x.size(2);
   const int Hh = (int)x.size(1);
   TORCH_CHECK(D % 8 == 0, "head dim must be a multiple of 8");
   TORCH_CHECK(cost.size(0) >= S && cost.size(1) == D / 2, "cos table shape");
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
-  auto o = at::empty_like(x);
+  auto o = at::empty(x.sizes(), x.options());
   const long total_quads = (long)x.size(0) * Hh * (D / 8);
   kt_rope(x.data_ptr(), o.data_ptr(), cost.data_ptr(), sint.data_ptr(),
-          total_quads, (int)S, Hh, D, (float)sign, cur_stream(x));
+          total_quads, (int)S, Hh, D, x.stride(0), x.stride(1), (float)sign,
+          cur_stream(x));
   return o;
 }
 

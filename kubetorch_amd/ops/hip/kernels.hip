@@ -211,7 +211,10 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
                             const float* __restrict__ cost,
                             const float* __restrict__ sint,
                             long total_quads, int S, int Hh, int D,
+                            long src_t_stride, long src_h_stride,
                             float sign) {
+  // x is read with explicit (token, head) strides so qkv-split views and
+  // transposed gradients need no .contiguous() copy; o is written packed.
   const int half = D >> 1;
   const int quads_per_head = D >> 3;  // 4 pairs per quad
   const long quads_per_tok = (long)Hh * quads_per_head;
@@ -223,8 +226,10 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
     const int qi = rem - h * quads_per_head;
     const int pos = (int)(tok % S);
     const size_t base = ((size_t)tok * Hh + h) * D + qi * 4;
-    vec4u x1 = *reinterpret_cast<const vec4u*>(x + base);
-    vec4u x2 = *reinterpret_cast<const vec4u*>(x + base + half);
+    const size_t sbase = (size_t)tok * src_t_stride +
+                         (size_t)h * src_h_stride + qi * 4;
+    vec4u x1 = *reinterpret_cast<const vec4u*>(x + sbase);
+    vec4u x2 = *reinterpret_cast<const vec4u*>(x + sbase + half);
     vec4f c = *reinterpret_cast<const vec4f*>(cost + (size_t)pos * half + qi * 4);
     vec4f s = *reinterpret_cast<const vec4f*>(sint + (size_t)pos * half + qi * 4);
     vec4u o1, o2;
@@ -473,12 +478,13 @@ void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
 }
 
 void kt_rope(const void* x, void* o, const void* cost, const void* sint,
-             long total_quads, int S, int Hh, int D, float sign,
-             hipStream_t stream) {
+             long total_quads, int S, int Hh, int D, long src_t_stride,
+             long src_h_stride, float sign, hipStream_t stream) {
   int grid = grid_for(total_quads, 256);
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(256), 0, stream,
                      (const u16*)x, (u16*)o, (const float*)cost,
-                     (const float*)sint, total_quads, S, Hh, D, sign);
+                     (const float*)sint, total_quads, S, Hh, D,
+                     src_t_stride, src_h_stride, sign);
 }
 
 void kt_swiglu_fwd(const void* gu, void* out, long N, int I,
