@@ -76,14 +76,27 @@ class Attention(nn.Module):
         q, k, v = qkv.split(
             [self.n_heads * hd, self.n_kv * hd, self.n_kv * hd], dim=-1
         )
-        q = ops.rope(q.view(B, S, self.n_heads, hd), cos, sin)
+        attn_impl = os.environ.get("KT_ATTN", "ck")
+        # The v3 FMHA kernel wants Q pre-scaled by softmax_scale*log2e (its
+        # LSE contract, ops/hip/bindings.cpp) — fold that into the RoPE
+        # kernel's fp32 output scaling so it costs nothing. Decide BEFORE
+        # RoPE runs.
+        use_v3 = (attn_impl == "ck"
+                  and ops.flash_attention_v3_supported(S, hd, x.device, x.dtype))
+        q_oscale = (hd ** -0.5) * ops.LOG2E if use_v3 else 1.0
+        q = ops.rope(q.view(B, S, self.n_heads, hd), cos, sin, oscale=q_oscale)
         k = ops.rope(k.view(B, S, self.n_kv, hd), cos, sin)
         v = v.view(B, S, self.n_kv, hd)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))
-        attn_impl = os.environ.get("KT_ATTN", "ck")
+        if use_v3:
+            # Fastest path: AITER-schedule CK v3 fwd (723 TF on the Llama
+            # shape) + AITER asm bwd, zero transpose copies, zero extra
+            # elementwise passes (scale folded into RoPE above).
+            o = ops.flash_attention(q, k, v, impl="v3")
+            return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         if (attn_impl in ("ck", "custom")
                 and ops.flash_attention_supported(q, k, v, True)):
-            # DEFAULT: CK-tile FMHA fwd + AITER asm bwd (KT_ATTN=torch for
+            # CK-tile FMHA fwd + AITER asm bwd (KT_ATTN=torch for
             # the SDPA path, KT_ATTN=custom for the in-tree rocWMMA kernel).
             # The CK path is stride-aware: the permuted [B,S,H,D] views go
             # in directly and O comes back in the same layout, so the whole

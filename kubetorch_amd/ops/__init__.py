@@ -171,44 +171,51 @@ def precompute_rope(seq_len, head_dim, base=500000.0, device=None):
     return freqs.cos().contiguous(), freqs.sin().contiguous()
 
 
-def _rope_ref(x, cos, sin, sign):
+def _rope_ref(x, cos, sin, sign, oscale=1.0):
     # x: [B, S, Hh, D]
     B, S, Hh, D = x.shape
     xf = x.float()
     x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
     c = cos[:S].view(1, S, 1, D // 2)
     s = sin[:S].view(1, S, 1, D // 2) * sign
-    o1 = x1 * c - x2 * s
-    o2 = x2 * c + x1 * s
+    o1 = (x1 * c - x2 * s) * oscale
+    o2 = (x2 * c + x1 * s) * oscale
     return torch.cat([o1, o2], dim=-1).to(x.dtype)
 
 
 class _Rope(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, cos, sin):
+    def forward(ctx, x, cos, sin, oscale):
         ctx.S = x.shape[1]
+        ctx.oscale = oscale
         ctx.save_for_backward(cos, sin)
         if x.is_cuda:
             B, S, Hh, D = x.shape
             xv = x.reshape(B * S, Hh, D)  # strided views pass through
-            return _ext().rope(xv, cos, sin, S, 1.0).view(x.shape)
-        return _rope_ref(x, cos, sin, 1.0)
+            return _ext().rope(xv, cos, sin, S, 1.0, oscale).view(x.shape)
+        return _rope_ref(x, cos, sin, 1.0, oscale)
 
     @staticmethod
     def backward(ctx, dy):
+        # RoPE is linear: y = oscale * R(x)  =>  dx = oscale * R^T(dy), so
+        # the same oscale rides along for free in the backward kernel.
         cos, sin = ctx.saved_tensors
         if dy.is_cuda:
             B, S, Hh, D = dy.shape
             dyv = dy.reshape(B * S, Hh, D)
-            dx = _ext().rope(dyv, cos, sin, S, -1.0).view(dy.shape)
+            dx = _ext().rope(dyv, cos, sin, S, -1.0, ctx.oscale).view(dy.shape)
         else:
-            dx = _rope_ref(dy, cos, sin, -1.0)
-        return dx, None, None
+            dx = _rope_ref(dy, cos, sin, -1.0, ctx.oscale)
+        return dx, None, None, None
 
 
-def rope(x, cos, sin):
-    """Apply rotate-half RoPE to x: [B, S, Hh, D] using [S, D/2] tables."""
-    return _Rope.apply(x, cos, sin)
+def rope(x, cos, sin, oscale=1.0):
+    """Apply rotate-half RoPE to x: [B, S, Hh, D] using [S, D/2] tables.
+
+    oscale multiplies the output in fp32 before the bf16 round (free in the
+    memory-bound kernel). Used to fold the attention softmax scale * log2e
+    into Q for the v3 FMHA path (flash_attention impl="v3")."""
+    return _Rope.apply(x, cos, sin, oscale)
 
 
 # ---------------------------------------------------------------------------
@@ -292,10 +299,21 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
 # ---------------------------------------------------------------------------
 # Flash attention: custom gfx950 forward + torch (AITER asm) backward
 # ---------------------------------------------------------------------------
+_LN2 = 0.6931471805599453
+LOG2E = 1.4426950408889634
+
+
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, impl):
-        if impl == "ck":
+        if impl == "v3":
+            # AITER-schedule CK v3 kernel (fastest fwd, profiles/). Contract:
+            # q arrives PRE-SCALED by scale*log2e (folded into the RoPE
+            # kernel's oscale), so softmax(scale*q0@kT) == softmax(ln2*q@kT)
+            # and the effective scale for fwd LSE and backward is ln2.
+            o, lse = _ext().attn_fwd_v3(q, k, v, scale, prescaled=True)
+            scale = _LN2
+        elif impl == "ck":
             # CK path is stride-aware: permuted [B,S,H,D] views go in as-is
             o, lse = _ext().attn_fwd_ck_tr(q, k, v, scale)
         else:
@@ -331,7 +349,9 @@ class _FlashAttention(torch.autograd.Function):
 def flash_attention(q, k, v, scale=None, impl="ck"):
     """Causal GQA attention fwd, bf16, D=128, layout [B, H, S, D], paired
     with torch's AITER asm backward. impl="ck" (default) runs the CK-tile
-    FMHA instantiation (1.56x AOTriton on the Llama shape — profiles/);
+    FMHA tr-load instantiation (1.8x AOTriton on the Llama shape —
+    profiles/); impl="v3" runs the AITER-schedule v3 kernel (2.0x) and
+    REQUIRES q pre-scaled by scale*log2e (fold it into ops.rope's oscale);
     impl="wmma" runs the in-tree rocWMMA kernel (ops/hip/attention.hip,
     requires S % 128 == 0)."""
     if scale is None:
@@ -343,6 +363,14 @@ def flash_attention_supported(q, k, v, is_causal):
     return (is_causal and q.is_cuda and q.dtype == torch.bfloat16
             and q.shape[-1] == 128 and q.shape[2] % 128 == 0
             and q.shape[2] >= 256 and hip_available())
+
+
+def flash_attention_v3_supported(S, head_dim, device, dtype):
+    """Shape gate for the v3 (prescaled-Q) path, checkable BEFORE RoPE so
+    the softmax scale can be folded into the RoPE output."""
+    return (head_dim == 128 and S % 256 == 0 and S >= 256
+            and device.type == "cuda" and dtype == torch.bfloat16
+            and hip_available())
 
 
 # ---------------------------------------------------------------------------

@@ -12,6 +12,7 @@
 #include <ck_tile/host/kernel_launch.hpp>
 #include <ck_tile/ops/epilogue.hpp>
 #include <ck_tile/ops/fmha.hpp>
+#include <ck_tile/ops/fmha_fwd_v3_impl.hpp>
 
 namespace {
 
@@ -100,7 +101,61 @@ void run_fmha(const void* q, const void* k, const void* v, void* o,
                                   K{}, grid, K::kBlockSize, 0, kargs));
 }
 
+// --- fmha v3 (the AITER schedule: 8 warps, M0=256, bf16) with LSE -------
+// the public fmha_fwd_v3() API hard-codes lse_ptr = nullptr; the kernel
+// itself supports kStoreLSE, so instantiate it directly for training.
+using V3Shape = ck_tile::TileFmhaShape<
+    ck_tile::sequence<256, 32, 128, 128, 32, 128>,
+    ck_tile::sequence<8, 1, 1>, ck_tile::sequence<32, 32, 16>,
+    ck_tile::sequence<8, 1, 1>, ck_tile::sequence<32, 32, 16>, true>;
+using V3Traits = ck_tile::TileFmhaFwdV3Traits<true,  // kPadSeqLenQ
+                                              true,  // kPadSeqLenK
+                                              false, // kPadHeadDimQ
+                                              false, // kPadHeadDimV
+                                              true,  // kStoreLSE (training)
+                                              -1>;
+using V3Mask = ck_tile::GenericAttentionMask<true, false>;
+using V3Problem = ck_tile::BlockFmhaFwdV3PipelineProblem<
+    qkv_t, qkv_t, qkv_t, float, float, float /* lse */, qkv_t /* P */,
+    float /* Oacc */, qkv_t /* O */, V3Shape, false /* varlen */, V3Mask,
+    V3Traits>;
+using V3Pipeline = ck_tile::BlockFmhaFwdV3Pipeline<V3Problem>;
+using V3Epilogue = ck_tile::Default2DEpilogue<
+    ck_tile::Default2DEpilogueProblem<float, qkv_t, true, true, true>>;
+using V3Kernel = ck_tile::FmhaFwdV3Kernel<V3Pipeline, V3Epilogue>;
+
 }  // namespace
+
+extern "C" void kt_attn_fwd_v3(const void* q, const void* k, const void* v,
+                               void* o, void* lse, int B, int Hq, int Hkv,
+                               int S, float scale, const long* strides,
+                               hipStream_t stream) {
+  FmhaStrides st;
+  memcpy(&st, strides, sizeof(st));
+  const ck_tile::index_t D = 128;
+  auto kargs = V3Kernel::MakeKargs(
+      q, k, v, lse, o, S, S, D, D, Hq, Hq / Hkv, scale,
+      (ck_tile::index_t)st.q_s, (ck_tile::index_t)st.k_s,
+      (ck_tile::index_t)st.v_s, (ck_tile::index_t)st.o_s,
+      (ck_tile::index_t)st.q_h, (ck_tile::index_t)st.k_h,
+      (ck_tile::index_t)st.v_h, S /* nhead_stride_lse */,
+      (ck_tile::index_t)st.o_h, (ck_tile::index_t)st.q_b,
+      (ck_tile::index_t)st.k_b, (ck_tile::index_t)st.v_b,
+      (ck_tile::index_t)Hq * S /* batch_stride_lse */,
+      (ck_tile::index_t)st.o_b, -1 /* window_left */, 0 /* window_right */,
+      (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
+      2 /* remap_opt (aiter default for this shape) */,
+      nullptr, nullptr);
+  dim3 grid = V3Kernel::GridSize(B, Hq, S, D);
+  constexpr dim3 blocks = V3Kernel::BlockSize();
+  ck_tile::stream_config cfg{};
+  cfg.stream_id_ = stream;
+  cfg.cold_niters_ = 0;
+  cfg.nrepeat_ = 1;
+  ck_tile::launch_kernel(cfg,
+                         ck_tile::make_kernel<V3Kernel::kBlockPerCu>(
+                             V3Kernel{}, grid, blocks, 0, kargs));
+}
 
 extern "C" void kt_attn_fwd_ck_tr(const void* q, const void* k,
                                   const void* v, void* o, void* lse, int B,

@@ -19,7 +19,7 @@ void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
                     hipStream_t stream);
 void kt_rope(const void* x, void* o, const void* cost, const void* sint,
              long total_quads, int S, int Hh, int D, long src_t_stride,
-             long src_h_stride, float sign, hipStream_t stream);
+             long src_h_stride, float sign, float oscale, hipStream_t stream);
 void kt_swiglu_fwd(const void* gu, void* out, long N, int I,
                    hipStream_t stream);
 void kt_swiglu_bwd(const void* dout, const void* gu, void* dgu, long N, int I,
@@ -39,6 +39,9 @@ void kt_attn_fwd_ck(const void* q, const void* k, const void* v, void* o,
 void kt_attn_fwd_ck_tr(const void* q, const void* k, const void* v, void* o,
                        void* lse, int B, int Hq, int Hkv, int S, float scale,
                        const long* strides, hipStream_t stream);
+void kt_attn_fwd_v3(const void* q, const void* k, const void* v, void* o,
+                    void* lse, int B, int Hq, int Hkv, int S, float scale,
+                    const long* strides, hipStream_t stream);
 }
 
 namespace {
@@ -106,7 +109,8 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
 }
 
 at::Tensor rope(const at::Tensor& x, const at::Tensor& cost,
-                const at::Tensor& sint, int64_t S, double sign) {
+                const at::Tensor& sint, int64_t S, double sign,
+                double oscale) {
   // x: [T, Hh, D] with T = B*S (strided views OK if the head dim is
   // contiguous — e.g. qkv-split slices / transposed grads); output packed.
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
@@ -122,7 +126,7 @@ at::Tensor rope(const at::Tensor& x, const at::Tensor& cost,
   const long total_quads = (long)x.size(0) * Hh * (D / 8);
   kt_rope(x.data_ptr(), o.data_ptr(), cost.data_ptr(), sint.data_ptr(),
           total_quads, (int)S, Hh, D, x.stride(0), x.stride(1), (float)sign,
-          cur_stream(x));
+          (float)oscale, cur_stream(x));
   return o;
 }
 
@@ -233,7 +237,7 @@ std::vector<at::Tensor> attn_fwd_ck_impl(const at::Tensor& q,
       v.stride(2), v.stride(1), v.stride(0),
       o.stride(2), o.stride(1), o.stride(0),
   };
-  auto fn = trload ? kt_attn_fwd_ck_tr : kt_attn_fwd_ck;
+  auto fn = trload ? kt_attn_fwd_ck_tr : kt_attn_fwd_ck;  // see also v3
   fn(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
      lse.data_ptr(), B, Hq, Hkv, S, (float)scale, strides, cur_stream(q));
   return {o, lse};
@@ -250,6 +254,40 @@ std::vector<at::Tensor> attn_fwd_ck_tr(const at::Tensor& q,
   return attn_fwd_ck_impl(q, k, v, scale, true);
 }
 
+std::vector<at::Tensor> attn_fwd_v3(const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& v, double scale,
+                                    bool prescaled) {
+  // AITER-schedule v3 kernel (8 warps, M0=256) with LSE output.
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128 && q.stride(3) == 1);
+  TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1);
+  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
+  const int Hkv = (int)k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
+  // LSE-correct scaling contract: the v3 pipeline tracks the rowmax `m` of
+  // the RAW gemm0 scores and its LSE epilogue computes m/log2e + log(l),
+  // which is only the natural-log LSE when the gemm0 scores are already in
+  // the scaled log2 domain. So Q must carry (scale * log2e) and the kernel
+  // gets scale_s = ln2 (MakeKargs multiplies by log2e -> effective in-kernel
+  // scale 1.0). O is mathematically unchanged; LSE becomes exact.
+  // prescaled=true: the caller already folded scale*log2e into Q (free via
+  // the RoPE kernel's oscale) -- skip the extra elementwise pass here.
+  auto qs = prescaled ? q : q.mul(scale * 1.4426950408889634);
+  auto o = at::empty_strided(q.sizes(), q.strides(), q.options());
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  const long strides[12] = {
+      qs.stride(2), qs.stride(1), qs.stride(0),
+      k.stride(2), k.stride(1), k.stride(0),
+      v.stride(2), v.stride(1), v.stride(0),
+      o.stride(2), o.stride(1), o.stride(0),
+  };
+  kt_attn_fwd_v3(qs.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                 lse.data_ptr(), B, Hq, Hkv, S, (float)0.6931471805599453,
+                 strides, cur_stream(q));
+  return {o, lse};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -259,9 +297,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "CK-tile FMHA fwd (bf16, D=128, causal, GQA) -> (o, lse)");
   mod.def("attn_fwd_ck_tr", &attn_fwd_ck_tr,
           "CK-tile FMHA fwd, gfx950 tr-load variant -> (o, lse)");
+  mod.def("attn_fwd_v3", &attn_fwd_v3,
+          "CK-tile FMHA v3 (AITER schedule) fwd with LSE -> (o, lse)",
+          pybind11::arg("q"), pybind11::arg("k"), pybind11::arg("v"),
+          pybind11::arg("scale"), pybind11::arg("prescaled") = false);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16)");
-  mod.def("rope", &rope, "RoPE rotate-half (bf16), sign=+1 fwd / -1 bwd");
+  mod.def("rope", &rope, "RoPE rotate-half (bf16), sign=+1 fwd / -1 bwd",
+          pybind11::arg("x"), pybind11::arg("cost"), pybind11::arg("sint"),
+          pybind11::arg("S"), pybind11::arg("sign"),
+          pybind11::arg("oscale") = 1.0);
   mod.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (bf16)");
   mod.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (bf16)");
   mod.def("cross_entropy_fwd_", &cross_entropy_fwd_,

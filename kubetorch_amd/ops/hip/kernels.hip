@@ -212,7 +212,10 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
                             const float* __restrict__ sint,
                             long total_quads, int S, int Hh, int D,
                             long src_t_stride, long src_h_stride,
-                            float sign) {
+                            float sign, float oscale) {
+  // oscale: free fp32 output scaling (applied before the bf16 round), used
+  // to fold the attention softmax scale * log2e into Q for the CK v3 FMHA
+  // kernel's scaled-log2-domain LSE contract (see bindings.cpp attn_fwd_v3).
   // x is read with explicit (token, head) strides so qkv-split views and
   // transposed gradients need no .contiguous() copy; o is written packed.
   const int half = D >> 1;
@@ -236,8 +239,8 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float a = bf2f(x1[j]), b = bf2f(x2[j]);
-      o1[j] = f2bf(a * c[j] - sign * b * s[j]);
-      o2[j] = f2bf(b * c[j] + sign * a * s[j]);
+      o1[j] = f2bf((a * c[j] - sign * b * s[j]) * oscale);
+      o2[j] = f2bf((b * c[j] + sign * a * s[j]) * oscale);
     }
     *reinterpret_cast<vec4u*>(o + base) = o1;
     *reinterpret_cast<vec4u*>(o + base + half) = o2;
@@ -479,12 +482,13 @@ void kt_rmsnorm_bwd(const void* dy, const void* ds, const void* x,
 
 void kt_rope(const void* x, void* o, const void* cost, const void* sint,
              long total_quads, int S, int Hh, int D, long src_t_stride,
-             long src_h_stride, float sign, hipStream_t stream) {
+             long src_h_stride, float sign, float oscale,
+             hipStream_t stream) {
   int grid = grid_for(total_quads, 256);
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(256), 0, stream,
                      (const u16*)x, (u16*)o, (const float*)cost,
                      (const float*)sint, total_quads, S, Hh, D,
-                     src_t_stride, src_h_stride, sign);
+                     src_t_stride, src_h_stride, sign, oscale);
 }
 
 void kt_swiglu_fwd(const void* gu, void* out, long N, int I,

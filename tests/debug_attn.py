@@ -34,10 +34,14 @@ def check(B, Hq, Hkv, S, tag, strided=False):
 
     o, lse = ops._ext().attn_fwd_ck(q, k, v, scale)
     o_tr, lse_tr = ops._ext().attn_fwd_ck_tr(q, k, v, scale)
+    o_v3, lse_v3 = ops._ext().attn_fwd_v3(q, k, v, scale)
     out_ref, lse_ref, seed, offset = aten_fwd(q, k_exp, v_exp, None, True,
                                               0.0, True, scale=scale)
     d = (o.float() - out_ref.float()).abs()
     dl = (lse - lse_ref).abs()
+    dv3 = (o_v3.float() - out_ref.float()).abs()
+    print(f"[{tag}] v3 fwd max={dv3.max().item():.4e} "
+          f"v3 lse max={(lse_v3 - lse_ref).abs().max().item():.4e}")
     dtr = (o_tr.float() - out_ref.float()).abs()
     print(f"[{tag}] tr fwd max={dtr.max().item():.4e} "
           f"tr lse max={(lse_tr - lse_ref).abs().max().item():.4e}")
@@ -102,12 +106,14 @@ def perf(B, Hq, Hkv, S, iters=20):
 
     t_mine = timeit(lambda: ops._ext().attn_fwd_ck(q, k, v, scale))
     t_tr = timeit(lambda: ops._ext().attn_fwd_ck_tr(q, k, v, scale))
+    t_v3 = timeit(lambda: ops._ext().attn_fwd_v3(q, k, v, scale))
     t_wmma = timeit(lambda: ops._ext().attn_fwd(q, k, v, scale))
     t_torch = timeit(lambda: aten_fwd(q, k_exp, v_exp, None, True, 0.0, True,
                                       scale=scale))
     flops = 4 * B * Hq * S * S * 128 * 0.5
     print(f"perf B{B} H{Hq}/{Hkv} S{S}: ck {flops/t_mine/1e12:.0f} TF  "
-          f"ck_tr {flops/t_tr/1e12:.0f} TF  wmma {flops/t_wmma/1e12:.0f} TF  "
+          f"ck_tr {flops/t_tr/1e12:.0f} TF  v3 {flops/t_v3/1e12:.0f} TF  "
+          f"wmma {flops/t_wmma/1e12:.0f} TF  "
           f"aten {flops/t_torch/1e12:.0f} TF")
 
 

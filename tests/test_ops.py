@@ -48,10 +48,11 @@ def ref_swiglu(gu):
     return F.silu(g) * u
 
 
-def _run_fwd_bwd(fn, *inputs):
+def _run_fwd_bwd(fn, *inputs, gout=None):
     ins = [t.detach().clone().requires_grad_(t.is_floating_point()) for t in inputs]
     out = fn(*ins)
-    gout = torch.randn_like(out)
+    if gout is None:
+        gout = torch.randn_like(out)
     out.backward(gout)
     grads = [t.grad for t in ins if t.is_floating_point()]
     return out.detach(), grads, gout
@@ -164,6 +165,27 @@ class TestRope:
     @pytest.mark.gpu
     def test_gpu(self):
         self._check("cuda")
+
+    def _check_oscale(self, device):
+        """rope(oscale=c) == c * rope(x) in fwd AND bwd (linearity)."""
+        torch.manual_seed(3)
+        B, S, H, D = 2, 32, 2, 64
+        c = 0.125  # exactly representable: no rounding ambiguity
+        cos, sin = ops.precompute_rope(S, D, base=10000.0, device=device)
+        x = torch.randn(B, S, H, D, dtype=BF16, device=device)
+        y, (dx,), gout = _run_fwd_bwd(
+            lambda xi: ops.rope(xi, cos, sin, oscale=c), x)
+        y0, (dx0,), _ = _run_fwd_bwd(
+            lambda xi: ops.rope(xi, cos, sin), x, gout=gout)
+        _assert_close(y, (y0.float() * c), msg="rope oscale fwd")
+        _assert_close(dx, (dx0.float() * c), msg="rope oscale bwd")
+
+    def test_oscale_cpu(self):
+        self._check_oscale("cpu")
+
+    @pytest.mark.gpu
+    def test_oscale_gpu(self):
+        self._check_oscale("cuda")
 
 
 class TestSwiGLU:
@@ -295,6 +317,54 @@ class TestFlashAttention:
         _assert_close(q.grad, qr.grad, msg="flash dq")
         _assert_close(k.grad, kr.grad, msg="flash dk")
         _assert_close(v.grad, vr.grad, msg="flash dv")
+
+    @pytest.mark.gpu
+    def test_v3_prescaled_fwd_bwd(self):
+        """impl="v3" contract: q pre-scaled by scale*log2e, effective scale
+        ln2; output, LSE and all grads must match the fp32 reference."""
+        torch.manual_seed(2)
+        B, Hq, Hkv, S, D = 2, 8, 2, 512, 128
+        scale = D ** -0.5
+        q0 = torch.randn(B, Hq, S, D, dtype=BF16, device="cuda")
+        k = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda",
+                        requires_grad=True)
+        v = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda",
+                        requires_grad=True)
+        # pre-scale in fp32 then round once, mirroring the RoPE oscale fold
+        q = (q0.float() * (scale * ops.LOG2E)).to(BF16).requires_grad_(True)
+        out = ops.flash_attention(q, k, v, impl="v3")
+        gout = torch.randn_like(out)
+        out.backward(gout)
+
+        g = Hq // Hkv
+        qr = q.detach().float().requires_grad_(True)
+        kr = k.detach().float().requires_grad_(True)
+        vr = v.detach().float().requires_grad_(True)
+        outr = F.scaled_dot_product_attention(
+            qr, kr.repeat_interleave(g, 1), vr.repeat_interleave(g, 1),
+            is_causal=True, scale=ops._LN2)
+        outr.backward(gout.float())
+        _assert_close(out, outr, msg="v3 fwd")
+
+        # Grad magnitudes here are ~8x larger than at the usual softmax scale
+        # (q carries scale*log2e, so dq is w.r.t. the prescaled q), so use a
+        # magnitude-aware atol. The bf16 aten EFFICIENT path itself shows the
+        # same absolute error vs fp32 at this scale (verified on-box).
+        def close_rel(a, b, msg):
+            atol = max(2e-2, 6e-3 * b.float().abs().max().item())
+            _assert_close(a, b, atol=atol, msg=msg)
+
+        close_rel(q.grad, qr.grad, "v3 dq")
+        close_rel(k.grad, kr.grad, "v3 dk")
+        close_rel(v.grad, vr.grad, "v3 dv")
+        # and the LSE itself against torch.logsumexp
+        o3, lse3 = ops._ext().attn_fwd_v3(q.detach(), k.detach(), v.detach(),
+                                          scale, prescaled=True)
+        s = (q.detach().float() @ k.detach().repeat_interleave(g, 1)
+             .float().transpose(-1, -2)) * ops._LN2
+        s = s + torch.full((S, S), float("-inf"), device="cuda").triu(1)
+        lse_ref = torch.logsumexp(s, dim=-1)
+        torch.testing.assert_close(lse3, lse_ref, rtol=1e-3, atol=1e-3)
 
     @pytest.mark.gpu
     def test_lse_matches_aten(self):
