@@ -32,6 +32,7 @@ _LAZY = {
     "get": ("kubetorch_amd.data_store.commands", "get"),
     "ls": ("kubetorch_amd.data_store.commands", "ls"),
     "rm": ("kubetorch_amd.data_store.commands", "rm"),
+    "get_broadcast": ("kubetorch_amd.data_store.commands", "get_broadcast"),
     "BroadcastWindow": ("kubetorch_amd.data_store.types", "BroadcastWindow"),
     "pod_ips": ("kubetorch_amd.serving.discovery", "pod_ips"),
     "ops": ("kubetorch_amd.ops", None),

@@ -132,6 +132,10 @@ def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
         from kubetorch_amd.data_store import gpu_store
 
         return gpu_store.get(key, dest, window=window)
+    if window is not None and _store_url() is not None:
+        # fs key + BroadcastWindow -> rolling tree broadcast (W pods share
+        # the download through completed peers instead of the store)
+        return get_broadcast(key, dest, fanout=window.fanout, timeout=timeout)
     url = _store_url()
     if url is None:
         srcp = _local_path(key)
@@ -162,6 +166,72 @@ def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
         with open(dest, "wb") as f:
             f.write(r.content)
     return dest
+
+
+def _fetch_from_peer(source, key, dest, timeout):
+    r = httpx.get(f"{source}/files/{key}", timeout=timeout)
+    r.raise_for_status()
+    dest = os.path.expanduser(dest or os.path.basename(key))
+    if r.headers.get("X-KT-Tar") == "1":
+        os.makedirs(dest, exist_ok=True)
+        with tarfile.open(fileobj=io.BytesIO(r.content), mode="r:gz") as tar:
+            tar.extractall(dest)  # noqa: S202 - trusted in-cluster peers
+    else:
+        os.makedirs(os.path.dirname(dest) or ".", exist_ok=True)
+        with open(dest, "wb") as f:
+            f.write(r.content)
+    return dest
+
+
+def get_broadcast(key, dest=None, serve=True, fanout=None, timeout=C.HTTP_TIMEOUT,
+                  poll=0.2, max_wait=600.0):
+    """Tree-broadcast get: W pods fetching the same key form a rolling tree
+    (store -> first `fanout` pods -> their children, exponential fan-out)
+    so the store serves O(fanout) downloads total instead of W. Reference
+    parity: DataStoreClient broadcast get + join_fs_broadcast.
+
+    serve=True registers this process as a source for later joiners (lazy
+    singleton BcastFileServer). Returns the local dest path."""
+    import time as _time
+
+    from kubetorch_amd.data_store import fileserve
+
+    url = _store_url()
+    if url is None:  # local-dir mode: nothing to offload
+        return get(key, dest, timeout=timeout)
+    if fanout is None:
+        fanout = int(os.environ.get("KT_FS_BCAST_FANOUT", "50"))
+    key = key.strip("/")
+    deadline = _time.monotonic() + max_wait
+    while True:
+        r = httpx.post(f"{url}/fsbcast/join",
+                       json={"key": key, "fanout": fanout}, timeout=timeout)
+        r.raise_for_status()
+        j = r.json()
+        if not j.get("wait"):
+            break
+        if _time.monotonic() > deadline:
+            raise TimeoutError(f"fs broadcast join timed out for {key}")
+        _time.sleep(poll)
+    parent = j["parent"]
+    try:
+        if j["source"] == "store":
+            out = get(key, dest, timeout=timeout)
+        else:
+            out = _fetch_from_peer(j["source"], key, dest, timeout)
+    except Exception:
+        # release the slot so the tree doesn't wedge on a failed child
+        httpx.post(f"{url}/fsbcast/complete",
+                   json={"key": key, "parent": parent}, timeout=timeout)
+        raise
+    serve_url = None
+    if serve:
+        fileserve.register_local(key, out)
+        serve_url = fileserve.ensure_server().url
+    httpx.post(f"{url}/fsbcast/complete",
+               json={"key": key, "parent": parent, "url": serve_url},
+               timeout=timeout)
+    return out
 
 
 def ls(prefix=""):

@@ -218,6 +218,70 @@ def broadcast_complete(group_id: str):
     return {"ok": True}
 
 
+_fsb = {}
+_fsb_lock = threading.Lock()
+
+
+@app.post("/fsbcast/join")
+async def fsbcast_join(request: Request):
+    """Filesystem tree-broadcast coordination (reference parity:
+    metadata-server join_fs_broadcast + pod-data-server fs-broadcast
+    tracking). Each joiner is assigned a source with spare capacity —
+    a pod that already completed the download (preferred, to offload the
+    store) or the store itself. fanout bounds concurrent children per
+    source. Returns {"source": url|"store", "parent": id} or
+    {"wait": true} when every source is saturated."""
+    body = await request.json()
+    key = body["key"].strip("/")
+    fanout = int(body.get("fanout", 50))
+    with _fsb_lock:
+        rec = _fsb.setdefault(key, {"sources": [], "active": {"store": 0},
+                                    "created": time.time()})
+        for url in rec["sources"]:  # completed pods first: offload the store
+            if rec["active"].get(url, 0) < fanout:
+                rec["active"][url] = rec["active"].get(url, 0) + 1
+                return {"source": url, "parent": url}
+        if rec["active"]["store"] < max(1, fanout):
+            rec["active"]["store"] += 1
+            return {"source": "store", "parent": "store"}
+        return {"wait": True}
+
+
+@app.post("/fsbcast/complete")
+async def fsbcast_complete(request: Request):
+    """A joiner finished downloading: release its parent's slot and (if it
+    advertises a serve url) register it as a source for later joiners."""
+    body = await request.json()
+    key = body["key"].strip("/")
+    with _fsb_lock:
+        rec = _fsb.get(key)
+        if rec is None:
+            return {"ok": True}
+        parent = body.get("parent")
+        if parent in rec["active"] and rec["active"][parent] > 0:
+            rec["active"][parent] -= 1
+        url = body.get("url")
+        if url and url not in rec["sources"]:
+            rec["sources"].append(url)
+    return {"ok": True}
+
+
+@app.get("/fsbcast/status")
+def fsbcast_status(key: str):
+    with _fsb_lock:
+        rec = _fsb.get(key.strip("/"))
+    if rec is None:
+        return JSONResponse({"error": "not found"}, status_code=404)
+    return rec
+
+
+@app.delete("/fsbcast/{key:path}")
+def fsbcast_clear(key: str):
+    with _fsb_lock:
+        _fsb.pop(key.strip("/"), None)
+    return {"ok": True}
+
+
 @app.post("/logs/push")
 async def logs_push(request: Request):
     body = await request.json()

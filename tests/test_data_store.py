@@ -337,3 +337,150 @@ def test_daemon_failure_auto_restart(tmp_path, monkeypatch):
     cli.register("recovered", t)
     r = cli.request({"cmd": "list"})
     assert "recovered" in r["keys"]
+
+
+# ---------------------------------------------------------------------------
+# Filesystem tree broadcast (reference: join_fs_broadcast rolling tree)
+# ---------------------------------------------------------------------------
+@pytest.fixture()
+def http_store(tmp_store):
+    """A live HTTP store service with KT_STORE_URL exported."""
+    import threading
+
+    import httpx
+    import uvicorn
+
+    from kubetorch_amd.data_store import server as store_server
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    store_server.DATA_ROOT = str(tmp_store / "fsbroot")
+    os.makedirs(store_server.DATA_ROOT, exist_ok=True)
+    config = uvicorn.Config(store_server.app, host="127.0.0.1", port=port,
+                            log_level="error")
+    server = uvicorn.Server(config)
+    threading.Thread(target=server.run, daemon=True).start()
+    url = f"http://127.0.0.1:{port}"
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        try:
+            if httpx.get(url + "/health", timeout=1).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.05)
+    old = os.environ.get("KT_STORE_URL")
+    os.environ["KT_STORE_URL"] = url
+    yield url
+    server.should_exit = True
+    if old is None:
+        os.environ.pop("KT_STORE_URL", None)
+    else:
+        os.environ["KT_STORE_URL"] = old
+
+
+def test_fs_broadcast_tree_chain(http_store, tmp_store):
+    """fanout=1: the 2nd and 3rd joiners must be fed by completed PEERS,
+    not the store (store egress stays O(1))."""
+    import httpx
+
+    from kubetorch_amd.data_store import commands as ds
+    from kubetorch_amd.data_store import fileserve
+
+    src = tmp_store / "bsrc"
+    src.mkdir()
+    (src / "w.bin").write_bytes(b"x" * 4096)
+    (src / "cfg.json").write_text("{}")
+    ds.put("ns/bcast1", src=str(src))
+
+    # joiner A: store is the only source
+    a_dest = str(tmp_store / "a")
+    out_a = ds.get_broadcast("ns/bcast1", a_dest, fanout=1)
+    assert (tmp_store / "a" / "w.bin").read_bytes() == b"x" * 4096
+    st = httpx.get(http_store + "/fsbcast/status",
+                   params={"key": "ns/bcast1"}).json()
+    assert len(st["sources"]) == 1  # A registered as a source
+    a_url = st["sources"][0]
+
+    # joiner B: must be assigned peer A (peers preferred over the store)
+    r = httpx.post(http_store + "/fsbcast/join",
+                   json={"key": "ns/bcast1", "fanout": 1})
+    assert r.json()["source"] == a_url
+    # while B is mid-download, A is saturated (fanout=1) and the store has
+    # capacity 1 -> C gets the store; D must wait
+    r2 = httpx.post(http_store + "/fsbcast/join",
+                    json={"key": "ns/bcast1", "fanout": 1})
+    assert r2.json()["source"] == "store"
+    r3 = httpx.post(http_store + "/fsbcast/join",
+                    json={"key": "ns/bcast1", "fanout": 1})
+    assert r3.json() == {"wait": True}
+    # B completes -> A has a free slot again
+    httpx.post(http_store + "/fsbcast/complete",
+               json={"key": "ns/bcast1", "parent": a_url})
+    r4 = httpx.post(http_store + "/fsbcast/join",
+                    json={"key": "ns/bcast1", "fanout": 1})
+    assert r4.json()["source"] == a_url
+
+    # the peer actually serves the bytes (fetch from A's BcastFileServer)
+    b_dest = str(tmp_store / "b")
+    ds._fetch_from_peer(a_url, "ns/bcast1", b_dest, 30)
+    assert (tmp_store / "b" / "w.bin").read_bytes() == b"x" * 4096
+    assert (tmp_store / "b" / "cfg.json").read_text() == "{}"
+    httpx.delete(http_store + "/fsbcast/ns/bcast1")
+    srv = fileserve._server
+    if srv is not None:
+        srv.close()
+        fileserve._server = None
+
+
+def test_fs_broadcast_concurrent(http_store, tmp_store):
+    """8 concurrent getters, fanout=2: everyone converges with correct
+    bytes and the store served at most 2 downloads directly."""
+    import threading
+
+    import httpx
+
+    from kubetorch_amd.data_store import commands as ds
+    from kubetorch_amd.data_store import fileserve
+
+    src = tmp_store / "csrc"
+    src.mkdir()
+    (src / "w.bin").write_bytes(b"y" * 10000)
+    ds.put("ns/bcast2", src=str(src))
+
+    store_hits = []
+    orig_get = ds.get
+
+    def counting_get(key, dest=None, **kw):
+        store_hits.append(dest)
+        return orig_get(key, dest, **kw)
+
+    ds.get = counting_get
+    try:
+        errs = []
+
+        def worker(i):
+            try:
+                d = str(tmp_store / f"w{i}")
+                ds.get_broadcast("ns/bcast2", d, fanout=2)
+                assert (tmp_store / f"w{i}" / "w.bin").read_bytes() == b"y" * 10000
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        threads = [threading.Thread(target=worker, args=(i,)) for i in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(60)
+        assert not errs, errs
+        # store handled at most its fanout's worth of direct downloads
+        # (every later joiner was fed by a completed peer)
+        assert len(store_hits) <= 4, store_hits
+    finally:
+        ds.get = orig_get
+        httpx.delete(http_store + "/fsbcast/ns/bcast2")
+        srv = fileserve._server
+        if srv is not None:
+            srv.close()
+            fileserve._server = None
