@@ -127,18 +127,40 @@ class Module:
         return self._http
 
     def _wait_ready(self, timeout=C.LAUNCH_TIMEOUT, reloaded=False):
-        """Poll /ready?launch_id until the pod finished loading this deploy.
-        If pods were hot-reloaded through the controller ack barrier, they are
-        ready by construction, but poll once to verify."""
+        """Poll /ready?launch_id until the pod finished loading this deploy,
+        streaming service events (pod scheduled/started/probe failures —
+        reference parity: K8s launch-event streaming) while waiting. If pods
+        were hot-reloaded through the controller ack barrier, they are ready
+        by construction, but poll once to verify."""
         deadline = time.time() + timeout
         delay = 0.05
+        ev_since = 0.0
+        last_ev_poll = 0.0
         while time.time() < deadline:
             if self.http.is_ready(launch_id=None if not reloaded else self.launch_id):
                 # accept pods that don't carry a launch_id (direct-env launch)
                 return
+            if self.stream_logs and time.time() - last_ev_poll > 1.0:
+                last_ev_poll = time.time()
+                ev_since = self._print_events(ev_since)
             time.sleep(delay)
             delay = min(delay * 1.5, 2.0)
+        self._print_events(ev_since)  # surface the failure reason
         raise LaunchError(f"service {self.name} not ready after {timeout}s")
+
+    def _print_events(self, since):
+        try:
+            evs = controller_client().service_events(
+                self.name, self.namespace, since=since)
+        except Exception:
+            return since
+        for e in evs:
+            tag = "!" if e.get("type") == "Warning" else "·"
+            pod = f" {e['pod']}" if e.get("pod") else ""
+            print(f"[kt] {tag}{pod} {e.get('reason', '')}: "
+                  f"{e.get('message', '')}")
+            since = max(since, e.get("ts", since))
+        return since
 
     # -- lifecycle ---------------------------------------------------------------
     def teardown(self):

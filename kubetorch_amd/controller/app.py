@@ -226,6 +226,18 @@ async def delete_workload(ns: str, name: str):
     return {"ok": True}
 
 
+@app.get("/controller/events/{ns}/{name}")
+def service_events(ns: str, name: str, since: float = 0.0):
+    """Pod lifecycle / K8s events for a service, polled by clients during
+    `.to()` launches (reference parity: launch-time K8s event streaming)."""
+    if not hasattr(HUB.driver, "get_events"):
+        return {"events": []}
+    try:
+        return {"events": HUB.driver.get_events(name, ns, since=since)}
+    except Exception as e:
+        return {"events": [], "error": str(e)}
+
+
 @app.post("/controller/pods/stream")
 async def pod_stream(request: Request):
     """Pod registration; the response is an endless NDJSON stream of pushes.

@@ -55,6 +55,33 @@ class LocalDriver:
     def __init__(self, controller_url=None):
         self.controller_url = controller_url
         self.services = {}  # (ns, name) -> [LocalPod]
+        self.events = {}    # (ns, name) -> [event dict] (launch streaming)
+        self._event_state = {}  # pod name -> last seen liveness
+
+    def _event(self, namespace, name, reason, message, pod=None):
+        self.events.setdefault((namespace, name), []).append({
+            "ts": time.time(), "type": "Normal", "reason": reason,
+            "message": message, "pod": pod,
+        })
+
+    def get_events(self, name, namespace, since=0.0):
+        """Pod lifecycle events for a service (reference parity: K8s events
+        streamed by the client during launch). Lazily emits exit events for
+        pods that died since the last poll."""
+        key = (namespace, name)
+        for p in self.services.get(key, []):
+            alive = p.alive()
+            was = self._event_state.get(p.name)
+            if was is None:
+                self._event_state[p.name] = alive
+            elif was and not alive:
+                self._event_state[p.name] = False
+                ev = self.events.setdefault(key, [])
+                ev.append({"ts": time.time(), "type": "Warning",
+                           "reason": "BackOff",
+                           "message": f"pod exited rc={p.proc.returncode}",
+                           "pod": p.name})
+        return [e for e in self.events.get(key, []) if e["ts"] > since]
 
     def apply(self, manifest, namespace, metadata=None, launch_id=None):
         name = manifest["metadata"]["name"]
@@ -63,6 +90,8 @@ class LocalDriver:
         pods = self.services.get(key, [])
         alive = [p for p in pods if p.alive()]
         if len(alive) == replicas:
+            self._event(namespace, name, "Reloaded",
+                        f"hot reload into {len(alive)} warm pod(s)")
             return [p.host for p in alive]  # warm pods: reload only
         for p in pods:
             p.kill()
@@ -119,12 +148,18 @@ class LocalDriver:
                 stderr=subprocess.DEVNULL,
             )
             new_pods.append(LocalPod(f"{name}-{i}", port, proc))
+            self._event(namespace, name, "Scheduled",
+                        f"assigned 127.0.0.1:{port}", pod=f"{name}-{i}")
+            self._event(namespace, name, "Started",
+                        "container started", pod=f"{name}-{i}")
+            self._event_state[f"{name}-{i}"] = True
         self.services[key] = new_pods
         return [p.host for p in new_pods]
 
     def delete(self, name, namespace):
         for p in self.services.pop((namespace, name), []):
             p.kill()
+        self.events.pop((namespace, name), None)
 
     def pods(self, name, namespace):
         return [p.host for p in self.services.get((namespace, name), [])
@@ -173,3 +208,33 @@ class K8sDriver:
             if ip:
                 hosts.append(f"{ip}:{C.SERVER_PORT}")
         return hosts
+
+    def get_events(self, name, namespace, since=0.0):
+        """K8s events for the service's objects (scheduling, image pulls,
+        probe failures) — what the client streams during launch."""
+        import datetime
+
+        out = subprocess.run(
+            [self.kubectl, "-n", namespace, "get", "events",
+             "--sort-by=.lastTimestamp", "-o", "json"],
+            capture_output=True,
+        )
+        if out.returncode != 0:
+            return []
+        evs = []
+        for it in json.loads(out.stdout).get("items", []):
+            obj = it.get("involvedObject", {}).get("name", "")
+            if not obj.startswith(name):
+                continue
+            raw_ts = (it.get("lastTimestamp") or it.get("eventTime") or "")
+            try:
+                ts = datetime.datetime.fromisoformat(
+                    raw_ts.replace("Z", "+00:00")).timestamp()
+            except ValueError:
+                ts = 0.0
+            if ts <= since:
+                continue
+            evs.append({"ts": ts, "type": it.get("type", "Normal"),
+                        "reason": it.get("reason", ""),
+                        "message": it.get("message", ""), "pod": obj})
+        return evs

@@ -86,6 +86,15 @@ class ControllerClient:
             "DELETE", f"/controller/workload/{namespace}/{name}"
         ).json()
 
+    def service_events(self, name, namespace, since=0.0):
+        """Launch/lifecycle events for a service's pods (streamed by
+        Module.to while waiting for readiness)."""
+        r = self._request(
+            "GET", f"/controller/events/{namespace}/{name}",
+            params={"since": since},
+        )
+        return r.json().get("events", [])
+
 
 _controller = None
 _lock = threading.Lock()

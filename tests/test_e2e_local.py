@@ -185,3 +185,20 @@ def test_inactivity_ttl_reaper():
     finally:
         os.environ.pop("KT_TTL_REAPER_INTERVAL", None)
         os.environ.pop("KT_TTL_GRACE", None)
+
+
+def test_launch_event_streaming(remote_fn):
+    """Service events (Scheduled/Started) are queryable during/after launch
+    and pod deaths surface as Warning events (reference: K8s launch-event
+    streaming in Module.to)."""
+    from kubetorch_amd.globals import controller_client
+
+    evs = controller_client().service_events(remote_fn.name,
+                                             remote_fn.namespace)
+    reasons = [e["reason"] for e in evs]
+    assert "Scheduled" in reasons and "Started" in reasons, evs
+    assert any(e.get("pod") for e in evs)
+    # incremental polling: since=last ts returns nothing new
+    last = max(e["ts"] for e in evs)
+    assert controller_client().service_events(
+        remote_fn.name, remote_fn.namespace, since=last) == []
