@@ -1,4 +1,5 @@
 from kubetorch_amd.models.llama import (  # noqa: F401
+    KVCache,
     Llama,
     LlamaConfig,
     llama3_8b,

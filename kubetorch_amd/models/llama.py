@@ -48,6 +48,34 @@ def llama_tiny(**overrides) -> LlamaConfig:
     return LlamaConfig(**d)
 
 
+class KVCache:
+    """Per-layer KV cache for autoregressive decode: preallocated
+    [B, n_kv, max_len, head_dim] buffers appended in place (no reallocation
+    per step, sized for 288 GB HBM3E budgets at batch)."""
+
+    def __init__(self, cfg: LlamaConfig, batch, max_len, device, dtype):
+        shape = (batch, cfg.n_kv_heads, max_len, cfg.head_dim)
+        self.k = [torch.zeros(shape, device=device, dtype=dtype)
+                  for _ in range(cfg.n_layers)]
+        self.v = [torch.zeros(shape, device=device, dtype=dtype)
+                  for _ in range(cfg.n_layers)]
+        self.pos = 0
+        self.max_len = max_len
+
+    def append(self, layer, k, v):
+        # k, v: [B, n_kv, S, hd] for the current chunk
+        S = k.shape[2]
+        self.k[layer][:, :, self.pos:self.pos + S] = k
+        self.v[layer][:, :, self.pos:self.pos + S] = v
+        return (self.k[layer][:, :, :self.pos + S],
+                self.v[layer][:, :, :self.pos + S])
+
+    def advance(self, S):
+        self.pos += S
+        if self.pos > self.max_len:
+            raise RuntimeError("KV cache overflow")
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim, eps):
         super().__init__()
@@ -67,7 +95,18 @@ class Attention(nn.Module):
         self.wqkv = nn.Linear(d, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False)
         self.wo = nn.Linear(cfg.n_heads * hd, d, bias=False)
 
-    def forward(self, x, cos, sin):
+    def _sdpa(self, q, k, v, causal):
+        if self.n_kv != self.n_heads:
+            try:
+                return F.scaled_dot_product_attention(
+                    q, k, v, is_causal=causal, enable_gqa=True)
+            except (TypeError, RuntimeError):
+                rep = self.n_heads // self.n_kv
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+        return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+
+    def forward(self, x, cos, sin, cache=None, layer=0):
         import os
 
         B, S, _ = x.shape
@@ -76,6 +115,21 @@ class Attention(nn.Module):
         q, k, v = qkv.split(
             [self.n_heads * hd, self.n_kv * hd, self.n_kv * hd], dim=-1
         )
+        if cache is not None:
+            # Decode path: cos/sin come pre-sliced at the cache position;
+            # prefill (pos==0) is causal over the chunk, decode steps (S==1)
+            # attend over the whole cache.
+            q = ops.rope(q.view(B, S, self.n_heads, hd), cos, sin)
+            k = ops.rope(k.view(B, S, self.n_kv, hd), cos, sin)
+            v = v.view(B, S, self.n_kv, hd)
+            q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+            kf, vf = cache.append(layer, k, v)
+            if cache.pos == 0:
+                o = self._sdpa(q, kf, vf, causal=True)
+            else:
+                assert S == 1, "chunked decode after prefill not supported"
+                o = self._sdpa(q, kf, vf, causal=False)
+            return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         attn_impl = os.environ.get("KT_ATTN", "ck")
         # The v3 FMHA kernel wants Q pre-scaled by softmax_scale*log2e (its
         # LSE contract, ops/hip/bindings.cpp) — fold that into the RoPE
@@ -104,16 +158,7 @@ class Attention(nn.Module):
             o = ops.flash_attention(
                 q, k, v, impl="wmma" if attn_impl == "custom" else "ck")
             return self.wo(o.transpose(1, 2).reshape(B, S, -1))
-        if self.n_kv != self.n_heads:
-            try:
-                o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
-            except (TypeError, RuntimeError):
-                rep = self.n_heads // self.n_kv
-                k = k.repeat_interleave(rep, dim=1)
-                v = v.repeat_interleave(rep, dim=1)
-                o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
-        else:
-            o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        o = self._sdpa(q, k, v, causal=True)
         o = o.transpose(1, 2).reshape(B, S, -1)
         return self.wo(o)
 
@@ -204,3 +249,51 @@ class Llama(nn.Module):
         """Forward + fused CE (logits buffer is consumed by the fused op)."""
         logits = self.forward(tokens)
         return ops.fused_cross_entropy(logits, targets)
+
+    def _forward_cached(self, tokens, cache):
+        """One forward chunk through the KV cache (prefill or decode step);
+        returns logits for the LAST position only."""
+        S = tokens.shape[1]
+        p = cache.pos
+        cos = self.rope_cos[p:p + S]
+        sin = self.rope_sin[p:p + S]
+        h = self.embed(tokens)
+        eps = self.cfg.norm_eps
+        for i, layer in enumerate(self.layers):
+            h = h + layer.attn(ops.rmsnorm(h, layer.attn_norm.weight, eps),
+                               cos, sin, cache=cache, layer=i)
+            h = h + layer.mlp(ops.rmsnorm(h, layer.mlp_norm.weight, eps))
+        cache.advance(S)
+        h = ops.rmsnorm(h[:, -1:].contiguous(), self.norm.weight, eps)
+        return self.lm_head(h)[:, -1]
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens, temperature=0.0, top_k=None,
+                 stop_token=None, max_len=None):
+        """Autoregressive decode with a preallocated KV cache: one prefill
+        chunk, then single-token steps. temperature=0 is greedy argmax.
+        Returns [B, prompt+new] token ids. (The reference ships no model
+        code — this is the serving half of the flagship workload, the
+        training half being bench.py's DDP step.)"""
+        B, S0 = tokens.shape
+        dev = tokens.device
+        dtype = self.embed.weight.dtype
+        cache = KVCache(self.cfg, B, max_len or min(self.cfg.max_seq_len,
+                                                    S0 + max_new_tokens),
+                        dev, dtype)
+        out = [tokens]
+        logits = self._forward_cached(tokens, cache)
+        for _ in range(max_new_tokens):
+            if temperature > 0:
+                lg = logits.float() / temperature
+                if top_k:
+                    kth = lg.topk(top_k, dim=-1).values[:, -1:]
+                    lg = lg.masked_fill(lg < kth, float("-inf"))
+                nxt = torch.multinomial(torch.softmax(lg, -1), 1)
+            else:
+                nxt = logits.argmax(-1, keepdim=True)
+            out.append(nxt)
+            if stop_token is not None and (nxt == stop_token).all():
+                break
+            logits = self._forward_cached(nxt, cache)
+        return torch.cat(out, dim=1)
