@@ -143,3 +143,29 @@ def test_generate_sampling_and_stop():
     first = model.generate(prompt, max_new_tokens=1)[:, -1].item()
     out2 = model.generate(prompt, max_new_tokens=6, stop_token=first)
     assert out2.shape[1] == 6  # stopped right after the first new token
+
+
+@pytest.mark.gpu
+def test_generate_gpu_cache_logits_match():
+    """bf16 GPU decode: per-step cached logits vs full-recompute logits.
+    (argmax equality is not robust in bf16, so compare the logits.)"""
+    model, cfg = _tiny_model("cuda")
+    model.eval()
+    B, S0 = 2, 7
+    prompt = torch.randint(0, cfg.vocab_size, (B, S0), device="cuda")
+    from kubetorch_amd.models import KVCache
+
+    cache = KVCache(cfg, B, 32, torch.device("cuda"), torch.bfloat16)
+    with torch.no_grad():
+        lg_cache = model._forward_cached(prompt, cache)
+        lg_full = model(prompt)[:, -1]
+        torch.testing.assert_close(lg_cache.float(), lg_full.float(),
+                                   rtol=5e-2, atol=5e-1)
+        nxt = lg_full.argmax(-1, keepdim=True)
+        lg_cache2 = model._forward_cached(nxt, cache)
+        toks = torch.cat([prompt, nxt], dim=1)
+        lg_full2 = model(toks)[:, -1]
+        torch.testing.assert_close(lg_cache2.float(), lg_full2.float(),
+                                   rtol=5e-2, atol=5e-1)
+    out = model.generate(prompt, max_new_tokens=4)
+    assert out.shape == (B, S0 + 4) and out.is_cuda
