@@ -78,7 +78,7 @@ class FlatDDP:
 
     def __init__(self, model, lr=3e-4, betas=(0.9, 0.95), eps=1e-8,
                  weight_decay=0.1, bucket_mb=256, process_group=None,
-                 overlap_optimizer=True, grad_accum_steps=1):
+                 overlap_optimizer=True, grad_accum_steps=1, clip_norm=None):
         self.model = model
         self.lr = lr
         self.betas = betas
@@ -86,6 +86,15 @@ class FlatDDP:
         self.weight_decay = weight_decay
         self.pg = process_group
         self.step_count = 0
+        # Global-norm clipping needs the WHOLE grad norm before any param
+        # update, so it forces the deferred-optimizer path (bucket
+        # all-reduces still overlap backward; only the AdamW launch moves
+        # to step()). The clip itself is free: the scale factor rides the
+        # fused kernel's grad_scale argument — no extra pass over grads.
+        self.clip_norm = clip_norm
+        self.last_grad_norm = None
+        if clip_norm is not None:
+            overlap_optimizer = False
         self._world = (
             dist.get_world_size(process_group) if dist.is_initialized() else 1
         )
@@ -223,6 +232,18 @@ class FlatDDP:
                 if b.work is not None:
                     b.work.wait()
                     b.work = None
+            if self.clip_norm is not None:
+                # flat grads hold the SUM over ranks; the norm of the mean
+                # grad is ||g_sum|| / world. Identical on every rank (the
+                # grads are already reduced), so no extra collective.
+                total_sq = sum(
+                    (b.flat_grad.float() ** 2).sum() for b in self.buckets
+                )
+                norm = float(total_sq.sqrt()) * grad_scale
+                self.last_grad_norm = norm
+                if norm > self.clip_norm:
+                    grad_scale *= self.clip_norm / (norm + 1e-6)
+            for b in self.buckets:
                 ops.adamw_(
                     b.flat_param, b.flat_grad, b.m, b.v, self.lr,
                     self.betas[0], self.betas[1], self.eps, self.weight_decay,

@@ -106,3 +106,61 @@ def test_flatddp_two_ranks_match_single_process():
         eng._world = 1
     single = torch.cat([b.flat_param.float() for b in eng.buckets])
     torch.testing.assert_close(results[0], single, rtol=2e-2, atol=2e-2)
+
+
+def test_grad_clipping_scales_update():
+    """clip_norm: the engine must apply AdamW to clipped grads; with a huge
+    clip it must match the unclipped engine exactly."""
+    from kubetorch_amd.models import Llama, llama_tiny
+    from kubetorch_amd.parallel import FlatDDP
+
+    def build(clip):
+        torch.manual_seed(0)
+        m = Llama(llama_tiny())
+        return m, FlatDDP(m, lr=1e-3, bucket_mb=4, clip_norm=clip)
+
+    torch.manual_seed(1)
+    x = torch.randint(0, 512, (2, 32))
+    y = torch.randint(0, 512, (2, 32))
+
+    m1, e1 = build(clip=1e9)       # effectively unclipped
+    m2, e2 = build(clip=None)
+    m3, e3 = build(clip=1e-3)      # aggressive clip
+    for m, e in ((m1, e1), (m2, e2), (m3, e3)):
+        loss = m.loss(x, y)
+        loss.backward()
+        e.step()
+    # huge clip == no clip
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1, p2, rtol=0, atol=0)
+    # aggressive clip must actually change the update
+    diff = sum((p1 - p3).abs().sum().item()
+               for p1, p3 in zip(m1.parameters(), m3.parameters()))
+    assert diff > 0
+    # reported norm matches autograd's own global grad norm
+    torch.manual_seed(0)
+    m4 = Llama(llama_tiny())
+    loss = m4.loss(x, y)
+    loss.backward()
+    ref_norm = torch.norm(torch.stack(
+        [p.grad.float().norm() for p in m4.parameters()])).item()
+    assert abs(e3.last_grad_norm - ref_norm) / ref_norm < 5e-2, (
+        e3.last_grad_norm, ref_norm)
+
+
+def test_lr_schedules():
+    from kubetorch_amd.parallel import (constant_with_warmup, warmup_cosine,
+                                        warmup_linear)
+
+    # warmup ramps linearly, peak at base, decays to min
+    assert warmup_cosine(0, 1.0, 10, 100) == pytest.approx(0.1)
+    assert warmup_cosine(9, 1.0, 10, 100) == pytest.approx(1.0)
+    assert warmup_cosine(10, 1.0, 10, 100) == pytest.approx(1.0)
+    assert warmup_cosine(55, 1.0, 10, 100, min_lr=0.1) == pytest.approx(0.55)
+    assert warmup_cosine(100, 1.0, 10, 100, min_lr=0.1) == pytest.approx(0.1)
+    assert warmup_cosine(500, 1.0, 10, 100, min_lr=0.1) == pytest.approx(0.1)
+    assert warmup_linear(55, 1.0, 10, 100) == pytest.approx(0.5)
+    assert constant_with_warmup(999, 3e-4, 10) == pytest.approx(3e-4)
+    # monotone decay after warmup
+    vals = [warmup_cosine(s, 1.0, 10, 100) for s in range(10, 101)]
+    assert all(a >= b for a, b in zip(vals, vals[1:]))
