@@ -37,6 +37,9 @@ def main():
     # ~14% end-to-end (profiles/r01_sdpa_ab.md) -> default efficient.
     ap.add_argument("--sdpa", type=str, default="efficient",
                     choices=["flash", "efficient", "math"])
+    ap.add_argument("--ckpt", action="store_true",
+                    help="per-layer activation checkpointing (memory for "
+                         "compute; off for the headline number)")
     args = ap.parse_args()
 
     from kubetorch_amd.models import Llama, llama3_8b, llama_tiny
@@ -66,6 +69,8 @@ def main():
             model = Llama(cfg)
     finally:
         torch.set_default_dtype(prev_dtype)
+    if args.ckpt:
+        model.gradient_checkpointing_enable()
     engine = FlatDDP(model, lr=args.lr, bucket_mb=args.bucket_mb)
     engine.broadcast_params(src=0)
     log(f"[bench rank{rank}] model+engine ready in {time.time()-t_build:.1f}s "

@@ -198,7 +198,16 @@ class Llama(nn.Module):
         cos, sin = ops.precompute_rope(cfg.max_seq_len, cfg.head_dim, cfg.rope_base)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        self.gradient_checkpointing = False
         self.reset_parameters()
+
+    def gradient_checkpointing_enable(self, enabled=True):
+        """Per-layer activation checkpointing: store only the residual
+        stream between layers and recompute each block in backward (frees
+        the attn/MLP activations — the lever for long-context or
+        large-batch runs past 288 GB)."""
+        self.gradient_checkpointing = enabled
+        return self
 
     def _apply(self, fn, recurse=True):
         # Keep the RoPE tables fp32 across model.to(bf16): recompute rather
@@ -231,6 +240,21 @@ class Llama(nn.Module):
         res = self.embed(tokens)
         delta = None
         eps = self.cfg.norm_eps
+        if self.gradient_checkpointing and torch.is_grad_enabled():
+            from torch.utils.checkpoint import checkpoint
+
+            def block_step(layer, d, r):
+                n1, r = ops.add_rmsnorm(d, r, layer.attn_norm.weight, eps)
+                a = layer.attn(n1, cos, sin)
+                n2, r = ops.add_rmsnorm(a, r, layer.mlp_norm.weight, eps)
+                return layer.mlp(n2), r
+
+            delta = torch.zeros_like(res)  # uniform (delta, res) carry
+            for layer in self.layers:
+                delta, res = checkpoint(block_step, layer, delta, res,
+                                        use_reentrant=False)
+            h, _ = ops.add_rmsnorm(delta, res, self.norm.weight, eps)
+            return self.lm_head(h)
         for layer in self.layers:
             if delta is None:
                 n1 = ops.rmsnorm(res, layer.attn_norm.weight, eps)
