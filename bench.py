@@ -136,6 +136,15 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
+        # achieved model FLOP/s (6*N per token fwd+bwd + causal attention)
+        n_params = sum(p.numel() for p in model.parameters())
+        att = 12 * cfg.n_layers * cfg.dim * S * 0.5  # per token, causal
+        flops_per_tok = 6 * n_params + 3 * att  # bwd ~2x fwd attention
+        tf = toks_per_sec * flops_per_tok / 1e12
+        log(f"[bench] ~{tf:.0f} TFLOP/s model FLOPs "
+            f"({100 * tf / 2500:.0f}% of 2.5 PF dense bf16 peak)")
+
+    if rank == 0:
         result = {
             "metric": "llama3_8b_ddp_bf16_tokens_per_sec" if args.model == "llama3-8b"
                       else "tiny_ddp_tokens_per_sec",
