@@ -96,10 +96,15 @@ def put(key, src, window=None):
     sd = _flatten(src)
     pack = bool(window and getattr(window, "pack", False))
     if pack:
+        from kubetorch_amd import ops as kt_ops
+
         dtypes = {t.dtype for t in sd.values()}
         if len(dtypes) != 1:
             raise ValueError("packed mode requires a single dtype")
-        flat = torch.cat([t.reshape(-1) for t in sd.values()])
+        # one pack kernel instead of torch.cat's per-tensor copies; segment
+        # starts 16B-aligned so the kernel runs pure uint4 vectors
+        flat, _ = kt_ops.pack_tensors(
+            [t.contiguous() for t in sd.values()])
         cli.register(f"{key}/__packed__", flat)
     else:
         for sub, t in sd.items():
@@ -146,17 +151,20 @@ def get(key, dest, window=None):
     if missing:
         raise KeyError(f"dest keys not in stored state dict: {sorted(missing)}")
     if meta.get("packed"):
-        total = sum(e["numel"] for e in entries.values())
+        from kubetorch_amd import ops as kt_ops
+
         dt = getattr(torch, next(iter(entries.values()))["dtype"])
         first = next(iter(sd.values()))
+        numels = [e["numel"] for e in entries.values()]
+        offs, total = kt_ops.aligned_offsets(numels, dt.itemsize)
         flat = torch.empty(total, dtype=dt, device=first.device)
         fetch(f"{key}/__packed__", flat)
-        off = 0
-        for sub, e in entries.items():  # sorted publish order
-            n = e["numel"]
-            if sub in sd:
-                sd[sub].copy_(flat[off:off + n].view(e["shape"]))
-            off += n
+        # scatter with one unpack kernel where possible (publish order)
+        wanted = [(sub, e, o) for (sub, e), o
+                  in zip(entries.items(), offs) if sub in sd]
+        kt_ops.unpack_tensors(
+            flat, [sd[sub] for sub, _e, _o in wanted],
+            offsets=[o for _s, _e, o in wanted])
         return dest
     for sub in sd:
         fetch(f"{key}/{sub}", sd[sub])

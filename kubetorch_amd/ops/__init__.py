@@ -374,6 +374,72 @@ def flash_attention_v3_supported(S, head_dim, device, dtype):
 
 
 # ---------------------------------------------------------------------------
+# Multi-tensor pack/unpack (GPU data plane: packed state-dict transfers)
+# ---------------------------------------------------------------------------
+PACK_ALIGN = 16  # bytes; segment starts in the flat buffer are 16B-aligned
+
+
+def aligned_offsets(numels, elem_size):
+    """Element offsets into the packed flat buffer with every segment start
+    16B-aligned (so the pack kernel runs pure uint4 copies). Returns
+    (offsets, total_elems)."""
+    step = PACK_ALIGN // elem_size
+    offs, cur = [], 0
+    for n in numels:
+        offs.append(cur)
+        cur += -(-n // step) * step  # round numel up to the alignment step
+    return offs, cur
+
+
+def pack_tensors(tensors, flat=None):
+    """Pack same-dtype contiguous GPU tensors into one flat buffer with a
+    single kernel (vs torch.cat's per-tensor copies). Returns (flat,
+    offsets) where offsets are element positions (aligned)."""
+    dt = tensors[0].dtype
+    es = dt.itemsize
+    numels = [t.numel() for t in tensors]
+    offs, total = aligned_offsets(numels, es)
+    dev = tensors[0].device
+    if flat is None:
+        flat = torch.zeros(total, dtype=dt, device=dev)
+    if dev.type == "cuda" and hip_available():
+        ptrs = torch.tensor([t.data_ptr() for t in tensors],
+                            dtype=torch.int64).to(dev, non_blocking=True)
+        nb = torch.tensor([n * es for n in numels],
+                          dtype=torch.int64).to(dev, non_blocking=True)
+        ob = torch.tensor([o * es for o in offs],
+                          dtype=torch.int64).to(dev, non_blocking=True)
+        _ext().pack_segments(flat, ptrs, nb, ob, True, max(numels) * es)
+    else:
+        for t, o, n in zip(tensors, offs, numels):
+            flat[o:o + n].copy_(t.reshape(-1))
+    return flat, offs
+
+
+def unpack_tensors(flat, tensors, offsets=None):
+    """Scatter a packed flat buffer back into pre-allocated tensors (the
+    inverse of pack_tensors) with one kernel on GPU."""
+    es = flat.dtype.itemsize
+    numels = [t.numel() for t in tensors]
+    if offsets is None:
+        offsets, _ = aligned_offsets(numels, es)
+    if (flat.is_cuda and hip_available()
+            and all(t.is_contiguous() for t in tensors)):
+        dev = flat.device
+        ptrs = torch.tensor([t.data_ptr() for t in tensors],
+                            dtype=torch.int64).to(dev, non_blocking=True)
+        nb = torch.tensor([n * es for n in numels],
+                          dtype=torch.int64).to(dev, non_blocking=True)
+        ob = torch.tensor([o * es for o in offsets],
+                          dtype=torch.int64).to(dev, non_blocking=True)
+        _ext().pack_segments(flat, ptrs, nb, ob, False, max(numels) * es)
+    else:
+        for t, o in zip(tensors, offsets):
+            t.copy_(flat[o:o + t.numel()].view(t.shape))
+    return tensors
+
+
+# ---------------------------------------------------------------------------
 # Fused AdamW on flat bf16 buckets (used by kubetorch_amd.parallel)
 # ---------------------------------------------------------------------------
 def adamw_(p, g, m, v, lr, beta1, beta2, eps, wd, step, grad_scale=1.0):

@@ -42,6 +42,9 @@ void kt_attn_fwd_ck_tr(const void* q, const void* k, const void* v, void* o,
 void kt_attn_fwd_v3(const void* q, const void* k, const void* v, void* o,
                     void* lse, int B, int Hq, int Hkv, int S, float scale,
                     const long* strides, hipStream_t stream);
+void kt_pack_segments(const void* ptrs, const void* nbytes, const void* offs,
+                      void* base, int nseg, long max_nbytes, int to_base,
+                      hipStream_t stream);
 }
 
 namespace {
@@ -288,6 +291,25 @@ std::vector<at::Tensor> attn_fwd_v3(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
+void pack_segments(const at::Tensor& base, const at::Tensor& ptrs,
+                   const at::Tensor& nbytes, const at::Tensor& offs,
+                   bool to_base, int64_t max_nbytes) {
+  // One-kernel state-dict pack (to_base=true) / unpack (false) between the
+  // flat `base` buffer and the tensors whose data pointers are in `ptrs`.
+  // Offsets must be 16B-aligned (ops.aligned_offsets). ptrs/nbytes/offs are
+  // int64 CUDA tensors (device-side descriptor table).
+  TORCH_CHECK(base.is_cuda() && base.is_contiguous());
+  TORCH_CHECK(ptrs.is_cuda() && ptrs.scalar_type() == at::kLong);
+  TORCH_CHECK(nbytes.is_cuda() && offs.is_cuda());
+  const int nseg = (int)ptrs.numel();
+  TORCH_CHECK(nbytes.numel() == nseg && offs.numel() == nseg);
+  TORCH_CHECK(nseg > 0 && nseg < 65536, "1..65535 segments");
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(base.device());
+  kt_pack_segments(ptrs.data_ptr(), nbytes.data_ptr(), offs.data_ptr(),
+                   base.data_ptr(), nseg, (long)max_nbytes, to_base ? 1 : 0,
+                   cur_stream(base));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -312,4 +334,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("cross_entropy_fwd_", &cross_entropy_fwd_,
           "Fused CE: returns per-row loss, overwrites logits with grad");
   mod.def("adamw_", &adamw_, "Fused AdamW on a flat bf16 bucket");
+  mod.def("pack_segments", &pack_segments,
+          "One-kernel multi-tensor pack/unpack vs a flat buffer");
 }

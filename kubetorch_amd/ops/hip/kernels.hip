@@ -445,6 +445,38 @@ __global__ void adamw_tail_kernel(u16* p, const u16* g, float* m, float* v,
 }
 
 // ---------------------------------------------------------------------------
+// Multi-tensor pack/unpack for the GPU data plane (state-dict transfers).
+// One kernel moves a whole same-dtype state dict between N scattered
+// tensors and one flat buffer (reference does torch.cat + N split copies —
+// SURVEY.md §2.9 #8 names this the hand-written-HIP candidate). Segment
+// starts in the flat buffer are 16 B-aligned (the host computes padded
+// offsets), and torch allocations are 256 B-aligned, so every copy runs as
+// uint4 (16 B) vectors with a byte tail. blockIdx.y = segment; x grid-strides
+// within it (small segments' blocks exit immediately — launch cost only).
+// ---------------------------------------------------------------------------
+typedef uint vec4ui __attribute__((ext_vector_type(4)));
+
+__global__ void pack_segments_kernel(const unsigned long long* __restrict__ ptrs,
+                                     const long* __restrict__ nbytes,
+                                     const long* __restrict__ offs,
+                                     char* __restrict__ base, int to_base) {
+  const int seg = blockIdx.y;
+  const long B = nbytes[seg];
+  char* flat = base + offs[seg];                       // 16 B aligned
+  char* t = reinterpret_cast<char*>(ptrs[seg]);        // torch alloc: aligned
+  char* dst = to_base ? flat : t;
+  const char* src = to_base ? (const char*)t : (const char*)flat;
+  const long nvec = B >> 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long)gridDim.x * blockDim.x) {
+    reinterpret_cast<vec4ui*>(dst)[i] =
+        reinterpret_cast<const vec4ui*>(src)[i];
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (B & 15))
+    dst[(nvec << 4) + threadIdx.x] = src[(nvec << 4) + threadIdx.x];
+}
+
+// ---------------------------------------------------------------------------
 // C launchers
 // ---------------------------------------------------------------------------
 static inline int grid_for(long work, int block, int cap = 2048) {
@@ -530,6 +562,16 @@ void kt_adamw(void* p, const void* g, void* m, void* v, long n, float lr,
                        (u16*)p, (const u16*)g, (float*)m, (float*)v, nvec << 3,
                        n, lr, beta1, beta2, eps, wd, bc1, bc2, grad_scale);
   }
+}
+
+void kt_pack_segments(const void* ptrs, const void* nbytes, const void* offs,
+                      void* base, int nseg, long max_nbytes, int to_base,
+                      hipStream_t stream) {
+  int gx = grid_for(max_nbytes >> 4, 256, 1024);
+  hipLaunchKernelGGL(pack_segments_kernel, dim3(gx, nseg), dim3(256), 0,
+                     stream, (const unsigned long long*)ptrs,
+                     (const long*)nbytes, (const long*)offs, (char*)base,
+                     to_base);
 }
 
 }  // extern "C"

@@ -388,3 +388,26 @@ def test_hip_extension_required_on_gpu():
     w = torch.ones(64, dtype=BF16, device="cuda")
     y = ops.rmsnorm(x, w)
     assert y.is_cuda and y.dtype == BF16
+
+
+@pytest.mark.gpu
+def test_pack_segments_kernel():
+    """One-kernel state-dict pack/unpack vs per-tensor reference (odd sizes
+    exercise the 16B alignment padding and the byte tail)."""
+    torch.manual_seed(5)
+    sizes = [(3,), (1000000,), (127, 33), (1,), (4096, 64), (5, 5, 5)]
+    ts = [torch.randn(s, dtype=BF16, device="cuda") for s in sizes]
+    flat, offs = ops.pack_tensors(ts)
+    # reference layout: aligned offsets, python copies
+    es = BF16.itemsize
+    r_offs, total = ops.aligned_offsets([t.numel() for t in ts], es)
+    assert offs == r_offs and flat.numel() == total
+    ref = torch.zeros(total, dtype=BF16, device="cuda")
+    for t, o in zip(ts, offs):
+        ref[o:o + t.numel()].copy_(t.reshape(-1))
+    assert torch.equal(flat, ref)
+    # unpack roundtrip into fresh tensors
+    outs = [torch.zeros_like(t) for t in ts]
+    ops.unpack_tensors(flat, outs, offsets=offs)
+    for a, b in zip(ts, outs):
+        assert torch.equal(a, b)
