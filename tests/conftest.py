@@ -12,8 +12,22 @@ import pytest
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
+def pytest_addoption(parser):
+    parser.addoption(
+        "--level", default=None,
+        choices=["unit", "minimal", "release", "gpu"],
+        help="reference-parity test levels: unit = fast in-process tests "
+             "only; minimal = + full-stack local-driver e2e; release = all "
+             "CPU tests; gpu = only GPU-marked tests. Default: no level "
+             "filtering (marker expressions like -m 'not gpu' still apply).")
+
+
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
+    config.addinivalue_line(
+        "markers",
+        "minimal: full-stack e2e through the local driver (subprocess "
+        "pods) — the reference's 'minimal' integration tier")
     config.addinivalue_line(
         "markers",
         "flaky_retry: rerun once on failure (multi-process e2e tests are "
@@ -39,6 +53,23 @@ def pytest_runtest_protocol(item, nextitem):
 
 
 def pytest_collection_modifyitems(config, items):
+    level = config.getoption("--level")
+    if level:
+        keep = []
+        for item in items:
+            is_gpu = "gpu" in item.keywords
+            is_minimal = "minimal" in item.keywords
+            if level == "gpu":
+                ok = is_gpu
+            elif level == "unit":
+                ok = not is_gpu and not is_minimal
+            elif level == "minimal":
+                ok = not is_gpu
+            else:  # release: everything CPU (gpu still gated below)
+                ok = not is_gpu
+            if ok:
+                keep.append(item)
+        items[:] = keep
     try:
         import torch
 

@@ -11,7 +11,7 @@ os.environ["KT_USERNAME"] = "clitest"
 import pytest  # noqa: E402
 import kubetorch_amd as kt  # noqa: E402
 
-pytestmark = pytest.mark.flaky_retry
+pytestmark = [pytest.mark.flaky_retry, pytest.mark.minimal]
 from kubetorch_amd.cli import app  # noqa: E402
 from tests.assets.summer import summer as summer_mod  # noqa: E402
 

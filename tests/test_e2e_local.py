@@ -15,7 +15,7 @@ os.environ["KT_USERNAME"] = "citest"
 
 import kubetorch_amd as kt  # noqa: E402
 
-pytestmark = pytest.mark.flaky_retry
+pytestmark = [pytest.mark.flaky_retry, pytest.mark.minimal]
 from tests.assets.summer import summer as summer_mod  # noqa: E402
 
 

@@ -15,7 +15,7 @@ os.environ["KT_USERNAME"] = "faulttest"
 
 import kubetorch_amd as kt  # noqa: E402
 
-pytestmark = pytest.mark.flaky_retry
+pytestmark = [pytest.mark.flaky_retry, pytest.mark.minimal]
 from kubetorch_amd.exceptions import WorkerMembershipChanged  # noqa: E402
 from tests.assets.summer import summer as summer_mod  # noqa: E402
 

@@ -12,7 +12,7 @@ os.environ["KT_USERNAME"] = "rltest"
 
 import kubetorch_amd as kt  # noqa: E402
 
-pytestmark = pytest.mark.flaky_retry
+pytestmark = [pytest.mark.flaky_retry, pytest.mark.minimal]
 from tests.assets.rl_services import rl_services  # noqa: E402
 
 
