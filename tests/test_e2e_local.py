@@ -107,6 +107,16 @@ class TestDistributed:
         assert all(r["master_addr"] == "127.0.0.1" for r in results)
         assert sorted({r["node_rank"] for r in results}) == [0, 1]
 
+    def test_worker_selection(self, dist_fn):
+        """workers=[0] runs only node 0's ranks; workers="any" runs one
+        node (reference: spmd worker-filter arg)."""
+        only0 = dist_fn(kt_workers=[0])
+        assert len(only0) == 2  # num_proc=2 on one node
+        assert all(r["node_rank"] == 0 for r in only0)
+        any_one = dist_fn(kt_workers="any")
+        assert len(any_one) == 2
+        assert len({r["node_rank"] for r in any_one}) == 1
+
     def test_gloo_allreduce(self, dist_fn):
         """User code runs torch.distributed through the env contract."""
         ddp = kt.fn(summer_mod.gloo_allreduce).to(

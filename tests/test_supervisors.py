@@ -1,0 +1,48 @@
+"""Unit tests for the distribution layer's per-framework env contracts and
+worker-selection logic (reference parity: test_distributed.py asserts the
+RANK/WORLD_SIZE/... env on every rank and the worker-filter arg)."""
+import json
+
+from kubetorch_amd.serving.supervisors import (
+    PROCESS_CLASSES,
+    JaxProcess,
+    TensorflowProcess,
+    TorchProcess,
+)
+
+HOSTS = ["10.0.0.1:32300", "10.0.0.2:32300", "10.0.0.3:32300"]
+
+
+def test_torch_env_contract():
+    env = TorchProcess.env_vars(HOSTS, node_rank=1, local_rank=3, num_proc=8)
+    assert env["WORLD_SIZE"] == "24"
+    assert env["RANK"] == "11"          # node_rank*num_proc + local_rank
+    assert env["LOCAL_RANK"] == "3"
+    assert env["NODE_RANK"] == "1"
+    assert env["LOCAL_WORLD_SIZE"] == "8"
+    assert env["MASTER_ADDR"] == "10.0.0.1"
+    assert env["POD_IPS"] == "10.0.0.1,10.0.0.2,10.0.0.3"
+    # RCCL/xGMI defaults ride along (incl. the dmabuf IPC requirement)
+    assert env.get("HSA_ENABLE_IPC_MODE_LEGACY") == "0"
+
+
+def test_jax_env_contract():
+    env = JaxProcess.env_vars(HOSTS, node_rank=2, local_rank=0, num_proc=1)
+    assert env["JAX_COORDINATOR_ADDRESS"] == "10.0.0.1:1234"
+    assert env["JAX_PROCESS_ID"] == "2"
+    assert env["JAX_NUM_PROCESSES"] == "3"
+    assert env["JAX_LOCAL_DEVICE_IDS"] == "0"
+
+
+def test_tf_env_contract():
+    env = TensorflowProcess.env_vars(HOSTS, node_rank=1, local_rank=0,
+                                     num_proc=1)
+    cfg = json.loads(env["TF_CONFIG"])
+    assert cfg["cluster"]["worker"] == [
+        "10.0.0.1:2222", "10.0.0.2:2222", "10.0.0.3:2222"]
+    assert cfg["task"] == {"type": "worker", "index": 1}
+
+
+def test_process_class_registry():
+    for name in ("pytorch", "jax", "tensorflow", "spmd"):
+        assert name in PROCESS_CLASSES
