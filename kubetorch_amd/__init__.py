@@ -17,6 +17,11 @@ _LAZY = {
     "fn": ("kubetorch_amd.client.fn", "fn"),
     "cls": ("kubetorch_amd.client.cls", "cls"),
     "app": ("kubetorch_amd.client.app", "app"),
+    "Fn": ("kubetorch_amd.client.fn", "Fn"),
+    "Cls": ("kubetorch_amd.client.cls", "Cls"),
+    "App": ("kubetorch_amd.client.app", "App"),
+    "config": ("kubetorch_amd.config", "config"),
+    "secret": ("kubetorch_amd.resources.secret", "secret_factory"),
     "Compute": ("kubetorch_amd.resources.compute", "Compute"),
     "Image": ("kubetorch_amd.resources.image", "Image"),
     "Volume": ("kubetorch_amd.resources.volume", "Volume"),
@@ -35,6 +40,18 @@ _LAZY = {
     "get_broadcast": ("kubetorch_amd.data_store.commands", "get_broadcast"),
     "BroadcastWindow": ("kubetorch_amd.data_store.types", "BroadcastWindow"),
     "pod_ips": ("kubetorch_amd.serving.discovery", "pod_ips"),
+    "distributed": ("kubetorch_amd.distributed", None),
+    # exception types users catch (reference exports these at top level)
+    "KubetorchError": ("kubetorch_amd.exceptions", "KubetorchError"),
+    "LaunchError": ("kubetorch_amd.exceptions", "LaunchError"),
+    "ImagePullError": ("kubetorch_amd.exceptions", "ImagePullError"),
+    "ResourceNotAvailableError": (
+        "kubetorch_amd.exceptions", "ResourceNotAvailableError"),
+    "PodTerminatedError": ("kubetorch_amd.exceptions", "PodTerminatedError"),
+    "WorkerMembershipChanged": (
+        "kubetorch_amd.exceptions", "WorkerMembershipChanged"),
+    "RemoteCallError": ("kubetorch_amd.exceptions", "RemoteCallError"),
+    "QuorumTimeout": ("kubetorch_amd.exceptions", "QuorumTimeout"),
     "ops": ("kubetorch_amd.ops", None),
     "models": ("kubetorch_amd.models", None),
     "parallel": ("kubetorch_amd.parallel", None),
