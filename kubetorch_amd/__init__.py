@@ -52,6 +52,7 @@ _LAZY = {
         "kubetorch_amd.exceptions", "WorkerMembershipChanged"),
     "RemoteCallError": ("kubetorch_amd.exceptions", "RemoteCallError"),
     "QuorumTimeout": ("kubetorch_amd.exceptions", "QuorumTimeout"),
+    "deep_breakpoint": ("kubetorch_amd.serving.pdb_ws", "deep_breakpoint"),
     "ops": ("kubetorch_amd.ops", None),
     "models": ("kubetorch_amd.models", None),
     "parallel": ("kubetorch_amd.parallel", None),
