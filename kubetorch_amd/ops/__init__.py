@@ -325,10 +325,25 @@ class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
+        import os
+
         q, k, v, o, lse = ctx.saved_tensors
         B, Hq, S, D = q.shape
         Hkv = k.shape[1]
         g = Hq // Hkv
+        if os.environ.get("KT_ATTN_BWD") == "ck" and S % 128 == 0:
+            # Opt-in: CK-tile GQA-native bwd (no KV expansion; dk/dv come
+            # back Hq-expanded and are group-summed). Round-2 default
+            # candidate — see profiles/ROUND2.md lever #1.
+            dq, dk_e, dv_e = _ext().attn_bwd_ck(
+                grad_out.contiguous(), q.contiguous(), k.contiguous(),
+                v.contiguous(), o.contiguous(), lse.contiguous(), ctx.scale)
+            if g > 1:
+                dk = dk_e.view(B, Hkv, g, S, D).sum(2)
+                dv = dv_e.view(B, Hkv, g, S, D).sum(2)
+            else:
+                dk, dv = dk_e, dv_e
+            return dq, dk, dv, None, None
         if g > 1:  # expand KV for the dense backward, then reduce over groups
             k_exp = k.repeat_interleave(g, dim=1)
             v_exp = v.repeat_interleave(g, dim=1)

@@ -17,7 +17,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 # module name -> sources (each builds an in-tree .so)
 MODULES = {
-    "_hip_ops": ["kernels.hip", "attention.hip", "attention_ck.hip", "bindings.cpp"],
+    "_hip_ops": ["kernels.hip", "attention.hip", "attention_ck.hip", "attention_ck_bwd.hip", "bindings.cpp"],
     "_hip_spill": ["spill.cpp"],
 }
 SO_PATH = os.path.join(OPS_DIR, "_hip_ops.so")  # primary (back-compat)
@@ -86,7 +86,7 @@ def _build_module(name, sources, verbose, force):
     for src in sources:
         obj = os.path.join(BUILD_DIR, os.path.splitext(src)[0] + ".o")
         src_flags = flags
-        if src in ("attention.hip", "attention_ck.hip"):
+        if src in ("attention.hip", "attention_ck.hip", "attention_ck_bwd.hip"):
             # rocWMMA/ck_tile need the __half conversions torch's flags
             # disable; these TUs have no torch headers -> drop the guards.
             src_flags = [f for f in flags

@@ -45,6 +45,10 @@ void kt_attn_fwd_v3(const void* q, const void* k, const void* v, void* o,
 void kt_pack_segments(const void* ptrs, const void* nbytes, const void* offs,
                       void* base, int nseg, long max_nbytes, int to_base,
                       hipStream_t stream);
+void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
+                    const void* o, const void* do_, const void* lse, void* d,
+                    void* dq_acc, void* dq, void* dk, void* dv, int B, int Hq,
+                    int Hkv, int S, float scale, hipStream_t stream);
 }
 
 namespace {
@@ -291,6 +295,37 @@ std::vector<at::Tensor> attn_fwd_v3(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
+std::vector<at::Tensor> attn_bwd_ck(const at::Tensor& grad_out,
+                                    const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& v, const at::Tensor& o,
+                                    const at::Tensor& lse, double scale) {
+  // CK-tile GQA-native FMHA backward (attention_ck_bwd.hip). Contiguous
+  // [B,H,S,128] bf16 in; returns (dq, dk, dv) with dk/dv Hq-EXPANDED — the
+  // autograd wrapper sums KV-head groups.
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(o.is_contiguous() && lse.is_contiguous());
+  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
+  const int Hkv = (int)k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0 && S % 128 == 0);
+  TORCH_CHECK(lse.sizes() == at::IntArrayRef({B, Hq, S}) &&
+              lse.scalar_type() == at::kFloat);
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(q.device());
+  auto go = grad_out.contiguous();
+  auto dq = at::empty_like(q);
+  auto dk = at::empty({B, Hq, S, 128}, k.options());
+  auto dv = at::empty({B, Hq, S, 128}, v.options());
+  auto d = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  // dq is accumulated atomically across key blocks -> fp32, zero-init
+  auto dq_acc = at::zeros({B, Hq, S, 128}, q.options().dtype(at::kFloat));
+  kt_attn_bwd_ck(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                 go.data_ptr(), lse.data_ptr(), d.data_ptr(),
+                 dq_acc.data_ptr(), dq.data_ptr(), dk.data_ptr(),
+                 dv.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
+  return {dq, dk, dv};
+}
+
 void pack_segments(const at::Tensor& base, const at::Tensor& ptrs,
                    const at::Tensor& nbytes, const at::Tensor& offs,
                    bool to_base, int64_t max_nbytes) {
@@ -336,4 +371,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_", &adamw_, "Fused AdamW on a flat bf16 bucket");
   mod.def("pack_segments", &pack_segments,
           "One-kernel multi-tensor pack/unpack vs a flat buffer");
+  mod.def("attn_bwd_ck", &attn_bwd_ck,
+          "CK-tile GQA-native FMHA backward -> (dq, dk_exp, dv_exp)");
 }

@@ -5,6 +5,8 @@ PyTorch fp32 implementations. GPU tests (-m gpu) compare the gfx950 HIP
 kernels against the same fp32 references (reference parity model:
 SURVEY.md §4 — kernel vs plain fp32 torch).
 """
+import os
+
 import pytest
 import torch
 import torch.nn.functional as F
@@ -411,3 +413,38 @@ def test_pack_segments_kernel():
     ops.unpack_tensors(flat, outs, offsets=offs)
     for a, b in zip(ts, outs):
         assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("KT_ATTN_BWD_TEST") != "1",
+                    reason="CK bwd WIP: causal mask orientation flipped "
+                           "(profiles/ROUND2.md lever #1) — enable with "
+                           "KT_ATTN_BWD_TEST=1 while fixing")
+def test_attn_bwd_ck_gqa_native():
+    """CK-tile GQA-native backward vs fp32 SDPA reference (opt-in path,
+    KT_ATTN_BWD=ck; round-2 default candidate)."""
+    torch.manual_seed(9)
+    B, Hq, Hkv, S, D = 2, 8, 2, 256, 128
+    scale = D ** -0.5
+    q = torch.randn(B, Hq, S, D, dtype=BF16, device="cuda")
+    k = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda")
+    v = torch.randn(B, Hkv, S, D, dtype=BF16, device="cuda")
+    # forward via the tr kernel for (o, lse)
+    o, lse = ops._ext().attn_fwd_ck_tr(q, k, v, scale)
+    gout = torch.randn_like(o)
+    dq, dk_e, dv_e = ops._ext().attn_bwd_ck(gout, q, k, v, o.contiguous(),
+                                            lse.contiguous(), scale)
+    g = Hq // Hkv
+    dk = dk_e.view(B, Hkv, g, S, D).sum(2)
+    dv = dv_e.view(B, Hkv, g, S, D).sum(2)
+
+    qr = q.float().requires_grad_(True)
+    kr = k.float().requires_grad_(True)
+    vr = v.float().requires_grad_(True)
+    outr = F.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(g, 1), vr.repeat_interleave(g, 1),
+        is_causal=True, scale=scale)
+    outr.backward(gout.float())
+    _assert_close(dq, qr.grad, msg="ck bwd dq")
+    _assert_close(dk, kr.grad, msg="ck bwd dk")
+    _assert_close(dv, vr.grad, msg="ck bwd dv")
