@@ -19,17 +19,29 @@ def train(steps: int = 50):
     with torch.device(dev):
         model = Llama(cfg)
     torch.set_default_dtype(prev)
-    engine = FlatDDP(model, lr=1e-4)
+    engine = FlatDDP(model, lr=1e-4, clip_norm=1.0)
     engine.broadcast_params(src=0)
 
-    x = torch.randint(0, cfg.vocab_size, (4, 4096), device=dev)
-    y = torch.randint(0, cfg.vocab_size, (4, 4096), device=dev)
+    from kubetorch_amd.data import ShardedLoader, TokenDataset, synthetic_tokens
+    from kubetorch_amd.parallel import warmup_cosine
+
+    ds = TokenDataset(synthetic_tokens(cfg.vocab_size, 4096 * 512), 4096)
+    loader = ShardedLoader(ds, batch=4, rank=rank, world=world,
+                           device=dev)  # pinned one-ahead H2D prefetch
+    it = iter(loader)
     for step in range(steps):
+        try:
+            x, y = next(it)
+        except StopIteration:
+            loader.set_epoch(loader.epoch + 1)
+            it = iter(loader)
+            x, y = next(it)
         loss = model.loss(x, y)
         loss.backward()
-        engine.step()
+        engine.step(lr=warmup_cosine(step, 1e-4, 10, steps))
         if rank == 0 and step % 10 == 0:
-            print(f"step {step}: loss {loss.item():.4f}")
+            print(f"step {step}: loss {loss.item():.4f} "
+                  f"grad_norm {engine.last_grad_norm:.2f}")
     if dist.is_initialized():
         dist.destroy_process_group()
     return {"rank": rank, "final_loss": loss.item()}
