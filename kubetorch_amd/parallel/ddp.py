@@ -114,6 +114,11 @@ class FlatDDP:
         # steps.
         self._copy_mode = grad_accum_steps == 1
         self.grad_accum_steps = grad_accum_steps
+        # fold mean-over-(ranks x micro-batches) into the fused kernel's
+        # grad scale: buckets hold the raw SUM, so the optimizer sees the
+        # mean gradient of the effective global batch (no LR surprises
+        # when grad_accum_steps changes).
+        self._grad_scale = 1.0 / (self._world * grad_accum_steps)
         self._micro_step = 0
         params = [p for p in model.parameters() if p.requires_grad]
         if not params:
@@ -209,7 +214,7 @@ class FlatDDP:
             ops.adamw_(
                 b.flat_param, b.flat_grad, b.m, b.v, self.lr,
                 self.betas[0], self.betas[1], self.eps, self.weight_decay,
-                self.step_count, 1.0 / self._world,
+                self.step_count, self._grad_scale,
             )
 
     # -- optimizer side ------------------------------------------------------
@@ -227,15 +232,15 @@ class FlatDDP:
             self._step_started = False
         else:
             self.step_count += 1
-            grad_scale = 1.0 / self._world
+            grad_scale = self._grad_scale
             for b in self.buckets:
                 if b.work is not None:
                     b.work.wait()
                     b.work = None
             if self.clip_norm is not None:
-                # flat grads hold the SUM over ranks; the norm of the mean
-                # grad is ||g_sum|| / world. Identical on every rank (the
-                # grads are already reduced), so no extra collective.
+                # flat grads hold the SUM over ranks (and micro-batches);
+                # the mean grad's norm is ||g_sum|| * grad_scale. Identical
+                # on every rank (grads already reduced) — no collective.
                 total_sq = sum(
                     (b.flat_grad.float() ** 2).sum() for b in self.buckets
                 )

@@ -101,9 +101,9 @@ def test_flatddp_two_ranks_match_single_process():
         for b, g in zip(eng.buckets, g0):
             b.flat_grad.add_(g)
         # emulate SUM all-reduce then 1/world scale inside step
-        eng._world = 2
+        eng._grad_scale = 0.5
         eng.step()
-        eng._world = 1
+        eng._grad_scale = 1.0
     single = torch.cat([b.flat_param.float() for b in eng.buckets])
     torch.testing.assert_close(results[0], single, rtol=2e-2, atol=2e-2)
 
@@ -164,3 +164,31 @@ def test_lr_schedules():
     # monotone decay after warmup
     vals = [warmup_cosine(s, 1.0, 10, 100) for s in range(10, 101)]
     assert all(a >= b for a, b in zip(vals, vals[1:]))
+
+
+def test_grad_accum_mean_semantics():
+    """grad_accum_steps=A over the same micro-batch repeated A times must
+    produce EXACTLY the single-step update (mean over micro-batches is
+    folded into the fused optimizer's grad scale — no hidden LR*A)."""
+    from kubetorch_amd.models import Llama, llama_tiny
+    from kubetorch_amd.parallel import FlatDDP
+
+    torch.manual_seed(3)
+    x = torch.randint(0, 512, (2, 32))
+    y = torch.randint(0, 512, (2, 32))
+
+    torch.manual_seed(0)
+    m1 = Llama(llama_tiny())
+    e1 = FlatDDP(m1, lr=1e-3, bucket_mb=4)
+    m1.loss(x, y).backward()
+    e1.step()
+
+    torch.manual_seed(0)
+    m2 = Llama(llama_tiny())
+    e2 = FlatDDP(m2, lr=1e-3, bucket_mb=4, grad_accum_steps=2)
+    m2.loss(x, y).backward()
+    m2.loss(x, y).backward()   # accumulates: bucket holds 2x the grad
+    e2.step()
+
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1, p2, rtol=0, atol=0)
