@@ -188,3 +188,45 @@ def test_gradient_checkpointing_grads_match():
         assert p2.grad is not None, n
         torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-5, atol=1e-6,
                                    msg=n)
+
+
+def test_hf_checkpoint_roundtrip(tmp_path):
+    """kt -> HF safetensors export -> load_hf_checkpoint must reproduce
+    the exact model (fused wqkv/gate_up split + concat are inverses)."""
+    import json
+
+    from safetensors.torch import save_file
+
+    from kubetorch_amd.models import convert
+
+    torch.manual_seed(21)
+    cfg = llama_tiny()
+    src = Llama(cfg)
+    hf_sd = convert.kt_to_hf_state_dict(src.state_dict(), cfg)
+    # HF checkpoints shard weights; emulate two shards
+    keys = sorted(hf_sd)
+    half = len(keys) // 2
+    save_file({k: hf_sd[k].contiguous() for k in keys[:half]},
+              str(tmp_path / "model-00001-of-00002.safetensors"))
+    save_file({k: hf_sd[k].contiguous() for k in keys[half:]},
+              str(tmp_path / "model-00002-of-00002.safetensors"))
+    (tmp_path / "config.json").write_text(json.dumps({
+        "hidden_size": cfg.dim, "num_hidden_layers": cfg.n_layers,
+        "num_attention_heads": cfg.n_heads,
+        "num_key_value_heads": cfg.n_kv_heads,
+        "intermediate_size": cfg.intermediate,
+        "vocab_size": cfg.vocab_size, "max_position_embeddings": cfg.max_seq_len,
+        "rope_theta": cfg.rope_base, "rms_norm_eps": cfg.norm_eps,
+    }))
+
+    model = convert.load_hf_checkpoint(str(tmp_path), dtype=torch.float32)
+    assert model.cfg.dim == cfg.dim and model.cfg.n_kv_heads == cfg.n_kv_heads
+    x = torch.randint(0, cfg.vocab_size, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(model(x), src(x), rtol=1e-5, atol=1e-5)
+    # tied-embeddings fallback: drop lm_head from the HF dict
+    hf_sd2 = dict(hf_sd)
+    hf_sd2.pop("lm_head.weight")
+    kt_sd = convert.hf_to_kt_state_dict(hf_sd2, cfg)
+    torch.testing.assert_close(kt_sd["lm_head.weight"],
+                               hf_sd["model.embed_tokens.weight"])
