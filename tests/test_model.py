@@ -230,3 +230,28 @@ def test_hf_checkpoint_roundtrip(tmp_path):
     kt_sd = convert.hf_to_kt_state_dict(hf_sd2, cfg)
     torch.testing.assert_close(kt_sd["lm_head.weight"],
                                hf_sd["model.embed_tokens.weight"])
+
+
+def test_tokenizer_hf_roundtrip(tmp_path):
+    """Train a tiny in-test BPE (offline) and round-trip through the
+    serving tokenizer wrapper."""
+    from tokenizers import Tokenizer as HFTok
+    from tokenizers.models import BPE
+    from tokenizers.pre_tokenizers import Whitespace
+    from tokenizers.trainers import BpeTrainer
+
+    tok = HFTok(BPE(unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    tok.train_from_iterator(
+        ["the quick brown fox", "jumps over the lazy dog",
+         "pack my box with five dozen liquor jugs"] * 20,
+        BpeTrainer(vocab_size=200, special_tokens=["[UNK]", "<s>", "</s>"]))
+    tok.save(str(tmp_path / "tokenizer.json"))
+
+    from kubetorch_amd.models.tokenizer import Tokenizer
+
+    t = Tokenizer.load(str(tmp_path))  # resolves the file inside the dir
+    ids = t.encode("the quick brown fox")
+    assert ids and all(isinstance(i, int) for i in ids)
+    assert "quick" in t.decode(ids)
+    assert t.vocab_size > 0
