@@ -93,11 +93,18 @@ void launch(hipStream_t stream, dim3 grid, Args&&... args) {
 // All tensors contiguous [B, H, S, 128] bf16 (Hq for q/o/do/dq, Hkv for
 // k/v; dk/dv are Hq-EXPANDED — the caller group-sums). lse/d: [B, Hq, S]
 // f32. dq_acc: [B, Hq, S, 128] f32 ZEROED by the caller (atomic accum).
+// mask_mode selects the causal-mask karg convention (runtime, so round-2's
+// flipped-mask A/B needs no recompile — profiles/ROUND2.md lever #1):
+//   0: window (-1, 0), MASK_FROM_TOP_LEFT     (matches the fwd kernels)
+//   1: window (-1, 0), MASK_FROM_BOTTOM_RIGHT
+//   2: window (0, -1), MASK_FROM_TOP_LEFT     (anti-causal at the
+//      constructor — causal if the trload pipeline masks the S^T tile)
 extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
                                const void* o, const void* do_, const void* lse,
                                void* d, void* dq_acc, void* dq, void* dk,
                                void* dv, int B, int Hq, int Hkv, int S,
-                               float scale, hipStream_t stream) {
+                               float scale, int mask_mode,
+                               hipStream_t stream) {
   const ck_tile::index_t D = 128;
   const ck_tile::index_t sq = (ck_tile::index_t)S * D;   // nhead stride q-side
   const ck_tile::index_t bq = (ck_tile::index_t)Hq * sq; // batch stride q-side
@@ -130,9 +137,12 @@ extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
         (ck_tile::index_t)Hq * S /* batch_stride_lsed */,
         bq /* batch_stride_dq_acc */, bq /* batch_stride_dk */,
         bq /* batch_stride_dv */, 0,
-        0 /* split_stride_dq_acc */, -1 /* window_left */,
-        0 /* window_right */,
-        (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
+        0 /* split_stride_dq_acc */,
+        (ck_tile::index_t)(mask_mode == 2 ? 0 : -1) /* window_left */,
+        (ck_tile::index_t)(mask_mode == 2 ? -1 : 0) /* window_right */,
+        (ck_tile::index_t)(mask_mode == 1
+            ? ck_tile::GenericAttentionMaskEnum::MASK_FROM_BOTTOM_RIGHT
+            : ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT),
         0.0f /* p_drop */, std::make_pair<uint64_t, uint64_t>(0, 0));
     launch<BwdKernel>(stream, BwdKernel::GridSize(B, Hq, S), kargs);
   }

@@ -2,6 +2,8 @@
 // Host-only translation unit; compiled by hipcc against torch-ROCm's
 // native HIP API surface (c10/hip). No CUDA-compat shims.
 
+#include <cstdlib>
+
 #include <torch/extension.h>
 
 // torch-ROCm registers GPU tensors under DeviceType::CUDA; its native HIP
@@ -48,7 +50,8 @@ void kt_pack_segments(const void* ptrs, const void* nbytes, const void* offs,
 void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
                     const void* o, const void* do_, const void* lse, void* d,
                     void* dq_acc, void* dq, void* dk, void* dv, int B, int Hq,
-                    int Hkv, int S, float scale, hipStream_t stream);
+                    int Hkv, int S, float scale, int mask_mode,
+                    hipStream_t stream);
 }
 
 namespace {
@@ -319,10 +322,16 @@ std::vector<at::Tensor> attn_bwd_ck(const at::Tensor& grad_out,
   auto d = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   // dq is accumulated atomically across key blocks -> fp32, zero-init
   auto dq_acc = at::zeros({B, Hq, S, 128}, q.options().dtype(at::kFloat));
+  // KT_CKBWD_MASK selects the causal-mask convention at runtime (WIP
+  // flipped-mask A/B, profiles/ROUND2.md): 0=top-left (fwd convention),
+  // 1=bottom-right, 2=swapped lr window.
+  int mask_mode = 0;
+  if (const char* mm = std::getenv("KT_CKBWD_MASK")) mask_mode = atoi(mm);
   kt_attn_bwd_ck(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                  go.data_ptr(), lse.data_ptr(), d.data_ptr(),
                  dq_acc.data_ptr(), dq.data_ptr(), dk.data_ptr(),
-                 dv.data_ptr(), B, Hq, Hkv, S, (float)scale, cur_stream(q));
+                 dv.data_ptr(), B, Hq, Hkv, S, (float)scale, mask_mode,
+                 cur_stream(q));
   return {dq, dk, dv};
 }
 
