@@ -5,3 +5,4 @@ from kubetorch_amd.models.llama import (  # noqa: F401
     llama3_8b,
     llama_tiny,
 )
+from kubetorch_amd.models.serving import BatchedGenerator  # noqa: F401

@@ -106,6 +106,19 @@ class Attention(nn.Module):
                 v = v.repeat_interleave(rep, dim=1)
         return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
 
+    def _sdpa_masked(self, q, k, v, mask):
+        """Boolean-masked SDPA (True = attend) for ragged decode batches
+        (models/serving.py: per-slot cache lengths)."""
+        if self.n_kv != self.n_heads:
+            try:
+                return F.scaled_dot_product_attention(
+                    q, k, v, attn_mask=mask, enable_gqa=True)
+            except (TypeError, RuntimeError):
+                rep = self.n_heads // self.n_kv
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+        return F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
+
     def forward(self, x, cos, sin, cache=None, layer=0):
         import os
 
