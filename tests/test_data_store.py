@@ -13,6 +13,9 @@ import time
 import pytest
 import torch
 
+# daemon-spawning tests are sensitive to CI host load spikes
+pytestmark = pytest.mark.flaky_retry
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
