@@ -42,6 +42,8 @@ class Module:
 
     @property
     def name(self):
+        if getattr(self, "_bound_full_name", None):
+            return self._bound_full_name  # bound to an existing service
         base = self._name or self.pointers["name"]
         return sanitize_name(f"{config.username}-{base}")
 
@@ -79,14 +81,79 @@ class Module:
         except Exception:
             pass  # no data store deployed: pods must have the code baked in
 
-    def to(self, compute: Compute = None, init_args=None):
-        """Deploy (or hot-reload) this module onto the compute."""
+    def _reload_prefixes(self, reload_prefixes):
+        """Default service-lookup order: username, current git branch, prod
+        (reference: Module.to get_if_exists / reload_prefixes)."""
+        if reload_prefixes:
+            return list(reload_prefixes)
+        prefixes = [config.username]
+        try:
+            import subprocess
+
+            br = subprocess.run(
+                ["git", "rev-parse", "--abbrev-ref", "HEAD"],
+                capture_output=True, text=True, timeout=5,
+            ).stdout.strip()
+            if br and br != "HEAD":
+                prefixes.append(br)
+        except Exception:
+            pass
+        prefixes.append("prod")
+        return prefixes
+
+    def _get_existing_service(self, reload_prefixes=None):
+        """Find an already-deployed service for this module under the
+        fallback prefixes. Returns the workload dict or None."""
+        base = self._name or self.pointers["name"]
+        candidates = [sanitize_name(f"{p}-{base}")
+                      for p in self._reload_prefixes(reload_prefixes)]
+        try:
+            workloads = controller_client().list_workloads(
+                self.namespace).get("workloads", [])
+        except Exception:
+            return None
+        by_name = {w["name"]: w for w in workloads}
+        for cand in candidates:
+            if cand in by_name:
+                return by_name[cand]
+        return None
+
+    def to(self, compute: Compute = None, init_args=None,
+           get_if_exists=False, reload_prefixes=None):
+        """Deploy (or hot-reload) this module onto the compute.
+
+        get_if_exists=True: before launching, look for an existing service
+        under the fallback prefixes (username -> git branch -> prod, or an
+        explicit reload_prefixes list) and bind to it instead of deploying
+        (reference parity: Module.to get_if_exists)."""
         if compute is not None:
             self.compute = compute
         if init_args is not None:
             self.init_args = init_args
         if self.compute is None:
             self.compute = Compute(cpus=1)
+        if get_if_exists:
+            w = self._get_existing_service(reload_prefixes)
+            if w is not None:
+                # bind exactly to the found service name (bypasses the
+                # username prefixing in .name)
+                found = w["name"]
+                self._bound_full_name = found
+                self.launch_id = w.get("launch_id")
+                self.service_hosts = w.get("pods") or []
+                if not self.service_hosts:
+                    try:
+                        full = controller_client()._request(
+                            "GET",
+                            f"/controller/workload/{self.namespace}/{found}",
+                        ).json()
+                        self.service_hosts = full.get("pods") or []
+                    except Exception:
+                        pass
+                self._http = None
+                if self.stream_logs:
+                    print(f"[kt] reusing existing service {found}")
+                return self
         t0 = time.time()
         md = self.metadata()
         self._sync_workdir(md)

@@ -212,3 +212,30 @@ def test_launch_event_streaming(remote_fn):
     last = max(e["ts"] for e in evs)
     assert controller_client().service_events(
         remote_fn.name, remote_fn.namespace, since=last) == []
+
+
+def test_get_if_exists_reuses_service(remote_fn):
+    """A second client binds to the already-deployed service under the
+    fallback prefixes instead of launching a new one (reference:
+    Module.to(get_if_exists=...) with username -> branch -> prod order)."""
+    from kubetorch_amd.client.fn import fn as make_fn
+
+    f2 = make_fn(summer_mod.summer)
+    f2.to(kt.Compute(cpus=1), get_if_exists=True,
+          reload_prefixes=["nosuch", "citest"])
+    try:
+        assert f2.name == remote_fn.name  # bound, not re-prefixed
+        assert f2.service_hosts, "expected the existing service's pods"
+        assert f2(4, 5) == 9  # calls go to the existing pods
+    finally:
+        pass  # do NOT teardown: the service belongs to remote_fn
+
+    # no match under unknown prefixes -> falls through to a fresh deploy
+    f3 = make_fn(summer_mod.summer, name="fresh-one")
+    f3.to(kt.Compute(cpus=1), get_if_exists=True,
+          reload_prefixes=["nosuch"])
+    try:
+        assert f3.name != remote_fn.name
+        assert f3(1, 1) == 2
+    finally:
+        f3.teardown()
