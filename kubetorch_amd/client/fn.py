@@ -9,7 +9,7 @@ class Fn(Module):
     def __call__(self, *args, **kwargs):
         opts = {}
         for key in ("workers", "restart_procs", "stream_logs",
-                    "stream_metrics", "timeout", "serialization"):
+                    "stream_metrics", "timeout", "serialization", "debug"):
             if f"kt_{key}" in kwargs:
                 opts[key] = kwargs.pop(f"kt_{key}")
         return self._call(args, kwargs, **opts)

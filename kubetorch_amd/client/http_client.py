@@ -128,7 +128,8 @@ class HTTPClient:
 
     def call(self, args=(), kwargs=None, method=None, serialization="pickle",
              stream_logs=False, stream_metrics=False, timeout=None,
-             workers=None, restart_procs=False, request_id=None):
+             workers=None, restart_procs=False, request_id=None,
+             debug=False):
         rid = request_id or uuid.uuid4().hex
         url = f"{self.base_url}/call/{self.name}"
         if method:
@@ -140,6 +141,11 @@ class HTTPClient:
         if restart_procs:
             params["restart_procs"] = "true"
         headers = {"X-Request-ID": rid, "X-Serialization": serialization}
+        if debug:
+            # breakpoints in this call wait for `kt debug <service>`
+            headers["X-KT-Debug"] = "1"
+            print(f"[kt] debug call: breakpoints in {self.name} will wait "
+                  f"for `kt debug` on the pod's KT_DEBUG_PORT")
         if serialization == "pickle":
             body = {"body": base64.b64encode(
                 pickle.dumps((tuple(args), kwargs or {}))).decode()}
