@@ -12,7 +12,8 @@ SUPPORTED_TRAINING_JOBS = ("pytorchjob", "tfjob", "mxjob", "xgboostjob")
 
 
 def build_pod_spec(service_name, image, command=None, env=None, cpus=None,
-                   memory=None, gpus=0, gpu_type=None, shared_memory="8Gi",
+                   memory=None, gpus=0, gpu_type=None, disk_size=None,
+                   shared_memory="8Gi",
                    volumes=(), secrets=(), node_selector=None,
                    gpu_anti_affinity=True, port=C.SERVER_PORT,
                    inactivity_ttl=None):
@@ -25,6 +26,9 @@ def build_pod_spec(service_name, image, command=None, env=None, cpus=None,
     if gpus:
         resources["requests"][C.GPU_RESOURCE] = str(gpus)
         resources["limits"][C.GPU_RESOURCE] = str(gpus)
+    if disk_size:
+        resources["requests"]["ephemeral-storage"] = str(disk_size)
+        resources["limits"]["ephemeral-storage"] = str(disk_size)
 
     env_list = [{"name": k, "value": str(v)} for k, v in (env or {}).items()]
     env_list += [

@@ -15,6 +15,7 @@ from kubetorch_amd.provisioning import manifests as M
 
 class Compute:
     def __init__(self, cpus=None, memory=None, gpus=0, gpu_type=None,
+                 gpu_memory=None, disk_size=None,
                  image=None, env=None, namespace=None, volumes=None,
                  secrets=None, shared_memory="8Gi", node_selector=None,
                  annotations=None, inactivity_ttl=None, queue=None,
@@ -24,6 +25,12 @@ class Compute:
         self.memory = memory
         self.gpus = gpus
         self.gpu_type = gpu_type
+        # fractional-GPU memory cap: a whole amd.com/gpu is still requested,
+        # the annotation caps usable HBM (reference: gpu_memory annotation)
+        self.gpu_memory = gpu_memory
+        if gpu_memory and not gpus:
+            self.gpus = 1
+        self.disk_size = disk_size
         self.image = image  # Image object or image id string
         self.env = dict(env or {})
         self.namespace = namespace or config.namespace
@@ -103,12 +110,15 @@ class Compute:
         pod_kw = dict(
             env=self.pod_env(), cpus=self.cpus, memory=self.memory,
             gpus=self.gpus, gpu_type=self.gpu_type,
+            disk_size=self.disk_size,
             shared_memory=self.shared_memory, volumes=self.volumes,
             secrets=self.secrets, node_selector=self.node_selector,
         )
         ann = dict(self.annotations)
         if self.inactivity_ttl:
             ann[C.INACTIVITY_TTL_ANNOTATION] = str(self.inactivity_ttl)
+        if self.gpu_memory:
+            ann["gpu-memory"] = str(self.gpu_memory)
         if self.kind == "knative":
             return M.build_knative_manifest(
                 service_name, self.namespace, self.image_id(),

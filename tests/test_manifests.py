@@ -145,3 +145,16 @@ def test_config_layering(tmp_path, monkeypatch):
     monkeypatch.setenv("KT_NAMESPACE", "from-env")
     assert cfg.namespace == "from-env"
     assert cfg.username == "bob"
+
+
+def test_gpu_memory_and_disk_size():
+    """gpu_memory: whole amd.com/gpu requested + gpu-memory annotation;
+    disk_size -> ephemeral-storage resources (reference parity)."""
+    c = Compute(gpu_memory="64Gi", disk_size="50Gi")
+    assert c.gpus == 1  # whole GPU still requested
+    m = c.to_manifest("svc", username="u")
+    pod = m["spec"]["template"]["spec"]
+    res = pod["containers"][0]["resources"]
+    assert res["limits"]["amd.com/gpu"] == "1"
+    assert res["requests"]["ephemeral-storage"] == "50Gi"
+    assert m["metadata"]["annotations"]["gpu-memory"] == "64Gi"
