@@ -52,6 +52,24 @@ class App(Module):
             super()._wait_ready(timeout=timeout or 900, reloaded=reloaded)
 
 
+    def wait(self, timeout=900, poll=1.0):
+        """Foreground mode: block until the app's pods exit (reference:
+        App deploy --follow / _wait_for_app_exit). Returns True if the
+        app finished within the timeout."""
+        import time as _t
+
+        from kubetorch_amd.globals import controller_client
+
+        deadline = _t.time() + timeout
+        while _t.time() < deadline:
+            w = controller_client().get_workload(self.name, self.namespace)
+            pods = (w or {}).get("pods") or []
+            if not pods:
+                return True
+            _t.sleep(poll)
+        return False
+
+
 def app(command, name=None, health_path=None, port=None):
     return App(command, name=sanitize_name(name) if name else None,
                health_path=health_path, port=port)

@@ -239,3 +239,14 @@ def test_get_if_exists_reuses_service(remote_fn):
         assert f3(1, 1) == 2
     finally:
         f3.teardown()
+
+
+def test_app_mode_run_and_wait():
+    """App mode: the user command is the pod main process; wait() returns
+    once it exits (reference: kt run + foreground follow)."""
+    a = kt.app("sleep 1 && echo app-done", name="shortapp")
+    a.to(kt.Compute(cpus=1))
+    try:
+        assert a.wait(timeout=60), "app did not finish"
+    finally:
+        a.teardown()
