@@ -193,13 +193,26 @@ def logs(service: str, namespace: str = typer.Option(None, "-n"),
 
 @app.command()
 def run(command: str, name: str = typer.Option("app"),
-        cpus: str = typer.Option(None), gpus: int = typer.Option(0)):
+        cpus: str = typer.Option(None), gpus: int = typer.Option(0),
+        follow: bool = typer.Option(False, "--follow", "-f",
+                                    help="block until the app exits, then "
+                                         "tear it down"),
+        timeout: int = typer.Option(3600)):
     """Deploy an arbitrary command as an app: `kt run 'python serve.py'`."""
     import kubetorch_amd as kt
 
     a = kt.app(command, name=name)
     a.to(kt.Compute(cpus=cpus, gpus=gpus))
     console.print(f"[green]running[/green] {a.name}")
+    if follow:
+        done = a.wait(timeout=timeout)
+        for e in a.logs(limit=200):
+            console.print(e.get("line", ""))
+        a.teardown()
+        if not done:
+            console.print(f"[red]timed out after {timeout}s[/red]")
+            raise typer.Exit(1)
+        console.print("[green]app finished[/green]")
 
 
 @app.command()
