@@ -89,6 +89,7 @@ def deploy(target: str,
     if not partials:
         console.print("[red]no @kt.compute-decorated callables found[/red]")
         raise typer.Exit(1)
+    mods = []
     for name, pm in partials.items():
         if not isinstance(pm, PartialModule):
             console.print(f"[red]{name} is not decorated[/red]")
@@ -97,8 +98,22 @@ def deploy(target: str,
         if workers and m.compute.distributed_config:
             m.compute.distributed_config["workers"] = workers
             m.compute.replicas = workers
-        m.to(m.compute)
-        console.print(f"[green]deployed[/green] {m.name}")
+        mods.append(m)
+    # parallel deploy (reference: kt deploy launches modules concurrently)
+    from concurrent.futures import ThreadPoolExecutor
+
+    with ThreadPoolExecutor(max_workers=min(8, len(mods))) as ex:
+        futs = {ex.submit(m.to, m.compute): m for m in mods}
+        errs = []
+        for fut, m in futs.items():
+            try:
+                fut.result()
+                console.print(f"[green]deployed[/green] {m.name}")
+            except Exception as e:  # noqa: BLE001
+                errs.append((m.name, e))
+                console.print(f"[red]failed[/red] {m.name}: {e}")
+    if errs:
+        raise typer.Exit(1)
 
 
 @app.command()
