@@ -56,6 +56,19 @@ def check():
     from kubetorch_amd.ops import hip_available
 
     console.print(f"gfx950 HIP extension built: {hip_available()}")
+    import os as _os
+
+    store = _os.environ.get("KT_STORE_URL")
+    if store:
+        try:
+            import httpx
+
+            httpx.get(f"{store}/health", timeout=5).raise_for_status()
+            console.print(f"[green]data store ok[/green] at {store}")
+        except Exception as e:
+            console.print(f"[yellow]data store unreachable:[/yellow] {e}")
+    else:
+        console.print("data store: local-dir mode (KT_STORE_URL unset)")
 
 
 @app.command()
