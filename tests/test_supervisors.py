@@ -46,3 +46,37 @@ def test_tf_env_contract():
 def test_process_class_registry():
     for name in ("pytorch", "jax", "tensorflow", "spmd"):
         assert name in PROCESS_CLASSES
+
+
+def test_k8s_driver_event_parsing(monkeypatch):
+    """K8sDriver.get_events: kubectl JSON -> filtered, since-windowed event
+    dicts (the cluster path that CPU CI can't run live)."""
+    import json
+    import subprocess
+    from types import SimpleNamespace
+
+    from kubetorch_amd.controller.drivers import K8sDriver
+
+    payload = {"items": [
+        {"involvedObject": {"name": "svc-a-0"}, "type": "Normal",
+         "reason": "Scheduled", "message": "assigned node1",
+         "lastTimestamp": "2026-09-11T10:00:00Z"},
+        {"involvedObject": {"name": "svc-a-0"}, "type": "Warning",
+         "reason": "BackOff", "message": "crash loop",
+         "lastTimestamp": "2026-09-11T10:05:00Z"},
+        {"involvedObject": {"name": "other-svc-0"}, "type": "Normal",
+         "reason": "Scheduled", "message": "not ours",
+         "lastTimestamp": "2026-09-11T10:01:00Z"},
+    ]}
+
+    def fake_run(args, **kw):
+        return SimpleNamespace(returncode=0,
+                               stdout=json.dumps(payload).encode())
+
+    monkeypatch.setattr(subprocess, "run", fake_run)
+    evs = K8sDriver().get_events("svc-a", "ns")
+    assert [e["reason"] for e in evs] == ["Scheduled", "BackOff"]
+    assert all(e["pod"].startswith("svc-a") for e in evs)
+    # since-window: only events after the first timestamp
+    later = K8sDriver().get_events("svc-a", "ns", since=evs[0]["ts"])
+    assert [e["reason"] for e in later] == ["BackOff"]
