@@ -40,6 +40,22 @@ def llama3_8b(**overrides) -> LlamaConfig:
     return LlamaConfig(**overrides)
 
 
+def llama3_70b(**overrides) -> LlamaConfig:
+    d = dict(dim=8192, n_layers=80, n_heads=64, n_kv_heads=8,
+             intermediate=28672, vocab_size=128256, max_seq_len=8192,
+             rope_base=500000.0)
+    d.update(overrides)
+    return LlamaConfig(**d)
+
+
+def llama2_7b(**overrides) -> LlamaConfig:
+    d = dict(dim=4096, n_layers=32, n_heads=32, n_kv_heads=32,
+             intermediate=11008, vocab_size=32000, max_seq_len=4096,
+             rope_base=10000.0)
+    d.update(overrides)
+    return LlamaConfig(**d)
+
+
 def llama_tiny(**overrides) -> LlamaConfig:
     """Small config for tests / smoke (runs on CPU)."""
     d = dict(dim=256, n_layers=2, n_heads=4, n_kv_heads=2, intermediate=512,
