@@ -32,6 +32,8 @@ def main():
     ap.add_argument("--model", type=str, default="llama3-8b",
                     choices=["llama3-8b", "tiny"])
     ap.add_argument("--bucket-mb", type=int, default=256)
+    ap.add_argument("--zero", action="store_true",
+                    help="ZeRO-1 optimizer-state sharding")
     ap.add_argument("--lr", type=float, default=1e-4)
     # AOTriton's "efficient" kernels beat its flash kernels on gfx950 by
     # ~14% end-to-end (profiles/r01_sdpa_ab.md) -> default efficient.
@@ -71,7 +73,8 @@ def main():
         torch.set_default_dtype(prev_dtype)
     if args.ckpt:
         model.gradient_checkpointing_enable()
-    engine = FlatDDP(model, lr=args.lr, bucket_mb=args.bucket_mb)
+    engine = FlatDDP(model, lr=args.lr, bucket_mb=args.bucket_mb,
+                     zero=args.zero)
     engine.broadcast_params(src=0)
     log(f"[bench rank{rank}] model+engine ready in {time.time()-t_build:.1f}s "
         f"({sum(p.numel() for p in model.parameters())/1e9:.2f}B params)")
