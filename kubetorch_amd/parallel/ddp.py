@@ -343,6 +343,9 @@ class FlatDDP:
                 dist.broadcast(b.flat_param, src=src, group=self.pg)
 
     def state_dict(self):
+        # NOTE: with zero=True, m/v hold only this rank's shard — resume
+        # requires the same world size (save per-rank, or all-gather the
+        # state first for a world-size-independent checkpoint).
         return {
             "step": self.step_count,
             "m": [b.m for b in self.buckets],
