@@ -102,7 +102,9 @@ class FlatDDP:
         # all-gather the updated params. Cuts optimizer state memory by
         # world (8 GB/B-param -> 1 GB/B-param at 8 GPUs) at the same
         # collective byte count as all-reduce. Deferred-optimizer mode
-        # (bucket collectives still overlap backward).
+        # (bucket collectives still overlap backward). NOTE: shards are
+        # sized at construction — elastic re-join with a NEW world size
+        # needs a fresh engine (load params via load_state_dict).
         self.zero = zero and self._world > 1
         if clip_norm is not None or self.zero:
             overlap_optimizer = False
