@@ -6,12 +6,13 @@ FFN uses the same fused SwiGLU op as the dense MLP. Two placements:
   * dense (default): all experts on every rank — token-grouped batched
     expert GEMMs, DP-compatible as-is.
   * expert parallel (ep_group): experts sharded across the group's ranks.
-    This first implementation exchanges activations with
-    all_gather(tokens) -> local-expert compute -> all_reduce(outputs),
-    which is correct on any backend (gloo CI included) and per-link bound
-    on xGMI like every collective here; the all_to_all token exchange
-    (less traffic for top_k << n_experts/world) is a round-2 swap kept
-    behind the same interface.
+    The EP group is sequence-replicated (every rank in it runs the same
+    tokens — the TP-style placement; compose with DP ACROSS groups, not
+    inside one): each rank computes its local experts and the partial
+    outputs are all-reduce summed. Correct on any backend (gloo CI
+    included); the all_to_all token exchange for data-sharded EP (less
+    traffic when top_k << n_experts/world) is a round-2 swap kept behind
+    the same interface.
 
 The reference ships no model code — this extends the
 flagship family the launchers run, and is the EP workload the SPMD
@@ -90,8 +91,8 @@ class MoEMLP(nn.Module):
         return out
 
     def _dispatch_ep(self, flat, sel, weights):
-        """EP: every rank sees all tokens (routing is replicated — the
-        router weights are identical across the EP group), computes only
+        """Sequence-replicated EP: every rank in the group runs the same
+        tokens (router weights identical across the group), computes only
         its local experts, then the partial outputs are summed."""
         import torch.distributed as dist
 
