@@ -9,6 +9,9 @@ import sys
 
 import pytest
 
+# spawns worker subprocesses via the supervisor: retry under CI load spikes
+pytestmark = pytest.mark.flaky_retry
+
 ASSETS = os.path.join(os.path.dirname(__file__), "assets", "summer")
 
 
