@@ -276,7 +276,13 @@ def sync_workdir_from_store():
     root = os.environ.get("KT_WORKDIR_BASE", "/workdir")
     dest = os.path.join(root, key.strip("/").replace("/", "_"))
     try:
-        get(key, dest)
+        # many pods pulling the same workdir (reload fan-out) form the
+        # broadcast tree instead of all hitting the store
+        workers = int(os.environ.get("KT_NUM_WORKERS", "1"))
+        if workers >= 4 and _store_url() is not None:
+            get_broadcast(key, dest)
+        else:
+            get(key, dest)
     except KeyError:
         return None
     rel = os.environ.get("KT_REL_PATH")
