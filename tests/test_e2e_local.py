@@ -197,21 +197,26 @@ def test_inactivity_ttl_reaper():
         os.environ.pop("KT_TTL_GRACE", None)
 
 
-def test_launch_event_streaming(remote_fn):
+def test_launch_event_streaming():
     """Service events (Scheduled/Started) are queryable during/after launch
     and pod deaths surface as Warning events (reference: K8s launch-event
-    streaming in Module.to)."""
+    streaming in Module.to). Deploys its own service so the assertion is
+    independent of fixture age (earlier tests may restart the controller)."""
     from kubetorch_amd.globals import controller_client
 
-    evs = controller_client().service_events(remote_fn.name,
-                                             remote_fn.namespace)
-    reasons = [e["reason"] for e in evs]
-    assert "Scheduled" in reasons and "Started" in reasons, evs
-    assert any(e.get("pod") for e in evs)
-    # incremental polling: since=last ts returns nothing new
-    last = max(e["ts"] for e in evs)
-    assert controller_client().service_events(
-        remote_fn.name, remote_fn.namespace, since=last) == []
+    f = kt.fn(summer_mod.summer, name="evt-probe").to(kt.Compute(cpus=1))
+    try:
+        assert f(1, 2) == 3
+        evs = controller_client().service_events(f.name, f.namespace)
+        reasons = [e["reason"] for e in evs]
+        assert "Scheduled" in reasons and "Started" in reasons, evs
+        assert any(e.get("pod") for e in evs)
+        # incremental polling: since=last ts returns nothing new
+        last = max(e["ts"] for e in evs)
+        assert controller_client().service_events(
+            f.name, f.namespace, since=last) == []
+    finally:
+        f.teardown()
 
 
 def test_get_if_exists_reuses_service(remote_fn):
