@@ -130,6 +130,13 @@ class FlatDDP:
         # when grad_accum_steps changes).
         self._grad_scale = 1.0 / (self._world * grad_accum_steps)
         self._micro_step = 0
+        for mod in model.modules():
+            if getattr(mod, "ep_world", 1) > 1:
+                raise ValueError(
+                    "FlatDDP replicates every parameter across the group, "
+                    "which would average DIFFERENT experts under expert "
+                    "parallelism. Run EP layers with a per-rank optimizer "
+                    "(or DP across EP groups) — see models/moe.py.")
         params = [p for p in model.parameters() if p.requires_grad]
         if not params:
             raise ValueError("model has no trainable parameters")

@@ -109,3 +109,14 @@ def test_moe_expert_parallel_matches_dense():
     x = torch.randn(2, 8, cfg.dim)
     ref = dense.layers[0].mlp(x)
     torch.testing.assert_close(outs[0], ref, rtol=1e-5, atol=1e-5)
+
+
+def test_flatddp_rejects_ep_models():
+    """FlatDDP would average DIFFERENT experts under EP — must refuse."""
+    from kubetorch_amd.parallel import FlatDDP
+
+    cfg = llama_tiny(n_layers=1)
+    model = convert_to_moe(Llama(cfg), n_experts=4, top_k=2)
+    model.layers[0].mlp.ep_world = 2  # simulate an EP-sharded layer
+    with pytest.raises(ValueError, match="expert"):
+        FlatDDP(model, lr=1e-3, bucket_mb=1)
