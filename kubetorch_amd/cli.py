@@ -173,11 +173,22 @@ def describe(service: str, namespace: str = typer.Option(None, "-n")):
     from kubetorch_amd.config import config as cfg
     from kubetorch_amd.globals import controller_client
 
-    w = controller_client().get_workload(service, namespace or cfg.namespace)
+    ns = namespace or cfg.namespace
+    w = controller_client().get_workload(service, ns)
     if not w:
         console.print("[red]not found[/red]")
         raise typer.Exit(1)
     console.print_json(json.dumps(w, default=str))
+    try:
+        evs = controller_client().service_events(service, ns)[-10:]
+    except Exception:
+        evs = []
+    if evs:
+        console.print("\n[bold]recent events[/bold]")
+        for e in evs:
+            tag = "!" if e.get("type") == "Warning" else "·"
+            console.print(f"  {tag} {e.get('reason', '')} "
+                          f"{e.get('pod') or ''}: {e.get('message', '')}")
 
 
 @app.command()
