@@ -2,6 +2,8 @@
 import os
 
 import pytest
+
+pytestmark = pytest.mark.flaky_retry
 import torch
 
 from kubetorch_amd.models import Llama, llama_tiny
