@@ -310,3 +310,23 @@ def test_batched_generator_continuous_admission():
     rid = eng2.submit([5, 6, 7], max_new_tokens=8, stop_token=first)
     out2 = eng2.run()
     assert out2[rid][-1] == first and len(out2[rid]) == 4
+
+
+def test_tokenizer_sentencepiece_roundtrip(tmp_path):
+    import sentencepiece as spm
+
+    corpus = tmp_path / "corpus.txt"
+    corpus.write_text("\n".join(
+        ["the quick brown fox jumps over the lazy dog",
+         "pack my box with five dozen liquor jugs"] * 30))
+    spm.SentencePieceTrainer.Train(
+        input=str(corpus), model_prefix=str(tmp_path / "tokenizer"),
+        vocab_size=80, model_type="bpe")
+
+    from kubetorch_amd.models.tokenizer import Tokenizer
+
+    t = Tokenizer.load(str(tmp_path / "tokenizer.model"))
+    assert t.kind == "sp"
+    ids = t.encode("the quick brown fox", add_bos=True)
+    assert ids[0] == t.bos_id
+    assert "quick" in t.decode(ids)
