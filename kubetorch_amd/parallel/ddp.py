@@ -49,7 +49,8 @@ def init_distributed(backend=None, timeout_s=300):
 
 class _Bucket:
     __slots__ = ("params", "flat_param", "flat_grad", "m", "v", "offsets",
-                 "pending", "work", "numel", "completions", "shard_grad")
+                 "pending", "work", "numel", "completions", "shard_grad",
+                 "ema")
 
     def __init__(self):
         self.params: List[torch.nn.Parameter] = []
@@ -79,7 +80,7 @@ class FlatDDP:
     def __init__(self, model, lr=3e-4, betas=(0.9, 0.95), eps=1e-8,
                  weight_decay=0.1, bucket_mb=256, process_group=None,
                  overlap_optimizer=True, grad_accum_steps=1, clip_norm=None,
-                 zero=False):
+                 zero=False, ema_decay=None):
         self.model = model
         self.lr = lr
         self.betas = betas
@@ -108,6 +109,9 @@ class FlatDDP:
         self.zero = zero and self._world > 1
         if clip_norm is not None or self.zero:
             overlap_optimizer = False
+        # EMA of the weights (fp32, one flat buffer per bucket): updated
+        # after every optimizer step; read back via ema_state_dict().
+        self.ema_decay = ema_decay
         # Overlap mode: fused AdamW for a bucket is launched on a side HIP
         # stream as soon as the bucket's grads are final (and all-reduced),
         # running concurrently with the remaining backward. Safe because a
@@ -179,6 +183,8 @@ class FlatDDP:
             if self.zero:
                 b.shard_grad = torch.zeros(b.numel // self._world,
                                            dtype=dtype, device=dev)
+            if self.ema_decay is not None:
+                b.ema = None  # lazily initialized from the first params
             for p, off in zip(b.params, b.offsets):
                 n = p.numel()
                 b.flat_param[off:off + n].copy_(p.data.reshape(-1))
@@ -277,6 +283,7 @@ class FlatDDP:
             if self.zero:
                 self._zero_step(grad_scale)
                 self._reset_buckets_after_step()
+                self._update_ema()
                 return
             if self.clip_norm is not None:
                 # flat grads hold the SUM over ranks (and micro-batches);
@@ -299,6 +306,30 @@ class FlatDDP:
             if not self._copy_mode:
                 b.flat_grad.zero_()
             b.pending = len(b.params)
+        self._update_ema()
+
+    def _update_ema(self):
+        if self.ema_decay is None:
+            return
+        d = self.ema_decay
+        for b in self.buckets:
+            if b.ema is None:
+                b.ema = b.flat_param.float()  # first step: copy
+            else:
+                b.ema.mul_(d).add_(b.flat_param.float(), alpha=1.0 - d)
+
+    @torch.no_grad()
+    def ema_state_dict(self):
+        """EMA weights as {param_name: fp32 tensor} (model's names)."""
+        if self.ema_decay is None:
+            raise ValueError("engine built without ema_decay")
+        name_of = {p: n for n, p in self.model.named_parameters()}
+        out = {}
+        for b in self.buckets:
+            ema = b.ema if b.ema is not None else b.flat_param.float()
+            for p, off in zip(b.params, b.offsets):
+                out[name_of[p]] = ema[off:off + p.numel()].view(p.shape).clone()
+        return out
 
     def _reset_buckets_after_step(self):
         for b in self.buckets:

@@ -265,3 +265,32 @@ def test_zero1_two_ranks_match_full_optimizer():
     n = min(results[0].numel(), single.numel())
     torch.testing.assert_close(results[0][:n], single[:n], rtol=2e-2,
                                atol=2e-2)
+
+
+def test_ema_tracking():
+    """EMA buffers follow d*ema + (1-d)*param after every step; the state
+    dict maps back to model parameter names/shapes."""
+    from kubetorch_amd.models import Llama, llama_tiny
+    from kubetorch_amd.parallel import FlatDDP
+
+    torch.manual_seed(4)
+    m = Llama(llama_tiny())
+    e = FlatDDP(m, lr=1e-2, bucket_mb=4, ema_decay=0.5)
+    x = torch.randint(0, 512, (2, 32))
+    y = torch.randint(0, 512, (2, 32))
+
+    m.loss(x, y).backward()
+    e.step()
+    p1 = {n: p.float().clone() for n, p in m.named_parameters()}
+    ema1 = e.ema_state_dict()
+    for n in p1:  # first step: ema == params
+        torch.testing.assert_close(ema1[n], p1[n])
+
+    m.loss(x, y).backward()
+    e.step()
+    p2 = {n: p.float().clone() for n, p in m.named_parameters()}
+    ema2 = e.ema_state_dict()
+    for n in p1:
+        torch.testing.assert_close(ema2[n], 0.5 * p1[n] + 0.5 * p2[n],
+                                   rtol=1e-5, atol=1e-6)
+    assert ema2[n].shape == p2[n].shape
