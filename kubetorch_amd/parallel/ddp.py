@@ -314,7 +314,8 @@ class FlatDDP:
         d = self.ema_decay
         for b in self.buckets:
             if b.ema is None:
-                b.ema = b.flat_param.float()  # first step: copy
+                # copy=True: .float() on fp32 params would ALIAS them
+                b.ema = b.flat_param.to(torch.float32, copy=True)
             else:
                 b.ema.mul_(d).add_(b.flat_param.float(), alpha=1.0 - d)
 
