@@ -155,9 +155,17 @@ class Attention(nn.Module):
             kf, vf = cache.append(layer, k, v)
             if cache.pos == 0:
                 o = self._sdpa(q, kf, vf, causal=True)
-            else:
-                assert S == 1, "chunked decode after prefill not supported"
+            elif S == 1:
                 o = self._sdpa(q, kf, vf, causal=False)
+            else:
+                # chunked prefill: query row i (position pos+i) attends
+                # keys [0, pos+i] — causal with a column offset
+                P = cache.pos
+                L = P + S
+                ar = torch.arange(L, device=x.device)
+                mask = (ar.view(1, L) <= (P + torch.arange(
+                    S, device=x.device)).view(S, 1)).view(1, 1, S, L)
+                o = self._sdpa_masked(q, kf, vf, mask)
             return self.wo(o.transpose(1, 2).reshape(B, S, -1))
         attn_impl = os.environ.get("KT_ATTN", "ck")
         # The v3 FMHA kernel wants Q pre-scaled by softmax_scale*log2e (its
@@ -322,12 +330,13 @@ class Llama(nn.Module):
 
     @torch.no_grad()
     def generate(self, tokens, max_new_tokens, temperature=0.0, top_k=None,
-                 stop_token=None, max_len=None):
-        """Autoregressive decode with a preallocated KV cache: one prefill
-        chunk, then single-token steps. temperature=0 is greedy argmax.
-        Returns [B, prompt+new] token ids. (The reference ships no model
-        code — this is the serving half of the flagship workload, the
-        training half being bench.py's DDP step.)"""
+                 stop_token=None, max_len=None, prefill_chunk=None):
+        """Autoregressive decode with a preallocated KV cache: prefill
+        (optionally in prefill_chunk-token pieces to bound latency spikes
+        on long prompts), then single-token steps. temperature=0 is greedy
+        argmax. Returns [B, prompt+new] token ids. (The reference ships no
+        model code — this is the serving half of the flagship workload,
+        the training half being bench.py's DDP step.)"""
         B, S0 = tokens.shape
         dev = tokens.device
         dtype = self.embed.weight.dtype
@@ -335,7 +344,12 @@ class Llama(nn.Module):
                                                     S0 + max_new_tokens),
                         dev, dtype)
         out = [tokens]
-        logits = self._forward_cached(tokens, cache)
+        if prefill_chunk and prefill_chunk < S0:
+            for s in range(0, S0, prefill_chunk):
+                logits = self._forward_cached(
+                    tokens[:, s:s + prefill_chunk], cache)
+        else:
+            logits = self._forward_cached(tokens, cache)
         for _ in range(max_new_tokens):
             if temperature > 0:
                 lg = logits.float() / temperature

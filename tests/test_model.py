@@ -330,3 +330,15 @@ def test_tokenizer_sentencepiece_roundtrip(tmp_path):
     ids = t.encode("the quick brown fox", add_bos=True)
     assert ids[0] == t.bos_id
     assert "quick" in t.decode(ids)
+
+
+def test_chunked_prefill_matches_full():
+    """Prefilling the prompt in 3-token chunks must produce the same greedy
+    output as the single-chunk prefill (offset-causal mask path)."""
+    torch.manual_seed(41)
+    cfg = llama_tiny()
+    model = Llama(cfg).eval()
+    prompt = torch.randint(0, cfg.vocab_size, (2, 10))
+    full = model.generate(prompt, max_new_tokens=6)
+    chunked = model.generate(prompt, max_new_tokens=6, prefill_chunk=3)
+    assert (full == chunked).all(), (full, chunked)
