@@ -31,11 +31,13 @@ class _Request:
 
 
 class BatchedGenerator:
-    def __init__(self, model, max_batch=8, max_len=None, device=None):
+    def __init__(self, model, max_batch=8, max_len=None, device=None,
+                 prefill_chunk=None):
         self.model = model
         cfg = model.cfg
         self.cfg = cfg
         self.max_batch = max_batch
+        self.prefill_chunk = prefill_chunk  # bound prefill latency spikes
         self.max_len = max_len or cfg.max_seq_len
         p = next(model.parameters())
         self.device = device or p.device
@@ -97,7 +99,13 @@ class BatchedGenerator:
 
         S0 = req.prompt.numel()
         cache = KVCache(self.cfg, 1, S0, self.device, self.dtype)
-        logits = self.model._forward_cached(req.prompt.view(1, -1), cache)
+        pc = self.prefill_chunk
+        if pc and pc < S0:
+            for s0 in range(0, S0, pc):
+                logits = self.model._forward_cached(
+                    req.prompt[s0:s0 + pc].view(1, -1), cache)
+        else:
+            logits = self.model._forward_cached(req.prompt.view(1, -1), cache)
         for i in range(self.cfg.n_layers):
             self.k[i][req.slot, :, :S0] = cache.k[i][0, :, :S0]
             self.v[i][req.slot, :, :S0] = cache.v[i][0, :, :S0]

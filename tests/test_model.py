@@ -342,3 +342,16 @@ def test_chunked_prefill_matches_full():
     full = model.generate(prompt, max_new_tokens=6)
     chunked = model.generate(prompt, max_new_tokens=6, prefill_chunk=3)
     assert (full == chunked).all(), (full, chunked)
+
+
+def test_batched_generator_chunked_prefill():
+    from kubetorch_amd.models.serving import BatchedGenerator
+
+    torch.manual_seed(43)
+    cfg = llama_tiny()
+    model = Llama(cfg).eval()
+    eng = BatchedGenerator(model, max_batch=2, max_len=64, prefill_chunk=4)
+    rid = eng.submit(list(range(1, 11)), max_new_tokens=5)
+    out = eng.run()
+    ref = model.generate(torch.tensor([list(range(1, 11))]), 5)[0].tolist()
+    assert out[rid] == ref
