@@ -101,3 +101,31 @@ def test_native_spill_roundtrip_gpu(tmp_path):
     for b, fp, m in zip(eng.buckets, before, before_m):
         assert torch.equal(b.flat_param, fp)
         assert torch.equal(b.m, m)
+
+
+def test_profile_step_logs_table(capsys):
+    """profile_step wraps a block and emits a kernel/op table + optional
+    chrome trace through the log stream."""
+    import torch
+
+    from kubetorch_amd.utils.profiling import profile_step
+
+    lines = []
+    with profile_step("unit", top=5, printer=lines.append):
+        a = torch.randn(64, 64)
+        (a @ a).sum().backward if False else (a @ a).sum()
+    out = "\n".join(lines)
+    assert "[kt-profile] unit:" in out
+    assert "Self CPU" in out  # the profiler table rendered
+
+
+def test_profile_step_chrome_trace(tmp_path):
+    import torch
+
+    from kubetorch_amd.utils.profiling import profile_step
+
+    lines = []
+    with profile_step("tr", trace_dir=str(tmp_path), printer=lines.append):
+        torch.randn(8, 8) @ torch.randn(8, 8)
+    traces = list(tmp_path.glob("tr_*.json"))
+    assert traces and traces[0].stat().st_size > 0
