@@ -42,6 +42,9 @@ def main():
     ap.add_argument("--ckpt", action="store_true",
                     help="per-layer activation checkpointing (memory for "
                          "compute; off for the headline number)")
+    ap.add_argument("--profile", action="store_true",
+                    help="profile one post-warmup step (top-kernel table "
+                         "to stderr; excluded from the timed region)")
     args = ap.parse_args()
 
     from kubetorch_amd.models import Llama, llama3_8b, llama_tiny
@@ -115,6 +118,17 @@ def main():
     for i in range(args.warmup):
         loss = one_step()
         log(f"[bench rank{rank}] warmup {i}: loss={loss.item():.4f}")
+
+    if args.profile:
+        # every rank takes the profiled step (collectives stay matched);
+        # only rank 0 prints the table
+        from kubetorch_amd.utils.profiling import profile_step
+
+        printer = log if rank == 0 else (lambda *_: None)
+        with profile_step("bench_step", top=20, printer=printer):
+            one_step()
+        if dev.type == "cuda":
+            torch.cuda.synchronize(dev)
 
     barrier_sync()
     t0 = time.perf_counter()
