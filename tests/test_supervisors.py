@@ -80,3 +80,37 @@ def test_k8s_driver_event_parsing(monkeypatch):
     # since-window: only events after the first timestamp
     later = K8sDriver().get_events("svc-a", "ns", since=evs[0]["ts"])
     assert [e["reason"] for e in later] == ["BackOff"]
+
+
+def test_pod_ips_local_ips_env(monkeypatch):
+    """pod_ips honors the KT_LOCAL_IPS fake-cluster contract and quorum."""
+    monkeypatch.setenv("KT_LOCAL_IPS", "127.0.0.1:1,127.0.0.1:2,127.0.0.1:3")
+    from kubetorch_amd.serving.discovery import pod_ips
+
+    peers = pod_ips(num_workers=3, timeout=5)
+    assert len(peers) == 3 and peers[0].startswith("127.0.0.1")
+
+
+def test_service_url_resolution(monkeypatch):
+    from kubetorch_amd.globals import service_url
+
+    assert service_url("svc", "ns", ["1.2.3.4:32300"]) == "http://1.2.3.4:32300"
+    monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    assert "svc.ns.svc.cluster.local" in service_url("svc", "ns")
+    monkeypatch.delenv("KUBERNETES_SERVICE_HOST")
+    assert service_url("svc", "ns").startswith("http://svc.ns:")
+
+
+def test_endpoint_and_stock_images():
+    from kubetorch_amd.resources import images
+    from kubetorch_amd.resources.endpoint import Endpoint
+
+    e = Endpoint(url="http://my-router:9000")
+    assert e.resolve(default_url="http://x") == "http://my-router:9000"
+    e2 = Endpoint(selector={"role": "head"}, port=8265)
+    assert e2.resolve(default_url="http://x") == "http://x"
+    assert e2.selector == {"role": "head"}
+
+    img = images.pytorch()
+    assert "rocm" in img.image_id.lower()
+    assert images.ray().image_id.startswith("rayproject/")
