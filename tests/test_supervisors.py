@@ -114,3 +114,32 @@ def test_endpoint_and_stock_images():
     img = images.pytorch()
     assert "rocm" in img.image_id.lower()
     assert images.ray().image_id.startswith("rayproject/")
+
+
+def test_inactivity_ttl_env_parse(monkeypatch):
+    from kubetorch_amd.serving.metrics import inactivity_ttl_seconds
+
+    for raw, want in (("120s", 120), ("5m", 300), ("2h", 7200),
+                      ("1d", 86400), ("90", 90), ("bogus", None)):
+        monkeypatch.setenv("KT_INACTIVITY_TTL", raw)
+        assert inactivity_ttl_seconds() == want, raw
+    monkeypatch.delenv("KT_INACTIVITY_TTL")
+    assert inactivity_ttl_seconds() is None
+
+
+def test_ringlog_ring_and_filters():
+    from kubetorch_amd.serving.log_capture import RingLog
+
+    r = RingLog(size=5)
+    for i in range(8):
+        r.append(f"line{i}", request_id="r1" if i % 2 else None)
+    entries = r.tail()
+    assert len(entries) == 5  # ring keeps the last `size`
+    assert entries[0]["line"] == "line3"
+    # seq windowing + request-id filtering
+    only_r1 = r.tail(request_id="r1")
+    assert all(e["request_id"] == "r1" for e in only_r1)
+    last_seq = entries[-1]["seq"]
+    assert r.tail(since=last_seq)[0]["line"] == "line7"
+    # wait_for returns immediately when new entries already exist
+    assert r.wait_for(since=0, timeout=0.1)
