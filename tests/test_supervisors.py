@@ -143,3 +143,29 @@ def test_ringlog_ring_and_filters():
     assert r.tail(since=last_seq)[0]["line"] == "line7"
     # wait_for returns immediately when new entries already exist
     assert r.wait_for(since=0, timeout=0.1)
+
+
+def test_extract_pointers_and_exception_roundtrip():
+    from tests.assets.summer import summer as summer_mod
+
+    from kubetorch_amd.client.pointers import extract_pointers
+    from kubetorch_amd.exceptions import (package_exception,
+                                          reconstruct_exception)
+
+    ptr = extract_pointers(summer_mod.summer)
+    assert ptr["name"] == "summer"
+    assert ptr["file_path"].endswith("summer.py")
+    assert ptr["project_root"]
+
+    # exception packaging -> reconstruction keeps the real class + traceback
+    try:
+        raise ValueError("kaboom 42")
+    except ValueError as e:
+        payload = package_exception(e)
+    exc = reconstruct_exception(payload)
+    assert isinstance(exc, ValueError) and "kaboom 42" in str(exc)
+    assert "ValueError" in exc.remote_traceback
+    # chained hop: re-packaging preserves the original remote traceback
+    payload2 = package_exception(exc)
+    exc2 = reconstruct_exception(payload2)
+    assert "kaboom 42" in exc2.remote_traceback
