@@ -169,3 +169,20 @@ def test_extract_pointers_and_exception_roundtrip():
     payload2 = package_exception(exc)
     exc2 = reconstruct_exception(payload2)
     assert "kaboom 42" in exc2.remote_traceback
+
+
+def test_secret_provider_presets_complete():
+    """All 14 reference provider presets exist and render env keys."""
+    from kubetorch_amd.resources.secret import PROVIDERS, secret_factory
+
+    expected = {"anthropic", "aws", "azure", "cohere", "gcp", "github",
+                "huggingface", "kubeconfig", "lambda", "langchain",
+                "openai", "pinecone", "ssh", "wandb"}
+    assert expected <= set(PROVIDERS), expected - set(PROVIDERS)
+    s = secret_factory("openai", values={"OPENAI_API_KEY": "sk-x"})
+    m = s.to_manifest("ns")
+    assert m["kind"] == "Secret"
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="known"):
+        secret_factory("not-a-provider")
