@@ -11,6 +11,19 @@ from rich.console import Console
 from rich.table import Table
 
 app = typer.Typer(help="kubetorch_amd: MI355X-native serverless ML dispatch")
+
+
+@app.callback(invoke_without_command=True)
+def _root(ctx: typer.Context,
+          version: bool = typer.Option(False, "--version", "-V")):
+    if version:
+        import kubetorch_amd
+
+        console.print(f"kubetorch_amd {kubetorch_amd.__version__}")
+        raise typer.Exit()
+    if ctx.invoked_subcommand is None:
+        console.print(ctx.get_help())
+        raise typer.Exit()
 console = Console()
 
 secrets_app = typer.Typer(help="manage secrets")
