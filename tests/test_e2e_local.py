@@ -255,3 +255,20 @@ def test_app_mode_run_and_wait():
         assert a.wait(timeout=60), "app did not finish"
     finally:
         a.teardown()
+
+
+def test_examples_run():
+    """The shipped examples stay runnable (01 hello, 05 serving decode)."""
+    import subprocess
+
+    env = dict(os.environ, KT_LOCAL_MODE="true", KT_USERNAME="exs",
+               PYTHONPATH=os.path.dirname(os.path.dirname(
+                   os.path.abspath(__file__))))
+    root = env["PYTHONPATH"]
+    for ex, needle in (("01_hello_world.py", "hello"),
+                       ("05_serving_decode.py", "generated:")):
+        r = subprocess.run(
+            [sys.executable, os.path.join(root, "examples", ex)],
+            env=env, capture_output=True, text=True, timeout=240)
+        assert r.returncode == 0, (ex, r.stdout[-500:], r.stderr[-500:])
+        assert needle in r.stdout, (ex, r.stdout[-300:])
