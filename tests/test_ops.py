@@ -448,3 +448,23 @@ def test_attn_bwd_ck_gqa_native():
     _assert_close(dq, qr.grad, msg="ck bwd dq")
     _assert_close(dk, kr.grad, msg="ck bwd dk")
     _assert_close(dv, vr.grad, msg="ck bwd dv")
+
+
+def test_attention_support_gates():
+    """Shape gates that pick the attention kernel path (pure logic)."""
+    dev_cpu = torch.device("cpu")
+    # v3 requires D=128, S%256==0, cuda+bf16 (+ext). On CPU: always False.
+    assert not ops.flash_attention_v3_supported(4096, 128, dev_cpu, BF16)
+    assert not ops.flash_attention_v3_supported(4096, 64, dev_cpu, BF16)
+    # the gate math itself (device/dtype aside): S%256 and D==128
+    import kubetorch_amd.ops as o
+
+    class FakeDev:
+        type = "cuda"
+
+    if o.hip_available():  # on a GPU box the gate goes live
+        assert o.flash_attention_v3_supported(512, 128, FakeDev, BF16)
+        assert not o.flash_attention_v3_supported(384, 128, FakeDev, BF16)
+        assert not o.flash_attention_v3_supported(512, 64, FakeDev, BF16)
+    q = torch.randn(1, 2, 128, 128, dtype=BF16)
+    assert not ops.flash_attention_supported(q, q, q, True)  # cpu tensor
