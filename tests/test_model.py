@@ -355,3 +355,19 @@ def test_batched_generator_chunked_prefill():
     out = eng.run()
     ref = model.generate(torch.tensor([list(range(1, 11))]), 5)[0].tolist()
     assert out[rid] == ref
+
+
+def test_kv_cache_overflow_and_submit_validation():
+    from kubetorch_amd.models import KVCache
+    from kubetorch_amd.models.serving import BatchedGenerator
+
+    cfg = llama_tiny()
+    cache = KVCache(cfg, 1, 8, torch.device("cpu"), torch.float32)
+    model = Llama(cfg)
+    model._forward_cached(torch.randint(0, cfg.vocab_size, (1, 8)), cache)
+    with pytest.raises(RuntimeError, match="overflow"):
+        model._forward_cached(torch.randint(0, cfg.vocab_size, (1, 1)), cache)
+
+    eng = BatchedGenerator(model, max_batch=1, max_len=16)
+    with pytest.raises(ValueError, match="exceeds"):
+        eng.submit(list(range(10)), max_new_tokens=10)
