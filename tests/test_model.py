@@ -371,3 +371,27 @@ def test_kv_cache_overflow_and_submit_validation():
     eng = BatchedGenerator(model, max_batch=1, max_len=16)
     with pytest.raises(ValueError, match="exceeds"):
         eng.submit(list(range(10)), max_new_tokens=10)
+
+
+def test_hf_convert_rejects_mismatched_dict(tmp_path):
+    import json
+
+    from safetensors.torch import save_file
+
+    from kubetorch_amd.models import convert
+
+    cfg = llama_tiny()
+    src = Llama(cfg)
+    hf_sd = convert.kt_to_hf_state_dict(src.state_dict(), cfg)
+    bad = {k: v for k, v in hf_sd.items()
+           if "layers.1" not in k}  # drop a whole layer
+    save_file({k: v.contiguous() for k, v in bad.items()},
+              str(tmp_path / "model.safetensors"))
+    (tmp_path / "config.json").write_text(json.dumps({
+        "hidden_size": cfg.dim, "num_hidden_layers": cfg.n_layers,
+        "num_attention_heads": cfg.n_heads,
+        "num_key_value_heads": cfg.n_kv_heads,
+        "intermediate_size": cfg.intermediate,
+        "vocab_size": cfg.vocab_size}))
+    with pytest.raises(KeyError):
+        convert.load_hf_checkpoint(str(tmp_path), dtype=torch.float32)
