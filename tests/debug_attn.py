@@ -1,6 +1,7 @@
 """GPU debug for the custom attention fwd: numerics vs torch (fwd+bwd via
 the aten efficient-attention kernels), LSE check, perf timing.
 PYTHONPATH=. python tests/debug_attn.py"""
+import sys
 import time
 
 import torch
@@ -125,3 +126,45 @@ if __name__ == "__main__":
     perf(4, 32, 8, 4096)
     perf(1, 32, 8, 2048)
     print("done")
+
+
+def bwd_mask_ab():
+    """Round-2 kickoff: sweep the CK bwd mask conventions in ONE gpu call.
+    Run: python tests/debug_attn.py bwd"""
+    import os
+
+    import torch.nn.functional as F
+
+    torch.manual_seed(9)
+    B, Hq, Hkv, S, D = 2, 8, 2, 256, 128
+    scale = D ** -0.5
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda")
+    o, lse = ops._ext().attn_fwd_ck_tr(q, k, v, scale)
+    gout = torch.randn_like(o)
+    g = Hq // Hkv
+    qr = q.float().requires_grad_(True)
+    kr = k.float().requires_grad_(True)
+    vr = v.float().requires_grad_(True)
+    outr = F.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(g, 1), vr.repeat_interleave(g, 1),
+        is_causal=True, scale=scale)
+    outr.backward(gout.float())
+    for mode in ("0", "1", "2"):
+        os.environ["KT_CKBWD_MASK"] = mode
+        dq, dk_e, dv_e = ops._ext().attn_bwd_ck(
+            gout, q, k, v, o.contiguous(), lse.contiguous(), scale)
+        dk = dk_e.view(B, Hkv, g, S, D).sum(2)
+        dv = dv_e.view(B, Hkv, g, S, D).sum(2)
+        print(f"mask_mode={mode}: "
+              f"dq {(dq.float()-qr.grad).abs().max():.3e} "
+              f"dk {(dk.float()-kr.grad).abs().max():.3e} "
+              f"dv {(dv.float()-vr.grad).abs().max():.3e} "
+              f"(dq row0 {dq[0,0,0].abs().max():.3e} "
+              f"rowN {dq[0,0,-1].abs().max():.3e})")
+    os.environ.pop("KT_CKBWD_MASK", None)
+
+
+if len(sys.argv) > 1 and sys.argv[1] == "bwd":
+    bwd_mask_ab()
