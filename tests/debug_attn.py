@@ -118,7 +118,9 @@ def perf(B, Hq, Hkv, S, iters=20):
           f"aten {flops/t_torch/1e12:.0f} TF")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "bwd":
+    pass  # skip the fwd suite; bwd_mask_ab runs below
+elif __name__ == "__main__":
     check(1, 2, 2, 256, "tiny MHA")
     check(1, 4, 2, 384, "small GQA")
     check(2, 8, 2, 1024, "mid GQA")
@@ -166,5 +168,5 @@ def bwd_mask_ab():
     os.environ.pop("KT_CKBWD_MASK", None)
 
 
-if len(sys.argv) > 1 and sys.argv[1] == "bwd":
+if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "bwd":
     bwd_mask_ab()
