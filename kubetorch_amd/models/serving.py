@@ -160,8 +160,8 @@ class BatchedGenerator:
             # (advanced indexing: [idx, :, lens] -> [B, n_kv, hd] rows)
             self.k[i][idx, :, lens] = k[:, 0]
             self.v[i][idx, :, lens] = v[:, 0]
-            kf = self.k[i][idx][:, :, :L]
-            vf = self.v[i][idx][:, :, :L]
+            kf = self.k[i][idx, :, :L]
+            vf = self.v[i][idx, :, :L]
             o = layer.attn._sdpa_masked(q.transpose(1, 2), kf, vf, mask)
             h = h + layer.attn.wo(o.transpose(1, 2).reshape(B, 1, -1))
             n2 = ops.rmsnorm(h, layer.mlp_norm.weight, eps)
