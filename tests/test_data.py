@@ -75,3 +75,26 @@ def test_synthetic_tokens_and_train_smoke():
     # smoke: training stepped over fresh shuffled batches without blowup
     # (monotone decrease needs repeated data — covered in test_model)
     assert len(losses) == 4 and all(np.isfinite(losses))
+
+
+def test_loader_errors_and_edges():
+    from kubetorch_amd.data import ShardedLoader, TokenDataset
+
+    with np.testing.assert_raises(ValueError):
+        TokenDataset(torch.arange(5), seq_len=10)  # too short
+    ds = TokenDataset(torch.arange(65), 8)
+    with np.testing.assert_raises(ValueError):
+        ShardedLoader(ds, batch=100)  # not enough for one batch
+    # 2-D input is flattened
+    ds2 = TokenDataset(torch.arange(64).reshape(8, 8), 4)
+    x, y = ds2.sample(0)
+    assert x.tolist() == [0, 1, 2, 3]
+
+
+def test_tokenizer_load_missing(tmp_path):
+    import pytest as _pytest
+
+    from kubetorch_amd.models.tokenizer import Tokenizer
+
+    with _pytest.raises(FileNotFoundError):
+        Tokenizer.load(str(tmp_path))  # dir without artifacts
