@@ -36,7 +36,7 @@ def main():
                     help="ZeRO-1 optimizer-state sharding")
     ap.add_argument("--lr", type=float, default=1e-4)
     # AOTriton's "efficient" kernels beat its flash kernels on gfx950 by
-    # ~14% end-to-end (profiles/r01_sdpa_ab.md) -> default efficient.
+    # ~14% end-to-end (profiles/r01_step_profile.md) -> default efficient.
     ap.add_argument("--sdpa", type=str, default="efficient",
                     choices=["flash", "efficient", "math"])
     ap.add_argument("--ckpt", action="store_true",
