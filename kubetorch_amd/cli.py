@@ -229,7 +229,11 @@ def teardown(service: str = typer.Argument(None),
 
 @app.command()
 def logs(service: str, namespace: str = typer.Option(None, "-n"),
-         limit: int = typer.Option(100)):
+         limit: int = typer.Option(100),
+         follow: bool = typer.Option(False, "--follow", "-f",
+                                     help="keep polling for new lines")):
+    import time as _time
+
     from kubetorch_amd.client.http_client import HTTPClient
     from kubetorch_amd.config import config as cfg
     from kubetorch_amd.globals import controller_client, service_url
@@ -239,8 +243,15 @@ def logs(service: str, namespace: str = typer.Option(None, "-n"),
     if not w:
         raise typer.Exit(1)
     client = HTTPClient(service_url(service, ns, w.get("pods")), service)
-    for e in client.logs(limit=limit):
-        console.print(f"[dim]{e['source']}[/dim] {e['line']}")
+    since = 0
+    while True:
+        entries = client.logs(since=since, limit=limit)
+        for e in entries:
+            console.print(f"[dim]{e['source']}[/dim] {e['line']}")
+            since = max(since, e.get("seq", since) + 1)
+        if not follow:
+            break
+        _time.sleep(1.0)
 
 
 @app.command()
