@@ -120,3 +120,21 @@ def test_flatddp_rejects_ep_models():
     model.layers[0].mlp.ep_world = 2  # simulate an EP-sharded layer
     with pytest.raises(ValueError, match="expert"):
         FlatDDP(model, lr=1e-3, bucket_mb=1)
+
+
+def test_moe_model_generates():
+    """MoE MLPs work through the KV-cache decode path (generate) and the
+    batching engine (layer.mlp is called identically)."""
+    torch.manual_seed(2)
+    cfg = llama_tiny()
+    model = convert_to_moe(Llama(cfg), n_experts=4, top_k=2, seed=9).eval()
+    prompt = torch.randint(0, cfg.vocab_size, (1, 6))
+    out = model.generate(prompt, max_new_tokens=5)
+    assert out.shape == (1, 11)
+    # cached decode == full recompute (argmax greedy, fp32 CPU)
+    toks = prompt.clone()
+    with torch.no_grad():
+        for _ in range(5):
+            nxt = model(toks)[:, -1].argmax(-1, keepdim=True)
+            toks = torch.cat([toks, nxt], dim=1)
+    assert (out == toks).all()
