@@ -186,3 +186,14 @@ def test_secret_provider_presets_complete():
 
     with _pytest.raises(ValueError, match="known"):
         secret_factory("not-a-provider")
+
+
+def test_config_set_persist(tmp_path):
+    from kubetorch_amd.config import KTConfig
+
+    path = tmp_path / "config"
+    cfg = KTConfig(path=str(path))
+    cfg.set("namespace", "teamspace", persist=True)
+    assert path.exists()
+    cfg2 = KTConfig(path=str(path))
+    assert cfg2.namespace == "teamspace"
