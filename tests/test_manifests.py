@@ -158,3 +158,20 @@ def test_gpu_memory_and_disk_size():
     assert res["limits"]["amd.com/gpu"] == "1"
     assert res["requests"]["ephemeral-storage"] == "50Gi"
     assert m["metadata"]["annotations"]["gpu-memory"] == "64Gi"
+
+
+def test_distribute_sets_replicas_and_spmd_env():
+    c = Compute(cpus=1).distribute("pytorch", workers=3, num_proc=2)
+    assert c.replicas == 3
+    assert c.distributed_config["type"] == "pytorch"
+    m = c.to_manifest("svc", username="u")
+    assert m["spec"]["replicas"] == 3
+
+
+def test_autoscale_bounds_in_manifest():
+    c = Compute(cpus=1).autoscale(min_scale=1, max_scale=6, target=5,
+                                  metric="concurrency")
+    m = c.to_manifest("svc", username="u")
+    ann = m["spec"]["template"]["metadata"]["annotations"]
+    assert ann["autoscaling.knative.dev/min-scale"] == "1"
+    assert ann["autoscaling.knative.dev/max-scale"] == "6"
