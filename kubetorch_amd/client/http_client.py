@@ -126,7 +126,7 @@ class HTTPClient:
         except httpx.HTTPError:
             return False
 
-    def call(self, args=(), kwargs=None, method=None, serialization="pickle",
+    def call(self, args=(), kwargs=None, method=None, serialization="json",
              stream_logs=False, stream_metrics=False, timeout=None,
              workers=None, restart_procs=False, request_id=None,
              debug=False):
@@ -151,6 +151,14 @@ class HTTPClient:
                 pickle.dumps((tuple(args), kwargs or {}))).decode()}
         else:
             body = {"args": list(args), "kwargs": kwargs or {}}
+            try:
+                json.dumps(body)
+            except (TypeError, ValueError) as e:
+                raise TypeError(
+                    "arguments are not JSON-serializable; call with "
+                    "serialization='pickle' and deploy with "
+                    "Compute(allowed_serialization=['json', 'pickle'])"
+                ) from e
         streamer = None
         mstreamer = None
         if stream_logs:

@@ -38,6 +38,7 @@ class Module:
         self.service_hosts = []
         self.launch_id = None
         self._http = None
+        self._exec_token = None  # per-deploy shared secret for run_bash
         self.stream_logs = bool(config.get("stream_logs"))
 
     @property
@@ -65,7 +66,11 @@ class Module:
                                    if self.compute else None),
             "image_setup": (self.compute.image_setup_contents()
                             if self.compute else ""),
+            "allowed_serialization": ",".join(
+                self.compute.allowed_serialization) if self.compute else "json",
         }
+        if self._exec_token:
+            md["exec_token"] = self._exec_token
         if self.init_args is not None:
             md["init_args"] = base64.b64encode(
                 pickle.dumps(self.init_args)).decode()
@@ -155,6 +160,7 @@ class Module:
                     print(f"[kt] reusing existing service {found}")
                 return self
         t0 = time.time()
+        self._exec_token = uuid.uuid4().hex
         md = self.metadata()
         self._sync_workdir(md)
         launch_id = uuid.uuid4().hex[:12]
@@ -243,9 +249,17 @@ class Module:
 
     # -- post-launch helpers (reference: compute.py ssh/pip_install/run_bash) --
     def run_bash(self, command, timeout=600):
-        """Run a bash command inside the service's (first) pod."""
+        """Run a bash command inside the service's (first) pod. Requires the
+        per-deploy exec token (set on the pod at deploy time); a service bound
+        via get_if_exists has no token and cannot exec."""
+        if not self._exec_token:
+            raise PermissionError(
+                "run_bash requires the exec token from this client's own "
+                "deploy (.to()); a service bound with get_if_exists cannot "
+                "be exec'd remotely")
         r = shared_client().post(self.http.base_url + "/exec",
                                  json={"command": command, "timeout": timeout},
+                                 headers={"X-KT-Exec-Token": self._exec_token},
                                  timeout=timeout + 30)
         r.raise_for_status()
         return r.json()

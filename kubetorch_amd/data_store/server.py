@@ -16,6 +16,8 @@ import io
 import json
 import os
 import tarfile
+
+from kubetorch_amd.utils.tar import safe_extractall
 import threading
 import time
 
@@ -55,7 +57,7 @@ async def put_file(key: str, request: Request):
     if request.headers.get("X-KT-Tar") == "1":
         os.makedirs(path, exist_ok=True)
         with tarfile.open(fileobj=io.BytesIO(body), mode="r:gz") as tar:
-            tar.extractall(path)  # noqa: S202 - trusted in-cluster clients
+            safe_extractall(tar, path)
     else:
         with open(path, "wb") as f:
             f.write(body)

@@ -20,7 +20,7 @@ class Compute:
                  secrets=None, shared_memory="8Gi", node_selector=None,
                  annotations=None, inactivity_ttl=None, queue=None,
                  launch_timeout=C.LAUNCH_TIMEOUT, endpoint=None,
-                 local=None):
+                 local=None, allowed_serialization=None):
         self.cpus = cpus
         self.memory = memory
         self.gpus = gpus
@@ -43,6 +43,10 @@ class Compute:
         self.queue = queue
         self.launch_timeout = launch_timeout
         self.endpoint = endpoint
+        # serialization formats the pod will accept for /call bodies.
+        # Default json (reference parity: KT_ALLOWED_SERIALIZATION); pass
+        # ["json", "pickle"] to allow pickled args from trusted clients.
+        self.allowed_serialization = list(allowed_serialization or ["json"])
         self.local = config.local_mode if local is None else local
         self.kind = "deployment"
         self.replicas = 1

@@ -23,6 +23,8 @@ def client():
     os.environ["KT_MODULE_NAME"] = "summer"
     os.environ["KT_MODULE_TYPE"] = "fn"
     os.environ["KT_LAUNCH_ID"] = "unit-lid-1"
+    os.environ["KT_ALLOWED_SERIALIZATION"] = "json,pickle"
+    os.environ["KT_EXEC_TOKEN"] = "unit-exec-token"
     os.environ.pop("KT_CONTROLLER_URL", None)
     os.environ.pop("KT_DISTRIBUTED_CONFIG", None)
     for m in list(sys.modules):
@@ -39,7 +41,8 @@ def client():
         sup.cleanup()
         http_server.STATE["supervisor"] = None
     for k in ("KT_FILE_PATH", "KT_PROJECT_ROOT", "KT_CLS_OR_FN_NAME",
-              "KT_MODULE_NAME", "KT_MODULE_TYPE", "KT_LAUNCH_ID"):
+              "KT_MODULE_NAME", "KT_MODULE_TYPE", "KT_LAUNCH_ID",
+              "KT_ALLOWED_SERIALIZATION", "KT_EXEC_TOKEN"):
         os.environ.pop(k, None)
 
 
@@ -83,10 +86,47 @@ def test_remote_exception_packaging(client):
 
 
 def test_exec_endpoint(client):
-    r = client.post("/exec", json={"command": "echo hi && exit 0"})
+    r = client.post("/exec", json={"command": "echo hi && exit 0"},
+                    headers={"X-KT-Exec-Token": "unit-exec-token"})
     assert r.status_code == 200
     assert r.json()["returncode"] == 0
     assert "hi" in r.json()["stdout"]
+
+
+def test_exec_requires_token(client):
+    # no token -> 403, wrong token -> 403 (route is RCE if left open)
+    r = client.post("/exec", json={"command": "id"})
+    assert r.status_code == 403
+    r = client.post("/exec", json={"command": "id"},
+                    headers={"X-KT-Exec-Token": "wrong"})
+    assert r.status_code == 403
+
+
+def test_exec_disabled_without_configured_token(client):
+    tok = os.environ.pop("KT_EXEC_TOKEN")
+    try:
+        r = client.post("/exec", json={"command": "id"},
+                        headers={"X-KT-Exec-Token": ""})
+        assert r.status_code == 403
+    finally:
+        os.environ["KT_EXEC_TOKEN"] = tok
+
+
+def test_pickle_serialization_default_denied(client):
+    # with no allowlist env, only json is accepted (reference parity:
+    # KT_ALLOWED_SERIALIZATION default "json")
+    allowed = os.environ.pop("KT_ALLOWED_SERIALIZATION")
+    try:
+        body = {"body": base64.b64encode(pickle.dumps(((7, 8), {}))).decode()}
+        r = client.post("/call/summer", json=body,
+                        headers={"X-Serialization": "pickle"})
+        assert r.status_code == 400
+        assert "not allowed" in r.json()["error"]["message"]
+        # json still works
+        r = client.post("/call/summer", json={"args": [1, 2], "kwargs": {}})
+        assert r.status_code == 200 and r.json()["result"] == 3
+    finally:
+        os.environ["KT_ALLOWED_SERIALIZATION"] = allowed
 
 
 def test_metrics_and_logs_endpoints(client):
