@@ -9,8 +9,13 @@
 //      buffer, dk/dv written per Q-head (host sums groups)
 //   3. FmhaBwdConvertQGradKernel: dq_acc f32 -> dq bf16
 //
-// gfx950: kUseTrLoad=true selects BlockFmhaBwdDQDKDVPipelineTrLoadKRKTRVR
-// (ds_read_b64_tr_b16 transposed-fragment loads) via the pipeline selector.
+// TWO pipeline instantiations, runtime-selected (pipe arg):
+//   0 "std":    kUseTrLoad=false -> BlockFmhaBwdDQDKDVPipelineKRKTRVRIGLP,
+//               classic 32x32x16 warp tiles (the widely-validated CK
+//               codegen config for hd128)
+//   1 "trload": kUseTrLoad=true -> BlockFmhaBwdDQDKDVPipelineTrLoadKRKTRVR
+//               (gfx950 ds_read_b64_tr_b16 transposed-fragment loads),
+//               16x16x32 warp tiles (kCM0PerLane==1 constraint)
 // Replaces torch's AITER asm bwd path (dense over expanded KV) —
 // profiles/ROUND2.md lever #1.
 #include <ck_tile/core.hpp>
@@ -22,17 +27,6 @@ namespace {
 
 using bf16 = ck_tile::bf16_t;
 
-// hd128 bwd tile: K/V block 128 keys resident, 32-query M-loop.
-// BlockTile = <M0, N0, K0, K1, K2, K3, K4, QKHeaddim, VHeaddim>
-// constraints: kM0 == kK1 == kK3 (contraction over the M tile).
-using BwdShape = ck_tile::TileFmhaBwdShape<
-    ck_tile::sequence<32, 128, 32, 32, 32, 32, 32, 128, 128>,
-    ck_tile::sequence<1, 4, 1>, ck_tile::sequence<16, 16, 32>,   // gemm0 S:   M0xN0
-    ck_tile::sequence<4, 1, 1>, ck_tile::sequence<16, 16, 32>,   // gemm1 dV:  N0xDv
-    ck_tile::sequence<1, 4, 1>, ck_tile::sequence<16, 16, 32>,   // gemm2 dP:  M0xN0
-    ck_tile::sequence<4, 1, 1>, ck_tile::sequence<16, 16, 32>,   // gemm3 dK:  N0xDq
-    ck_tile::sequence<1, 4, 1>, ck_tile::sequence<16, 16, 32>>;  // gemm4 dQ:  M0xDq
-
 using BwdTraits = ck_tile::TileFmhaBwdTraits<
     0 /* kPadHeadDimQ: exact 128 */, 0 /* kPadHeadDimV */,
     ck_tile::BlockAttentionBiasEnum::NO_BIAS, false /* kHasBiasGrad */>;
@@ -40,23 +34,49 @@ using BwdTraits = ck_tile::TileFmhaBwdTraits<
 using BwdMask = ck_tile::SimplifiedGenericAttentionMask<true>;
 using BwdDropout = ck_tile::BlockDropoutBwd<false, false, false>;
 
-using BwdProblem = ck_tile::BlockFmhaBwdPipelineProblem<
-    bf16 /* Q */, bf16 /* K */, bf16 /* V */, bf16 /* Gemm */,
-    float /* LSE */, float /* Acc */, float /* D */, bf16 /* Bias */,
-    uint8_t /* RandVal */, bf16 /* O */, bf16 /* OGrad */, bf16 /* QGrad */,
-    bf16 /* KGrad */, bf16 /* VGrad */, bf16 /* BiasGrad */, BwdShape,
-    false /* kIsGroupMode */, false /* kIsDeterministic */, BwdMask,
-    BwdDropout, true /* kUseTrLoad: gfx950 */, BwdTraits>;
+// hd128 bwd tile: K/V block 128 keys resident, 32-query M-loop.
+// BlockTile = <M0, N0, K0, K1, K2, K3, K4, QKHeaddim, VHeaddim>
+// constraints: kM0 == kK1 == kK3 (contraction over the M tile).
+// kK0 and kK2 MUST equal the head dim: the KR/VR pipelines issue gemm0
+// (Q@K^T) and gemm2 (dO@V^T) as a SINGLE block-gemm over a register tile
+// read as (kN0, kK0)/(kN0, kK2) — a smaller kK0 silently contracts only
+// the first kK0 of 128 head dims (diagnosed on-box: chunk regression
+// showed coeffs [1,1,1,1,0,...] at kK0=32, gpurun_out/bwd_probe4.log).
+// Only gemm4 (dS@K over kN0 keys) has an explicit k4 loop.
+// Block-warp M dims follow gemm1/3 M=N0 (dV/dK partition keys across
+// warps), gemm0/2/4 M=M0.
+template <bool UseTrLoad, int WtM, int WtN, int WtK, int M0 = 32>
+struct BwdConfig {
+  using Shape = ck_tile::TileFmhaBwdShape<
+      ck_tile::sequence<M0, 128, 128, M0, 128, M0, 32, 128, 128>,
+      ck_tile::sequence<1, 4, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm0 S
+      ck_tile::sequence<4, 1, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm1 dV
+      ck_tile::sequence<1, 4, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm2 dP
+      ck_tile::sequence<4, 1, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm3 dK
+      ck_tile::sequence<1, 4, 1>, ck_tile::sequence<WtM, WtN, WtK>>; // gemm4 dQ
 
-using BwdPipeline = ck_tile::BlockFmhaBwdDQDKDVPipeline<BwdProblem>;
+  using Problem = ck_tile::BlockFmhaBwdPipelineProblem<
+      bf16 /* Q */, bf16 /* K */, bf16 /* V */, bf16 /* Gemm */,
+      float /* LSE */, float /* Acc */, float /* D */, bf16 /* Bias */,
+      uint8_t /* RandVal */, bf16 /* O */, bf16 /* OGrad */, bf16 /* QGrad */,
+      bf16 /* KGrad */, bf16 /* VGrad */, bf16 /* BiasGrad */, Shape,
+      false /* kIsGroupMode */, false /* kIsDeterministic */, BwdMask,
+      BwdDropout, UseTrLoad, BwdTraits>;
 
-using KGradEpilogue = ck_tile::Default2DEpilogue<
-    ck_tile::Default2DEpilogueProblem<float, bf16, false, false>>;
-using VGradEpilogue = ck_tile::Default2DEpilogue<
-    ck_tile::Default2DEpilogueProblem<float, bf16, false, false>>;
+  using Pipeline = ck_tile::BlockFmhaBwdDQDKDVPipeline<Problem>;
 
-using BwdKernel =
-    ck_tile::FmhaBwdDQDKDVKernel<BwdPipeline, KGradEpilogue, VGradEpilogue>;
+  using KGradEpilogue = ck_tile::Default2DEpilogue<
+      ck_tile::Default2DEpilogueProblem<float, bf16, false, false>>;
+  using VGradEpilogue = ck_tile::Default2DEpilogue<
+      ck_tile::Default2DEpilogueProblem<float, bf16, false, false>>;
+
+  using Kernel = ck_tile::FmhaBwdDQDKDVKernel<Pipeline, KGradEpilogue,
+                                              VGradEpilogue>;
+};
+
+using StdKernel = BwdConfig<false, 32, 32, 16>::Kernel;
+using TrKernel = BwdConfig<true, 16, 16, 32>::Kernel;
+using TrKernel64 = BwdConfig<true, 16, 16, 32, 64>::Kernel;  // pipe=2
 
 // --- D = rowsum(dO*O) -------------------------------------------------------
 using DotTraits = ck_tile::TileFmhaBwdOGradDotOTraits<
@@ -88,23 +108,11 @@ void launch(hipStream_t stream, dim3 grid, Args&&... args) {
                K{}, grid, K::BlockSize(), 0, std::forward<Args>(args)...));
 }
 
-}  // namespace
-
-// All tensors contiguous [B, H, S, 128] bf16 (Hq for q/o/do/dq, Hkv for
-// k/v; dk/dv are Hq-EXPANDED — the caller group-sums). lse/d: [B, Hq, S]
-// f32. dq_acc: [B, Hq, S, 128] f32 ZEROED by the caller (atomic accum).
-// mask_mode selects the causal-mask karg convention (runtime, so round-2's
-// flipped-mask A/B needs no recompile — profiles/ROUND2.md lever #1):
-//   0: window (-1, 0), MASK_FROM_TOP_LEFT     (matches the fwd kernels)
-//   1: window (-1, 0), MASK_FROM_BOTTOM_RIGHT
-//   2: window (0, -1), MASK_FROM_TOP_LEFT     (anti-causal at the
-//      constructor — causal if the trload pipeline masks the S^T tile)
-extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
-                               const void* o, const void* do_, const void* lse,
-                               void* d, void* dq_acc, void* dq, void* dk,
-                               void* dv, int B, int Hq, int Hkv, int S,
-                               float scale, int mask_mode,
-                               hipStream_t stream) {
+template <typename MainKernel>
+void run_bwd(const void* q, const void* k, const void* v, const void* o,
+             const void* do_, const void* lse, void* d, void* dq_acc, void* dq,
+             void* dk, void* dv, int B, int Hq, int Hkv, int S, float scale,
+             int mask_mode, hipStream_t stream) {
   const ck_tile::index_t D = 128;
   const ck_tile::index_t sq = (ck_tile::index_t)S * D;   // nhead stride q-side
   const ck_tile::index_t bq = (ck_tile::index_t)Hq * sq; // batch stride q-side
@@ -120,7 +128,7 @@ extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
     launch<DotKernel>(stream, DotKernel::GridSize(B, Hq, S), kargs);
   }
   {  // 2: dq_acc / dk / dv
-    auto kargs = BwdKernel::MakeKargsImpl(
+    auto kargs = MainKernel::MakeKargsImpl(
         q, k, v, nullptr /* bias */, lse, do_, d, nullptr /* randval */,
         dk, dv, nullptr /* dbias */, dq_acc,
         S /* seqlen_q */, S /* seqlen_k */, D, D, Hq, Hq / Hkv, scale,
@@ -144,7 +152,7 @@ extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
             ? ck_tile::GenericAttentionMaskEnum::MASK_FROM_BOTTOM_RIGHT
             : ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT),
         0.0f /* p_drop */, std::make_pair<uint64_t, uint64_t>(0, 0));
-    launch<BwdKernel>(stream, BwdKernel::GridSize(B, Hq, S), kargs);
+    launch<MainKernel>(stream, MainKernel::GridSize(B, Hq, S), kargs);
   }
   {  // 3: dq = bf16(dq_acc)
     auto kargs = CvtKernel::MakeKargs(
@@ -153,5 +161,33 @@ extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
         bq /* batch_stride_dq */, bq /* batch_stride_dq_acc */,
         0 /* split_stride_dq_acc */);
     launch<CvtKernel>(stream, CvtKernel::GridSize(B, Hq, S), kargs);
+  }
+}
+
+}  // namespace
+
+// All tensors contiguous [B, H, S, 128] bf16 (Hq for q/o/do/dq, Hkv for
+// k/v; dk/dv are Hq-EXPANDED — the caller group-sums). lse/d: [B, Hq, S]
+// f32. dq_acc: [B, Hq, S, 128] f32 ZEROED by the caller (atomic accum).
+// mask_mode selects the causal-mask karg convention (runtime A/B):
+//   0: window (-1, 0), MASK_FROM_TOP_LEFT     (matches the fwd kernels)
+//   1: window (-1, 0), MASK_FROM_BOTTOM_RIGHT
+//   2: window (0, -1), MASK_FROM_TOP_LEFT     (anti-causal)
+// pipe: 0 = std (KRKTRVR IGLP, 32x32x16), 1 = trload (gfx950, 16x16x32).
+extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
+                               const void* o, const void* do_, const void* lse,
+                               void* d, void* dq_acc, void* dq, void* dk,
+                               void* dv, int B, int Hq, int Hkv, int S,
+                               float scale, int mask_mode, int pipe,
+                               hipStream_t stream) {
+  if (pipe == 2) {
+    run_bwd<TrKernel64>(q, k, v, o, do_, lse, d, dq_acc, dq, dk, dv, B, Hq,
+                        Hkv, S, scale, mask_mode, stream);
+  } else if (pipe == 1) {
+    run_bwd<TrKernel>(q, k, v, o, do_, lse, d, dq_acc, dq, dk, dv, B, Hq, Hkv,
+                      S, scale, mask_mode, stream);
+  } else {
+    run_bwd<StdKernel>(q, k, v, o, do_, lse, d, dq_acc, dq, dk, dv, B, Hq, Hkv,
+                       S, scale, mask_mode, stream);
   }
 }

@@ -50,7 +50,7 @@ void kt_pack_segments(const void* ptrs, const void* nbytes, const void* offs,
 void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
                     const void* o, const void* do_, const void* lse, void* d,
                     void* dq_acc, void* dq, void* dk, void* dv, int B, int Hq,
-                    int Hkv, int S, float scale, int mask_mode,
+                    int Hkv, int S, float scale, int mask_mode, int pipe,
                     hipStream_t stream);
 }
 
@@ -324,13 +324,19 @@ std::vector<at::Tensor> attn_bwd_ck(const at::Tensor& grad_out,
   auto dq_acc = at::zeros({B, Hq, S, 128}, q.options().dtype(at::kFloat));
   // KT_CKBWD_MASK selects the causal-mask convention at runtime (WIP
   // flipped-mask A/B, profiles/ROUND2.md): 0=top-left (fwd convention),
-  // 1=bottom-right, 2=swapped lr window.
+  // 1=bottom-right, 2=swapped lr window. KT_CKBWD_PIPE selects the
+  // pipeline: 0=std (KRKTRVR IGLP, 32x32x16 warp tiles), 1=trload
+  // (gfx950 transposed-fragment loads, 16x16x32).
   int mask_mode = 0;
   if (const char* mm = std::getenv("KT_CKBWD_MASK")) mask_mode = atoi(mm);
+  // default pipe 1 (gfx950 trload, 311 TF measured) — fastest correct
+  // in-tree pipeline; pipe 0 = classic IGLP (221 TF), 2 = M0=64 (203 TF)
+  int pipe = 1;
+  if (const char* pp = std::getenv("KT_CKBWD_PIPE")) pipe = atoi(pp);
   kt_attn_bwd_ck(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                  go.data_ptr(), lse.data_ptr(), d.data_ptr(),
                  dq_acc.data_ptr(), dq.data_ptr(), dk.data_ptr(),
-                 dv.data_ptr(), B, Hq, Hkv, S, (float)scale, mask_mode,
+                 dv.data_ptr(), B, Hq, Hkv, S, (float)scale, mask_mode, pipe,
                  cur_stream(q));
   return {dq, dk, dv};
 }

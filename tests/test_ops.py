@@ -416,13 +416,11 @@ def test_pack_segments_kernel():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(os.environ.get("KT_ATTN_BWD_TEST") != "1",
-                    reason="CK bwd WIP: causal mask orientation flipped "
-                           "(profiles/ROUND2.md lever #1) — enable with "
-                           "KT_ATTN_BWD_TEST=1 while fixing")
 def test_attn_bwd_ck_gqa_native():
     """CK-tile GQA-native backward vs fp32 SDPA reference (opt-in path,
-    KT_ATTN_BWD=ck; round-2 default candidate)."""
+    KT_ATTN_BWD=ck). Root cause of the round-1 failure was kK0/kK2=32
+    truncating the head-dim contraction to 32 of 128 dims (fixed:
+    kK0=kK2=128; the mask was never wrong — gpurun_out/bwd_probe4.log)."""
     torch.manual_seed(9)
     B, Hq, Hkv, S, D = 2, 8, 2, 256, 128
     scale = D ** -0.5
