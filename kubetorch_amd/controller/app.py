@@ -111,14 +111,54 @@ async def _ttl_reaper(interval=None):
                     pass
 
 
+async def _pod_monitor(interval=None):
+    """Auto re-provision: watch every registered workload and re-apply its
+    manifest when live pods < desired replicas (the reconciliation loop a
+    K8s Deployment controller provides; the local driver needs its own so
+    a killed worker pod auto-heals without a client re-deploy — BASELINE
+    config 4's 'auto re-provision'). Respawned pods join the next call's
+    rendezvous (per-call process groups = elastic re-join)."""
+    if interval is None:
+        interval = float(os.environ.get("KT_POD_MONITOR_INTERVAL", "1.0"))
+    backoff = {}  # (ns, name) -> last respawn ts
+    while True:
+        await asyncio.sleep(interval)
+        if HUB.driver_name != "local":
+            continue  # K8s reconciles its own Deployments
+        for (ns, name), w in list(HUB.workloads.items()):
+            manifest = w.get("manifest") or {}
+            if not manifest:
+                continue
+            desired = manifest.get("spec", {}).get("replicas", 1)
+            try:
+                alive = len(HUB.driver.pods(name, ns))
+            except Exception:
+                continue
+            if alive >= desired:
+                continue
+            if time.time() - backoff.get((ns, name), 0) < 3.0:
+                continue
+            backoff[(ns, name)] = time.time()
+            try:
+                await asyncio.to_thread(
+                    HUB.driver.apply, manifest, ns, w.get("metadata"),
+                    w.get("launch_id"))
+                HUB.driver._event(ns, name, "Respawned",
+                                  f"re-provisioned {desired - alive} pod(s)")
+            except Exception:
+                pass
+
+
 from contextlib import asynccontextmanager
 
 
 @asynccontextmanager
 async def _lifespan(app):
     task = asyncio.create_task(_ttl_reaper())
+    mon = asyncio.create_task(_pod_monitor())
     yield
     task.cancel()
+    mon.cancel()
 
 
 app = FastAPI(lifespan=_lifespan)

@@ -57,6 +57,8 @@ class LocalDriver:
         self.services = {}  # (ns, name) -> [LocalPod]
         self.events = {}    # (ns, name) -> [event dict] (launch streaming)
         self._event_state = {}  # pod name -> last seen liveness
+        self._master_ports = {}  # (ns, name) -> rendezvous port (stable
+        # across pod replacement so re-joined ranks find the same master)
 
     def _event(self, namespace, name, reason, message, pod=None):
         self.events.setdefault((namespace, name), []).append({
@@ -84,6 +86,12 @@ class LocalDriver:
         return [e for e in self.events.get(key, []) if e["ts"] > since]
 
     def apply(self, manifest, namespace, metadata=None, launch_id=None):
+        """Reconcile the service to the manifest's replica count. Warm
+        (alive) pods are kept — a hot reload reaches them through the
+        controller push channel — and only dead/missing pods are spawned
+        (the K8s-Deployment-controller behavior the reference delegates to
+        Kubernetes; here the control plane does its own reconciliation,
+        which is what makes mid-step pod death auto-heal)."""
         name = manifest["metadata"]["name"]
         replicas = manifest.get("spec", {}).get("replicas", 1)
         key = (namespace, name)
@@ -92,15 +100,24 @@ class LocalDriver:
         if len(alive) == replicas:
             self._event(namespace, name, "Reloaded",
                         f"hot reload into {len(alive)} warm pod(s)")
+            self.services[key] = alive
             return [p.host for p in alive]  # warm pods: reload only
-        for p in pods:
-            p.kill()
-        ports = [_free_port() for _ in range(replicas)]
-        peer_list = ",".join(f"127.0.0.1:{p}" for p in ports)
-        master_port = _free_port()  # one rendezvous port for the whole service
+        if len(alive) > replicas:  # scale down: drop the newest extras
+            for p in alive[replicas:]:
+                p.kill()
+            alive = alive[:replicas]
+        n_new = replicas - len(alive)
+        used_idx = {int(p.name.rsplit("-", 1)[1]) for p in alive}
+        new_idx = [i for i in range(replicas + len(used_idx))
+                   if i not in used_idx][:n_new]
+        ports = [_free_port() for _ in range(n_new)]
+        peer_list = ",".join([p.host for p in alive]
+                             + [f"127.0.0.1:{p}" for p in ports])
+        # one stable rendezvous port for the whole service lifetime
+        master_port = self._master_ports.setdefault(key, _free_port())
         new_pods = []
         md = metadata or {}
-        for i, port in enumerate(ports):
+        for i, port in zip(new_idx, ports):
             env = dict(os.environ)
             env.update({
                 "KT_SERVER_PORT": str(port),
@@ -153,13 +170,14 @@ class LocalDriver:
             self._event(namespace, name, "Started",
                         "container started", pod=f"{name}-{i}")
             self._event_state[f"{name}-{i}"] = True
-        self.services[key] = new_pods
-        return [p.host for p in new_pods]
+        self.services[key] = alive + new_pods
+        return [p.host for p in self.services[key]]
 
     def delete(self, name, namespace):
         for p in self.services.pop((namespace, name), []):
             p.kill()
         self.events.pop((namespace, name), None)
+        self._master_ports.pop((namespace, name), None)
 
     def pods(self, name, namespace):
         return [p.host for p in self.services.get((namespace, name), [])

@@ -95,6 +95,11 @@ class MembershipMonitor:
     def stop(self):
         self._stop.set()
 
+    def rebase(self, baseline):
+        """Accept the current peer set as the new normal (start of a new
+        call): prior joins/leaves no longer count as a change."""
+        self._baseline = set(baseline)
+
     def subscribe(self, callback):
         with self._lock:
             self._subs.append(callback)

@@ -278,6 +278,11 @@ class SPMDSupervisor(DistributedSupervisor):
 
         if not self.monitor._thread or not self.monitor._thread.is_alive():
             self.monitor.start(hosts)
+        else:
+            # membership changes BETWEEN calls are the expected elastic path
+            # (a respawned pod joins this call's rendezvous); only changes
+            # during the call should abort — re-baseline to this call's set
+            self.monitor.rebase(hosts)
         aborted = []
         self.monitor.subscribe(aborted.append)
         executor = ThreadPoolExecutor(max_workers=min(64, len(hosts) + 4))

@@ -44,15 +44,29 @@ def test_kill_pod_mid_call_then_rejoin():
         from kubetorch_amd.controller.app import HUB
 
         pods = HUB.driver.services[("default", f.name)]
-        pods[1].kill()
+        killed = pods[1]
+        killed.kill()
         t.join(90)
         assert not t.is_alive(), "call did not abort after pod death"
         assert "error" in result, f"expected abort, got {result}"
         err = result["error"]
         assert isinstance(err, WorkerMembershipChanged), repr(err)
 
-        # re-deploy: controller re-provisions (respawn), next call succeeds
-        f.to()
-        assert f(3, delay=0, kt_restart_procs=True) == [3, 3]
+        # AUTOMATIC re-provision: the controller's pod monitor respawns the
+        # dead pod (no client re-deploy) and the next call's rendezvous
+        # re-forms the group with the replacement (elastic re-join)
+        deadline = time.time() + 60
+        while time.time() < deadline:
+            alive = HUB.driver.pods(f.name, "default")
+            if len(alive) == 2 and killed.host not in alive:
+                break
+            time.sleep(0.5)
+        else:
+            raise AssertionError(
+                f"pod monitor did not respawn: {HUB.driver.pods(f.name, 'default')}")
+        # respawned pod must be ready before it can serve the fan-out
+        assert f(3, delay=0, kt_restart_procs=True, kt_timeout=120) == [3, 3]
+        evs = HUB.driver.events[("default", f.name)]
+        assert any(e["reason"] == "Respawned" for e in evs)
     finally:
         f.teardown()
