@@ -50,3 +50,52 @@ class CudaInference:
         kt.get("rlgpu/w", dest)
         torch.cuda.synchronize()
         return float(dest.float().sum().item())
+
+
+def elastic_gpu_step(v, delay=0.0):
+    """Distributed step for the GPU elastic drill: every rank does real
+    CUDA work, then all ranks agree on the result through a per-call
+    process group (gloo on a 1-GPU box: both worker pods share cuda:0, so
+    RCCL can't host two ranks; the PG lifecycle is identical). The delay
+    keeps the call in flight long enough to kill a pod mid-step."""
+    import os
+    import time
+
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    dev = torch.device("cuda", 0)
+    a = torch.full((512, 512), float(v), device=dev, dtype=torch.bfloat16)
+    out = (a @ torch.eye(512, device=dev, dtype=torch.bfloat16)).float().mean()
+    if delay:
+        time.sleep(delay)
+    if world > 1:
+        if dist.is_initialized():
+            dist.destroy_process_group()  # fresh group per call (elastic)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        t = torch.tensor([out.item()])
+        dist.all_reduce(t)
+        dist.destroy_process_group()
+        return {"rank": rank, "world": world, "sum": float(t.item())}
+    return {"rank": rank, "world": world, "sum": float(out.item())}
+
+
+def rccl_group_lifecycle(cycles=3):
+    """In-step RCCL group destroy/reinit on gfx950 (world=1 on a 1-GPU
+    box): validates the create -> collective -> destroy -> recreate cycle
+    the per-call elastic model relies on, with device tensors over the
+    nccl(=RCCL) backend."""
+    import torch.distributed as dist
+
+    results = []
+    for i in range(cycles):
+        store = dist.TCPStore("127.0.0.1", 29650 + i, 1, True)
+        dist.init_process_group("nccl", store=store, rank=0, world_size=1)
+        t = torch.full((1024,), float(i + 1), device="cuda")
+        dist.all_reduce(t)
+        dist.barrier()
+        results.append(float(t.sum().item()))
+        dist.destroy_process_group()
+        del store
+    return results
