@@ -376,6 +376,31 @@ class FlatDDP:
             b.pending = len(b.params)
             b.work = None
 
+    # -- in-step elastic support (parallel/elastic.py) -----------------------
+    def abort_comm(self):
+        """Forget all in-flight comm state after a collective failure: the
+        async works reference a dead process group, and flat grads may hold
+        partial sums. The interrupted step is re-run by the caller after
+        the group re-forms."""
+        for b in self.buckets:
+            b.work = None
+            b.completions = 0
+            b.pending = len(b.params)
+            b.flat_grad.zero_()
+        self._micro_step = 0
+        self._step_started = False
+        if self._copy_mode:
+            for p in self._param_bucket:
+                p.grad = None
+
+    def set_world(self, world):
+        """Adopt a new group size after an in-step re-join (zero=False
+        only: ZeRO shards are sized by world at construction)."""
+        if self.zero:
+            raise ValueError("ZeRO-1 engine cannot change world in place")
+        self._world = max(1, int(world))
+        self._grad_scale = 1.0 / (self._world * self.grad_accum_steps)
+
     @torch.no_grad()
     def broadcast_params(self, src=0):
         """Sync initial params across ranks (one collective per bucket)."""
