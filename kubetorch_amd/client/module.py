@@ -274,8 +274,30 @@ class Module:
         return self.run_bash(
             f"{shlex.quote(sys.executable)} -m pip install {pkgs} {extra_args}")
 
+    def _lb_client(self):
+        """Round-robin over the service's live pods for autoscaled
+        (knative-kind) services — the role the K8s Service/Knative
+        activator plays in-cluster. Pod list refreshed from the
+        controller every 2 s so scale-ups receive traffic."""
+        now = time.time()
+        if now - getattr(self, "_lb_ts", 0) > 2.0:
+            try:
+                w = controller_client().get_workload(self.name, self.namespace)
+                hosts = w.get("pods") or []
+            except Exception:
+                hosts = []
+            self._lb_hosts = hosts or self.service_hosts
+            self._lb_ts = now
+        self._lb_i = getattr(self, "_lb_i", -1) + 1
+        host = self._lb_hosts[self._lb_i % len(self._lb_hosts)]
+        return HTTPClient(f"http://{host}", self.pointers["name"])
+
     def _call(self, args, kwargs, method=None, **opts):
-        return self.http.call(
+        client = self.http
+        if (self.compute is not None and self.compute.kind == "knative"
+                and self.compute.local):
+            client = self._lb_client()
+        return client.call(
             args=args, kwargs=kwargs, method=method,
             stream_logs=opts.pop("stream_logs", self.stream_logs),
             **opts,

@@ -111,6 +111,96 @@ async def _ttl_reaper(interval=None):
                     pass
 
 
+def _parse_duration(raw, default=0.0):
+    if raw in (None, ""):
+        return default
+    raw = str(raw).strip().lower()
+    mult = {"s": 1, "m": 60, "h": 3600}.get(raw[-1])
+    try:
+        return float(raw[:-1]) * mult if mult else float(raw)
+    except ValueError:
+        return default
+
+
+def _autoscale_spec(manifest):
+    """Extract the Knative KPA annotations from a knative-Service manifest
+    (None for non-autoscaled workloads)."""
+    ann = (manifest.get("spec", {}).get("template", {})
+           .get("metadata", {}).get("annotations", {}))
+    if "autoscaling.knative.dev/target" not in ann:
+        return None
+    g = lambda k, d: ann.get(f"autoscaling.knative.dev/{k}", d)  # noqa: E731
+    return {
+        "target": max(1, int(float(g("target", 100)))),
+        "metric": g("metric", "concurrency"),
+        "min": max(1, int(g("min-scale", 1) or 1)),
+        "max": int(g("max-scale", 0) or 0),
+        "scale_down_delay": _parse_duration(g("scale-down-delay", "0s")),
+        "window": _parse_duration(g("window", "60s"), 60.0),
+    }
+
+
+async def _autoscaler(interval=None):
+    """KPA-style concurrency autoscaler for the local driver (the role
+    Knative's autoscaler plays in-cluster; the chart ships Knative config
+    for the K8s path — charts/kubetorch-amd/knative/). Scales a workload's
+    replica count on the summed kt_active_requests of its pods:
+    desired = ceil(in_flight / target), clamped to [min, max], with
+    scale-down held back by scale-down-delay."""
+    import httpx
+
+    if interval is None:
+        interval = float(os.environ.get("KT_AUTOSCALER_INTERVAL", "1.0"))
+    last_above = {}   # (ns, name) -> ts the signal last justified >min
+    while True:
+        await asyncio.sleep(interval)
+        if HUB.driver_name != "local":
+            continue
+        for (ns, name), w in list(HUB.workloads.items()):
+            manifest = w.get("manifest") or {}
+            spec = _autoscale_spec(manifest)
+            if spec is None:
+                continue
+            try:
+                pods = HUB.driver.pods(name, ns)
+            except Exception:
+                continue
+            in_flight = 0.0
+            async with httpx.AsyncClient(timeout=3) as client:
+                for p in pods:
+                    try:
+                        r = await client.get(f"http://{p}/metrics")
+                        for line in r.text.splitlines():
+                            if line.startswith("kt_active_requests"):
+                                in_flight += float(line.split()[-1])
+                    except Exception:
+                        pass
+            want = -(-int(in_flight) // spec["target"])  # ceil
+            want = max(spec["min"], want)
+            if spec["max"]:
+                want = min(spec["max"], want)
+            cur = w.get("desired_replicas", len(pods) or 1)
+            key = (ns, name)
+            if want >= cur:
+                last_above[key] = time.time()
+            if want > cur or (want < cur and time.time()
+                              - last_above.get(key, 0) > spec["scale_down_delay"]):
+                w["desired_replicas"] = want
+                scaled = dict(manifest)
+                scaled.setdefault("spec", {})
+                scaled["spec"] = dict(scaled["spec"], replicas=want)
+                try:
+                    await asyncio.to_thread(
+                        HUB.driver.apply, scaled, ns, w.get("metadata"),
+                        w.get("launch_id"))
+                    HUB.driver._event(ns, name, "Autoscaled",
+                                      f"{cur} -> {want} replicas "
+                                      f"(in_flight={int(in_flight)}, "
+                                      f"target={spec['target']})")
+                except Exception:
+                    pass
+
+
 async def _pod_monitor(interval=None):
     """Auto re-provision: watch every registered workload and re-apply its
     manifest when live pods < desired replicas (the reconciliation loop a
@@ -129,7 +219,8 @@ async def _pod_monitor(interval=None):
             manifest = w.get("manifest") or {}
             if not manifest:
                 continue
-            desired = manifest.get("spec", {}).get("replicas", 1)
+            desired = w.get("desired_replicas") or \
+                manifest.get("spec", {}).get("replicas", 1)
             try:
                 alive = len(HUB.driver.pods(name, ns))
             except Exception:
@@ -156,9 +247,11 @@ from contextlib import asynccontextmanager
 async def _lifespan(app):
     task = asyncio.create_task(_ttl_reaper())
     mon = asyncio.create_task(_pod_monitor())
+    scaler = asyncio.create_task(_autoscaler())
     yield
     task.cancel()
     mon.cancel()
+    scaler.cancel()
 
 
 app = FastAPI(lifespan=_lifespan)

@@ -235,16 +235,20 @@ app = FastAPI(lifespan=lifespan)
 async def metrics_middleware(request: Request, call_next):
     t0 = time.time()
     path = request.url.path.split("/")[1] or "root"
-    metrics.ACTIVE_REQUESTS.inc()
+    # the in-flight gauge counts USER calls only — metrics/health polls
+    # must not feed the autoscaler's concurrency signal
+    is_call = path in ("call", "spmd")
+    if is_call:
+        metrics.ACTIVE_REQUESTS.inc()
     try:
         resp = await call_next(request)
         metrics.HTTP_REQUESTS.labels(path=path, status=resp.status_code).inc()
         return resp
     finally:
-        metrics.ACTIVE_REQUESTS.dec()
-        metrics.HTTP_DURATION.labels(path=path).observe(time.time() - t0)
-        if path in ("call", "spmd"):
+        if is_call:
+            metrics.ACTIVE_REQUESTS.dec()
             metrics.touch_activity()
+        metrics.HTTP_DURATION.labels(path=path).observe(time.time() - t0)
 
 
 @app.get("/health")

@@ -66,3 +66,7 @@ def deploy_child_and_call():
         return child(21)
     finally:
         child.teardown()
+
+
+def pod_name():
+    return os.environ.get("POD_NAME", "unknown")
