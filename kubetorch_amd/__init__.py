@@ -53,6 +53,8 @@ _LAZY = {
     "RemoteCallError": ("kubetorch_amd.exceptions", "RemoteCallError"),
     "QuorumTimeout": ("kubetorch_amd.exceptions", "QuorumTimeout"),
     "deep_breakpoint": ("kubetorch_amd.serving.pdb_ws", "deep_breakpoint"),
+    "secret_factory": ("kubetorch_amd.resources.secret", "secret_factory"),
+    "globals": ("kubetorch_amd.globals", None),
     "ops": ("kubetorch_amd.ops", None),
     "models": ("kubetorch_amd.models", None),
     "parallel": ("kubetorch_amd.parallel", None),

@@ -409,11 +409,32 @@ def notebook(service: str = typer.Option("notebook"),
 
 @secrets_app.command("create")
 def secrets_create(name: str, provider: str = typer.Option(None),
-                   values_json: str = typer.Option("{}", "--values")):
+                   values_json: str = typer.Option("{}", "--values"),
+                   namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
     from kubetorch_amd.resources.secret import Secret
 
     s = Secret(name, values=json.loads(values_json), provider=provider)
-    console.print_json(json.dumps(s.to_manifest("default"), default=str))
+    controller_client().put_secret(s, namespace)
+    console.print(f"[green]secret {s.k8s_name} created[/green] "
+                  f"(keys: {sorted(s.values)})")
+
+
+@secrets_app.command("list")
+def secrets_list(namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
+
+    for s in controller_client().list_secrets(namespace):
+        console.print(f"{s['name']}  ({s['k8s_name']}, keys: {s['keys']})")
+
+
+@secrets_app.command("delete")
+def secrets_delete(name: str,
+                   namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
+
+    controller_client().delete_secret(name, namespace)
+    console.print(f"[green]secret {name} deleted[/green]")
 
 
 @volumes_app.command("create")

@@ -163,13 +163,20 @@ class Module:
         self._exec_token = uuid.uuid4().hex
         md = self.metadata()
         self._sync_workdir(md)
+        # secrets must exist before pods reference them (reference:
+        # _upload_secrets_list before the manifest apply)
+        for s in (self.compute.secrets or []):
+            controller_client().put_secret(s, self.namespace)
         launch_id = uuid.uuid4().hex[:12]
         manifest = self.compute.to_manifest(self.name, username=config.username,
                                             module=self.pointers["name"])
+        service_config = {"kind": self.compute.kind}
+        if self.compute.endpoint is not None:
+            service_config["endpoint"] = self.compute.endpoint.to_service_config()
         resp = controller_client().deploy(
             name=self.name, namespace=self.namespace, manifest=manifest,
             metadata=md, launch_id=launch_id,
-            service_config={"kind": self.compute.kind},
+            service_config=service_config,
             timeout=self.compute.launch_timeout,
         )
         self.launch_id = resp.get("launch_id", launch_id)
@@ -191,7 +198,11 @@ class Module:
     def _base_url(self):
         from kubetorch_amd.globals import service_url
 
-        return service_url(self.name, self.namespace, self.service_hosts)
+        default = service_url(self.name, self.namespace, self.service_hosts)
+        ep = self.compute.endpoint if self.compute else None
+        if ep is not None:
+            return ep.resolve(default_url=default, hosts=self.service_hosts)
+        return default
 
     @property
     def http(self) -> HTTPClient:
@@ -263,6 +274,29 @@ class Module:
                                  timeout=timeout + 30)
         r.raise_for_status()
         return r.json()
+
+    def ssh(self, command=None, pod_index=0):
+        """Shell access to a service pod (reference: Compute.ssh). With
+        `command`: run it and return {returncode, stdout, stderr} (exec
+        route, token-gated). Without: attach an interactive shell —
+        kubectl exec -it in-cluster; on the local driver pods are local
+        processes, so a command is required."""
+        if command is not None:
+            return self.run_bash(command)
+        if self.compute is not None and self.compute.local:
+            raise RuntimeError(
+                "interactive ssh targets a cluster pod; local-driver pods "
+                "are subprocesses on this machine — pass command=... "
+                "instead")
+        import subprocess
+
+        pods = (self.workload() or {}).get("pods") or self.service_hosts
+        if not pods:
+            raise RuntimeError(f"no pods found for {self.name}")
+        pod_name = f"{self.name}-{pod_index}"
+        return subprocess.call(
+            ["kubectl", "-n", self.namespace, "exec", "-it",
+             pod_name, "--", "bash"])
 
     def pip_install(self, packages, extra_args=""):
         if isinstance(packages, str):

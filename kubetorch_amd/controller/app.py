@@ -296,6 +296,21 @@ async def deploy(request: Request):
             return JSONResponse(
                 {"error": f"driver apply failed: {e}"}, status_code=500
             )
+        if HUB.driver_name == "k8s" and manifest.get("kind") == "Deployment":
+            # route + discovery Services (the reference controller creates
+            # these; a custom endpoint selector narrows call routing)
+            from kubetorch_amd.provisioning.manifests import (
+                build_service_manifests,
+            )
+
+            ep = (body.get("service_config") or {}).get("endpoint") or {}
+            svc, headless = build_service_manifests(
+                name, ns, selector=ep.get("selector"))
+            for m in (svc, headless):
+                try:
+                    await asyncio.to_thread(HUB.driver.apply, m, ns)
+                except Exception:
+                    pass
 
     # hot-reload connected pods of this service and wait for acks
     pods = HUB.pods_for_service(name, ns)
@@ -317,6 +332,37 @@ async def deploy(request: Request):
 
     return {"ok": True, "launch_id": launch_id, "hosts": hosts,
             "reloaded_pods": acked, "driver": HUB.driver_name}
+
+
+@app.post("/controller/secrets/{ns}")
+async def put_secret(ns: str, request: Request):
+    """Create/update a secret through the control plane (reference parity:
+    kubernetes_secrets_client.py — the client never needs direct K8s
+    credentials). Body: {name, values, as_env, mount_path}."""
+    body = await request.json()
+    try:
+        await asyncio.to_thread(HUB.driver.apply_secret, body, ns)
+    except Exception as e:
+        return JSONResponse({"error": f"secret apply failed: {e}"},
+                            status_code=500)
+    return {"ok": True, "name": body.get("name")}
+
+
+@app.get("/controller/secrets/{ns}")
+def list_secrets(ns: str):
+    try:
+        return {"secrets": HUB.driver.list_secrets(ns)}
+    except Exception as e:
+        return JSONResponse({"error": str(e)}, status_code=500)
+
+
+@app.delete("/controller/secrets/{ns}/{name}")
+async def delete_secret(ns: str, name: str):
+    try:
+        await asyncio.to_thread(HUB.driver.delete_secret, name, ns)
+    except Exception as e:
+        return JSONResponse({"error": str(e)}, status_code=500)
+    return {"ok": True}
 
 
 @app.post("/controller/workload")

@@ -59,6 +59,7 @@ class LocalDriver:
         self._event_state = {}  # pod name -> last seen liveness
         self._master_ports = {}  # (ns, name) -> rendezvous port (stable
         # across pod replacement so re-joined ranks find the same master)
+        self.secrets = {}  # (ns, k8s_name) -> {name, values, as_env, mount_path}
 
     def _event(self, namespace, name, reason, message, pod=None):
         self.events.setdefault((namespace, name), []).append({
@@ -136,6 +137,7 @@ class LocalDriver:
                 env["KT_PEERS_URL"] = self.controller_url
             for k, v in (md.get("env") or {}).items():
                 env[k] = str(v)
+            self._inject_secrets(manifest, namespace, env)
             # module metadata -> env contract (same as controller push)
             if md.get("file_path"):
                 env[C.ENV_FILE_PATH] = md["file_path"]
@@ -178,6 +180,65 @@ class LocalDriver:
             p.kill()
         self.events.pop((namespace, name), None)
         self._master_ports.pop((namespace, name), None)
+
+    # -- secrets (reference: controller-side kubernetes_secrets_client) ----
+    def apply_secret(self, spec, namespace):
+        k8s_name = spec.get("k8s_name") or \
+            f"kt-secret-{spec['name']}".lower().replace("_", "-")
+        self.secrets[(namespace, k8s_name)] = dict(spec, k8s_name=k8s_name)
+
+    def list_secrets(self, namespace):
+        return [{"name": s["name"], "k8s_name": k,
+                 "keys": sorted(s.get("values", {}))}
+                for (ns, k), s in self.secrets.items() if ns == namespace]
+
+    def delete_secret(self, name, namespace):
+        k8s_name = f"kt-secret-{name}".lower().replace("_", "-")
+        self.secrets.pop((namespace, k8s_name), None)
+        self.secrets.pop((namespace, name), None)
+
+    def _inject_secrets(self, manifest, namespace, env):
+        """Materialize the pod template's secret references into the local
+        pod's environment: envFrom secretRefs become env vars; secret
+        volumes are written under a scratch dir with
+        KT_SECRET_MOUNT_<NAME>=<dir> pointing at it (the stand-in for the
+        in-cluster file mount)."""
+        try:
+            spec = (manifest.get("spec", {}).get("template", {})
+                    .get("spec", {})) or {}
+            containers = spec.get("containers") or [{}]
+            c = containers[0]
+            for ref in c.get("envFrom", []) or []:
+                sname = ref.get("secretRef", {}).get("name")
+                sec = self.secrets.get((namespace, sname))
+                if sec:
+                    for k, v in sec.get("values", {}).items():
+                        env[k] = str(v)
+            vol_secrets = {v["name"]: v["secret"]["secretName"]
+                           for v in spec.get("volumes", []) or []
+                           if "secret" in v}
+            for vm in c.get("volumeMounts", []) or []:
+                sname = vol_secrets.get(vm["name"])
+                if not sname:
+                    continue
+                sec = self.secrets.get((namespace, sname))
+                if not sec:
+                    continue
+                import tempfile
+
+                d = os.path.join(tempfile.gettempdir(), "kt-local-secrets",
+                                 namespace, sname)
+                os.makedirs(d, exist_ok=True)
+                for k, v in sec.get("values", {}).items():
+                    path = os.path.join(d, k)
+                    with open(path, "w") as f:
+                        f.write(str(v))
+                    os.chmod(path, 0o600)
+                env_key = "KT_SECRET_MOUNT_" + \
+                    sec["name"].upper().replace("-", "_")
+                env[env_key] = d
+        except Exception:
+            pass
 
     def pods(self, name, namespace):
         return [p.host for p in self.services.get((namespace, name), [])
@@ -226,6 +287,41 @@ class K8sDriver:
             if ip:
                 hosts.append(f"{ip}:{C.SERVER_PORT}")
         return hosts
+
+    def apply_secret(self, spec, namespace):
+        import base64
+
+        k8s_name = spec.get("k8s_name") or \
+            f"kt-secret-{spec['name']}".lower().replace("_", "-")
+        manifest = {
+            "apiVersion": "v1", "kind": "Secret",
+            "metadata": {"name": k8s_name, "namespace": namespace},
+            "type": "Opaque",
+            "data": {k: base64.b64encode(str(v).encode()).decode()
+                     for k, v in (spec.get("values") or {}).items()},
+        }
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "apply", "-f", "-"],
+            input=json.dumps(manifest).encode(), check=True,
+            capture_output=True)
+
+    def list_secrets(self, namespace):
+        out = subprocess.run(
+            [self.kubectl, "-n", namespace, "get", "secrets", "-o", "json"],
+            check=True, capture_output=True)
+        items = json.loads(out.stdout).get("items", [])
+        return [{"name": it["metadata"]["name"].replace("kt-secret-", "", 1),
+                 "k8s_name": it["metadata"]["name"],
+                 "keys": sorted((it.get("data") or {}))}
+                for it in items
+                if it["metadata"]["name"].startswith("kt-secret-")]
+
+    def delete_secret(self, name, namespace):
+        k8s_name = f"kt-secret-{name}".lower().replace("_", "-")
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "delete", "secret", k8s_name,
+             "--ignore-not-found"],
+            check=True, capture_output=True)
 
     def get_events(self, name, namespace, since=0.0):
         """K8s events for the service's objects (scheduling, image pulls,

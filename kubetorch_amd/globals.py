@@ -68,6 +68,25 @@ class ControllerClient:
             raise LaunchError(str(data))
         return data
 
+    def put_secret(self, secret, namespace):
+        """Create/update a Secret through the controller (reference:
+        controller-side kubernetes_secrets_client)."""
+        r = self._request(
+            "POST", f"/controller/secrets/{namespace}",
+            json={"name": secret.name, "k8s_name": secret.k8s_name,
+                  "values": secret.values, "as_env": secret.as_env,
+                  "mount_path": secret.mount_path})
+        r.raise_for_status()
+        return r.json()
+
+    def list_secrets(self, namespace):
+        return self._request(
+            "GET", f"/controller/secrets/{namespace}").json().get("secrets", [])
+
+    def delete_secret(self, name, namespace):
+        return self._request(
+            "DELETE", f"/controller/secrets/{namespace}/{name}").json()
+
     def register_workload(self, name, namespace, **body):
         return self._request(
             "POST", "/controller/workload",

@@ -394,16 +394,20 @@ __global__ void adamw_kernel(u16* __restrict__ p, const u16* __restrict__ g,
                              long n, float lr, float beta1, float beta2,
                              float eps, float wd, float bc1, float bc2,
                              float grad_scale) {
+  // Pure streaming op (every byte touched exactly once): nontemporal
+  // loads/stores bypass L2 retention so the 22 B/param of HBM traffic
+  // doesn't thrash the cache the overlapped backward is using, and the
+  // stores avoid read-for-ownership (measured 4.6 TB/s before).
   const long nvec = n >> 3;
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < nvec;
        t += (long)gridDim.x * blockDim.x) {
     const size_t i = (size_t)t * 8;
-    vec8u pv = *reinterpret_cast<const vec8u*>(p + i);
-    vec8u gv = *reinterpret_cast<const vec8u*>(g + i);
-    vec4f m0 = *reinterpret_cast<const vec4f*>(m + i);
-    vec4f m1 = *reinterpret_cast<const vec4f*>(m + i + 4);
-    vec4f v0 = *reinterpret_cast<const vec4f*>(v + i);
-    vec4f v1 = *reinterpret_cast<const vec4f*>(v + i + 4);
+    vec8u pv = __builtin_nontemporal_load(reinterpret_cast<const vec8u*>(p + i));
+    vec8u gv = __builtin_nontemporal_load(reinterpret_cast<const vec8u*>(g + i));
+    vec4f m0 = __builtin_nontemporal_load(reinterpret_cast<const vec4f*>(m + i));
+    vec4f m1 = __builtin_nontemporal_load(reinterpret_cast<const vec4f*>(m + i + 4));
+    vec4f v0 = __builtin_nontemporal_load(reinterpret_cast<const vec4f*>(v + i));
+    vec4f v1 = __builtin_nontemporal_load(reinterpret_cast<const vec4f*>(v + i + 4));
     vec8u po;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -420,11 +424,11 @@ __global__ void adamw_kernel(u16* __restrict__ p, const u16* __restrict__ g,
       if (j < 4) { m0[j] = mj; v0[j] = vj; }
       else { m1[j - 4] = mj; v1[j - 4] = vj; }
     }
-    *reinterpret_cast<vec8u*>(p + i) = po;
-    *reinterpret_cast<vec4f*>(m + i) = m0;
-    *reinterpret_cast<vec4f*>(m + i + 4) = m1;
-    *reinterpret_cast<vec4f*>(v + i) = v0;
-    *reinterpret_cast<vec4f*>(v + i + 4) = v1;
+    __builtin_nontemporal_store(po, reinterpret_cast<vec8u*>(p + i));
+    __builtin_nontemporal_store(m0, reinterpret_cast<vec4f*>(m + i));
+    __builtin_nontemporal_store(m1, reinterpret_cast<vec4f*>(m + i + 4));
+    __builtin_nontemporal_store(v0, reinterpret_cast<vec4f*>(v + i));
+    __builtin_nontemporal_store(v1, reinterpret_cast<vec4f*>(v + i + 4));
   }
 }
 

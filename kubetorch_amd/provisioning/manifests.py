@@ -130,20 +130,25 @@ def build_deployment_manifest(service_name, namespace, image, replicas=1,
     }
 
 
-def build_service_manifests(service_name, namespace, port=C.SERVER_PORT):
-    """ClusterIP service + headless service (peer discovery)."""
+def build_service_manifests(service_name, namespace, port=C.SERVER_PORT,
+                            selector=None):
+    """ClusterIP service + headless service (peer discovery). A custom
+    `selector` routes call traffic to a pod subset (e.g. the Ray head —
+    reference: Endpoint(selector=...)); the headless service keeps the
+    full pod set so rank discovery still sees every worker."""
     base = {
         "apiVersion": "v1",
         "kind": "Service",
         "metadata": {"name": service_name, "namespace": namespace,
                      "labels": _labels(service_name)},
         "spec": {
-            "selector": {C.SERVICE_LABEL: service_name},
+            "selector": dict(selector or {C.SERVICE_LABEL: service_name}),
             "ports": [{"port": port, "targetPort": port}],
         },
     }
     headless = copy.deepcopy(base)
     headless["metadata"]["name"] = f"{service_name}-headless"
+    headless["spec"]["selector"] = {C.SERVICE_LABEL: service_name}
     headless["spec"]["clusterIP"] = "None"
     return base, headless
 

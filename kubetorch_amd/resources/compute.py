@@ -149,3 +149,65 @@ class Compute:
         if img is not None and hasattr(img, "contents"):
             return img.contents()
         return ""
+
+    # -- getter/setter surface (reference: compute.py:934-956 — setters
+    # -- also mutate a BYO manifest in place so from_manifest computes stay
+    # -- editable without rebuilding) ---------------------------------------
+    def _pod_spec_of_raw(self):
+        m = self._raw_manifest
+        if m is None:
+            return None
+        spec = m.get("spec", {})
+        if "template" in spec:          # Deployment
+            return spec["template"].get("spec")
+        if "pytorchReplicaSpecs" in spec:  # PyTorchJob: first replica spec
+            for rs in spec["pytorchReplicaSpecs"].values():
+                return rs.get("template", {}).get("spec")
+        return None
+
+    def _patch_resource(self, key, value, request_only=False):
+        pod = self._pod_spec_of_raw()
+        if not pod:
+            return
+        for c in pod.get("containers", []):
+            res = c.setdefault("resources", {})
+            res.setdefault("requests", {})[key] = str(value)
+            if not request_only:
+                res.setdefault("limits", {})[key] = str(value)
+
+    def set_gpus(self, gpus):
+        self.gpus = gpus
+        self._patch_resource(C.GPU_RESOURCE, gpus)
+        return self
+
+    def set_cpus(self, cpus):
+        self.cpus = cpus
+        self._patch_resource("cpu", cpus, request_only=True)
+        return self
+
+    def set_memory(self, memory):
+        self.memory = memory
+        self._patch_resource("memory", memory)
+        return self
+
+    def set_gpu_type(self, gpu_type):
+        self.gpu_type = gpu_type
+        pod = self._pod_spec_of_raw()
+        if pod is not None:
+            pod.setdefault("nodeSelector", {})[C.GPU_PRODUCT_LABEL] = gpu_type
+        return self
+
+    def set_image(self, image):
+        self.image = image
+        pod = self._pod_spec_of_raw()
+        if pod is not None:
+            for c in pod.get("containers", []):
+                c["image"] = self.image_id()
+        return self
+
+    def set_replicas(self, replicas):
+        self.replicas = replicas
+        if self._raw_manifest is not None and \
+                "replicas" in self._raw_manifest.get("spec", {}):
+            self._raw_manifest["spec"]["replicas"] = replicas
+        return self

@@ -70,3 +70,15 @@ def deploy_child_and_call():
 
 def pod_name():
     return os.environ.get("POD_NAME", "unknown")
+
+
+def read_env(key):
+    return os.environ.get(key)
+
+
+def read_secret_file(secret_name, key):
+    d = os.environ.get(f"KT_SECRET_MOUNT_{secret_name}")
+    if not d:
+        return None
+    with open(os.path.join(d, key)) as f:
+        return f.read()

