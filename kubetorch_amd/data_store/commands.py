@@ -88,8 +88,56 @@ def put_delta(key, src, timeout=C.HTTP_TIMEOUT):
             "removed": len(removed)}
 
 
-def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True):
-    """Store a file/dir (by path) or publish GPU tensors under `key`."""
+def _self_host():
+    host = os.environ.get("KT_SELF_HOST")
+    if host:
+        return host
+    import socket
+
+    try:
+        ip = socket.gethostbyname(socket.gethostname())
+    except socket.gaierror:
+        ip = "127.0.0.1"
+    return f"{ip}:{os.environ.get('KT_SERVER_PORT', C.SERVER_PORT)}"
+
+
+def localfs_registry_path():
+    """Per-pod registry of locally-served keys; the pod's http_server
+    process reads it to answer /localfiles/{key} (worker processes and
+    the server share the pod filesystem)."""
+    import tempfile
+
+    tag = os.environ.get("KT_SERVER_PORT", str(C.SERVER_PORT))
+    return os.path.join(tempfile.gettempdir(), f"kt-localfs-{tag}.json")
+
+
+def _register_local_serve(key, path):
+    import json
+
+    reg_path = localfs_registry_path()
+    reg = {}
+    if os.path.exists(reg_path):
+        try:
+            with open(reg_path) as f:
+                reg = json.load(f)
+        except (OSError, ValueError):
+            reg = {}
+    reg[key.strip("/")] = path
+    tmp = reg_path + ".tmp"
+    with open(tmp, "w") as f:
+        json.dump(reg, f)
+    os.replace(tmp, reg_path)
+
+
+def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True,
+        locale="store"):
+    """Store a file/dir (by path) or publish GPU tensors under `key`.
+
+    locale="store" copies the data into the namespace store pod;
+    locale="local" registers the key WITHOUT copying (zero-copy): this
+    pod serves the file on demand and getters fetch it p2p, falling back
+    to the store (and de-registering the source) if this pod dies.
+    (Reference parity: data_store_client.py local locale, :175/:325.)"""
     if _is_gpu_data(src):
         from kubetorch_amd.data_store import gpu_store
 
@@ -98,6 +146,17 @@ def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True):
     if not os.path.exists(src):
         raise FileNotFoundError(src)
     url = _store_url()
+    if locale == "local":
+        if url is None:
+            raise ValueError("locale='local' needs a store service for the "
+                             "key registry (KT_STORE_URL)")
+        src_abs = os.path.abspath(src)
+        _register_local_serve(key, src_abs)
+        httpx.post(f"{url}/meta/localfs/{key}",
+                   json={"host": _self_host(), "path": src_abs,
+                         "is_dir": os.path.isdir(src_abs)},
+                   timeout=timeout).raise_for_status()
+        return {"key": key, "locale": "local", "host": _self_host()}
     if url is not None and _delta and os.path.isdir(src):
         return put_delta(key, src, timeout=timeout)
     if url is None:
@@ -154,6 +213,17 @@ def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
             os.makedirs(os.path.dirname(dest) or ".", exist_ok=True)
             shutil.copy2(srcp, dest)
         return dest
+    # p2p-first: a local-locale source serves the key directly; if the
+    # source is gone, de-register it and fall back to the store copy
+    # (reference: get w/ remove_source retry, data_store_client.py:325)
+    m = httpx.get(f"{url}/meta/localfs/{key}", timeout=timeout)
+    if m.status_code == 200:
+        source = m.json().get("host")
+        try:
+            return _fetch_from_peer(f"http://{source}", key, dest,
+                                    timeout, route="localfiles")
+        except (httpx.HTTPError, OSError):
+            httpx.delete(f"{url}/meta/localfs/{key}", timeout=timeout)
     r = httpx.get(f"{url}/files/{key}", timeout=timeout)
     if r.status_code == 404:
         raise KeyError(f"no such key: {key}")
@@ -170,8 +240,8 @@ def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
     return dest
 
 
-def _fetch_from_peer(source, key, dest, timeout):
-    r = httpx.get(f"{source}/files/{key}", timeout=timeout)
+def _fetch_from_peer(source, key, dest, timeout, route="files"):
+    r = httpx.get(f"{source}/{route}/{key}", timeout=timeout)
     r.raise_for_status()
     dest = os.path.expanduser(dest or os.path.basename(key))
     if r.headers.get("X-KT-Tar") == "1":

@@ -442,6 +442,39 @@ async def exec_route(request: Request):
             "stderr": res.stderr[-20000:]}
 
 
+@app.get("/localfiles/{key:path}")
+def serve_local_file(key: str):
+    """Serve a key this pod registered with locale='local' (zero-copy
+    put: the data never moved to the store; peers fetch it from here).
+    Only explicitly registered keys are served — this is not a general
+    file server."""
+    import io as _io
+    import tarfile as _tarfile
+
+    from kubetorch_amd.data_store.commands import localfs_registry_path
+
+    reg_path = localfs_registry_path()
+    reg = {}
+    if os.path.exists(reg_path):
+        try:
+            with open(reg_path) as f:
+                reg = json.load(f)
+        except (OSError, ValueError):
+            reg = {}
+    path = reg.get(key.strip("/"))
+    if not path or not os.path.exists(path):
+        return JSONResponse({"error": f"key {key!r} not registered here"},
+                            status_code=404)
+    if os.path.isdir(path):
+        buf = _io.BytesIO()
+        with _tarfile.open(fileobj=buf, mode="w:gz") as tar:
+            tar.add(path, arcname=".")
+        return Response(buf.getvalue(), media_type="application/gzip",
+                        headers={"X-KT-Tar": "1"})
+    with open(path, "rb") as f:
+        return Response(f.read(), media_type="application/octet-stream")
+
+
 @app.get("/app/status")
 def app_status():
     proc = STATE.get("app_proc")
