@@ -76,6 +76,11 @@ def _flatten(sd):
     return {k: sd[k] for k in sorted(sd)}
 
 
+# key -> (flat buffer, signature): packed publish buffers stay allocated
+# and IPC-registered so repeat publishes only run the pack kernel
+_PACKED_CACHE = {}
+
+
 # -- public api ----------------------------------------------------------------
 def put(key, src, window=None):
     """Publish a tensor or state dict. Tensors stay in the owner's memory
@@ -102,10 +107,23 @@ def put(key, src, window=None):
         if len(dtypes) != 1:
             raise ValueError("packed mode requires a single dtype")
         # one pack kernel instead of torch.cat's per-tensor copies; segment
-        # starts 16B-aligned so the kernel runs pure uint4 vectors
-        flat, _ = kt_ops.pack_tensors(
-            [t.contiguous() for t in sd.values()])
-        cli.register(f"{key}/__packed__", flat)
+        # starts 16B-aligned so the kernel runs pure uint4 vectors.
+        # Re-publishes of the same key/shapes (the RL weight-sync hot
+        # loop) reuse the cached flat buffer AND its hipIpc registration:
+        # the r01-measured 14 GB/s packed put was dominated by the fresh
+        # 256 MB allocation + hipIpcGetMemHandle per publish, not the copy.
+        sig = (tuple(t.numel() for t in sd.values()),
+               next(iter(dtypes)), len(sd))
+        cached = _PACKED_CACHE.get(key)
+        if cached is not None and cached[1] == sig:
+            flat = cached[0]
+            kt_ops.pack_tensors([t.contiguous() for t in sd.values()],
+                                flat=flat)
+        else:
+            flat, _ = kt_ops.pack_tensors(
+                [t.contiguous() for t in sd.values()])
+            cli.register(f"{key}/__packed__", flat)
+            _PACKED_CACHE[key] = (flat, sig)
     else:
         for sub, t in sd.items():
             cli.register(f"{key}/{sub}", t)
