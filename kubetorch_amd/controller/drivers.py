@@ -251,20 +251,47 @@ class LocalDriver:
 
 class K8sDriver:
     """Applies manifests via kubectl (the controller pod has RBAC for this;
-    reference ships this logic inside its closed-source controller image)."""
+    reference ships this logic inside its closed-source controller image).
+    apply() verifies the rollout actually converged (kubectl rollout
+    status for Deployments, pod readiness for job kinds) instead of
+    fire-and-forgetting the manifest."""
 
-    def __init__(self, kubectl="kubectl"):
+    ROLLOUT_KINDS = {"Deployment", "StatefulSet", "DaemonSet"}
+
+    def __init__(self, kubectl="kubectl",
+                 rollout_timeout=None):
         self.kubectl = kubectl
+        self.rollout_timeout = rollout_timeout or int(
+            os.environ.get("KT_ROLLOUT_TIMEOUT", "600"))
 
     def available(self):
         return shutil.which(self.kubectl) is not None
 
     def apply(self, manifest, namespace, metadata=None, launch_id=None):
         payload = json.dumps(manifest)
-        subprocess.run(
+        r = subprocess.run(
             [self.kubectl, "-n", namespace, "apply", "-f", "-"],
-            input=payload.encode(), check=True, capture_output=True,
+            input=payload.encode(), capture_output=True,
         )
+        if r.returncode != 0:
+            raise RuntimeError(
+                f"kubectl apply failed rc={r.returncode}: "
+                f"{r.stderr.decode(errors='replace')[-800:]}")
+        kind = manifest.get("kind", "")
+        name = manifest.get("metadata", {}).get("name", "")
+        if kind in self.ROLLOUT_KINDS:
+            rs = subprocess.run(
+                [self.kubectl, "-n", namespace, "rollout", "status",
+                 f"{kind.lower()}/{name}",
+                 f"--timeout={self.rollout_timeout}s"],
+                capture_output=True,
+            )
+            if rs.returncode != 0:
+                raise RuntimeError(
+                    f"rollout of {kind}/{name} did not converge within "
+                    f"{self.rollout_timeout}s: "
+                    f"{rs.stderr.decode(errors='replace')[-800:]}")
+            return self.pods(name, namespace)
         return []
 
     def delete(self, name, namespace, kind="deployment"):
