@@ -27,6 +27,49 @@ def parse_steps(contents: str):
     return steps
 
 
+def _pip_requirements_satisfied(cmd):
+    """True if `cmd` is a plain `pip install` whose requirements are all
+    already importable at a matching version (reference parity: the
+    pip-freeze diff that keeps warm reloads from re-resolving installed
+    deps). Anything unparseable (urls, -r files, editable installs,
+    options with values) conservatively returns False."""
+    try:
+        toks = shlex.split(cmd)
+    except ValueError:
+        return False
+    if "install" not in toks or not any("pip" in t for t in toks):
+        return False
+    specs = []
+    for t in toks[toks.index("install") + 1:]:
+        if t.startswith("-"):
+            return False  # options may change resolution; just run it
+        if any(c in t for c in ":/@"):
+            return False  # url / path / vcs spec
+        specs.append(t)
+    if not specs:
+        return False
+    from importlib import metadata
+
+    for spec in specs:
+        for op in ("==", ">=", "<=", "~=", "!=", ">", "<"):
+            if op in spec:
+                name, _, want = spec.partition(op)
+                break
+        else:
+            name, op, want = spec, None, None
+        name = name.strip().split("[")[0]
+        try:
+            have = metadata.version(name)
+        except metadata.PackageNotFoundError:
+            return False
+        if op == "==" and have != want:
+            return False
+        if op in (">=", "<=", "~=", "!=", ">", "<") and op != "==":
+            # only exact pins are verified offline; ranges -> run pip
+            return False
+    return True
+
+
 def run_step(kind, payload, app_state=None):
     if kind == "FROM":
         return  # base image is fixed at pod creation
@@ -38,6 +81,9 @@ def run_step(kind, payload, app_state=None):
         cmd = payload.replace(f"${PIP_CMD_VAR}",
                               os.environ.get(PIP_CMD_VAR, _default_pip()))
         cmd = os.path.expandvars(cmd)
+        if _pip_requirements_satisfied(cmd):
+            print(f"[image-setup] skip (already installed): {cmd}")
+            return
         res = subprocess.run(["bash", "-lc", cmd], capture_output=True,
                              text=True)
         if res.returncode != 0:

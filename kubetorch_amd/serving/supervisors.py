@@ -285,25 +285,26 @@ class SPMDSupervisor(DistributedSupervisor):
             self.monitor.rebase(hosts)
         aborted = []
         self.monitor.subscribe(aborted.append)
-        executor = ThreadPoolExecutor(max_workers=min(64, len(hosts) + 4))
+        executor = ThreadPoolExecutor(max_workers=4)
         try:
             remote_hosts = hosts[1:]
             futs = []
-            # remote fan-out (flat; tree for very large worlds)
-            from kubetorch_amd.serving.remote_pool import call_worker_subcall
+            # remote fan-out on the asyncio engine (200-concurrency cap,
+            # one loop thread regardless of world size); flat, or tree for
+            # very large worlds
+            from kubetorch_amd.serving.remote_pool import fanout
 
+            eng = fanout()
             if len(hosts) >= TREE_THRESHOLD:
                 children = remote_hosts[:TREE_FANOUT]
                 subtrees = [remote_hosts[i::TREE_FANOUT] for i in range(TREE_FANOUT)]
                 for child, subtree in zip(children, subtrees):
-                    futs.append(executor.submit(
-                        call_worker_subcall, child, body, method, hosts,
-                        [child] + subtree[1:], timeout))
+                    futs.append(eng.submit(child, body, method, hosts,
+                                           [child] + subtree[1:], timeout))
             else:
                 for h in remote_hosts:
-                    futs.append(executor.submit(
-                        call_worker_subcall, h, body, method, hosts, None,
-                        timeout))
+                    futs.append(eng.submit(h, body, method, hosts, None,
+                                           timeout))
             local_fut = executor.submit(
                 self._run_local_ranks, body, method, hosts, 0, timeout)
             futs.append(local_fut)

@@ -1,0 +1,75 @@
+"""AsyncFanout engine: 300 concurrent subcalls complete in ~2 slow-server
+rounds (concurrency cap 200), not serially — the scale property the
+reference gets from its asyncio remote_worker_pool subprocess."""
+import base64
+import pickle
+import socket
+import threading
+import time
+
+import pytest
+from fastapi import FastAPI, Request
+
+pytestmark = pytest.mark.flaky_retry
+
+stub = FastAPI()
+STATE = {"active": 0, "peak": 0, "lock": threading.Lock()}
+
+
+@stub.get("/health")
+def health():
+    return {"status": "ok"}
+
+
+@stub.post("/spmd/subcall")
+async def subcall(request: Request):
+    import asyncio
+
+    with STATE["lock"]:
+        STATE["active"] += 1
+        STATE["peak"] = max(STATE["peak"], STATE["active"])
+    try:
+        await asyncio.sleep(0.2)
+        return {"result": base64.b64encode(pickle.dumps(["ok"])).decode()}
+    finally:
+        with STATE["lock"]:
+            STATE["active"] -= 1
+
+
+@pytest.mark.timeout(120)
+def test_300_subcalls_run_concurrently():
+    import uvicorn
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    server = uvicorn.Server(uvicorn.Config(stub, host="127.0.0.1", port=port,
+                                           log_level="error"))
+    threading.Thread(target=server.run, daemon=True).start()
+    import httpx
+
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        try:
+            if httpx.get(f"http://127.0.0.1:{port}/health",
+                         timeout=1).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.05)
+
+    from kubetorch_amd.serving.remote_pool import (MAX_CONCURRENT_SUBCALLS,
+                                                   fanout)
+
+    eng = fanout()
+    host = f"127.0.0.1:{port}"
+    t0 = time.perf_counter()
+    futs = [eng.submit(host, "", None, [host], None, 30) for _ in range(300)]
+    results = [f.result(60) for f in futs]
+    wall = time.perf_counter() - t0
+    assert all(r == ["ok"] for r in results)
+    # serial would be 60 s; 200-wide concurrency -> 2 rounds of 0.2 s + ovh
+    assert wall < 10.0, f"fan-out too slow ({wall:.1f}s): not concurrent"
+    assert STATE["peak"] > 50, f"peak concurrency only {STATE['peak']}"
+    assert STATE["peak"] <= MAX_CONCURRENT_SUBCALLS + 1
+    server.should_exit = True

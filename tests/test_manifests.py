@@ -119,6 +119,24 @@ def test_image_setup_interpreter_caching(tmp_path, monkeypatch):
     assert marker.read_text().count("one") == 1
 
 
+def test_pip_freeze_diff_skips_satisfied_installs():
+    """Reference parity: pip-freeze diff — a pip install whose exact pins
+    are already present is skipped; ranges/urls/options always run."""
+    from importlib import metadata
+
+    from kubetorch_amd.serving.image_setup import _pip_requirements_satisfied
+
+    have = metadata.version("pytest")
+    assert _pip_requirements_satisfied(f"pip install pytest=={have}")
+    assert _pip_requirements_satisfied("python -m pip install pytest numpy")
+    assert not _pip_requirements_satisfied("pip install not-a-real-pkg-xyz")
+    assert not _pip_requirements_satisfied("pip install pytest==0.0.1")
+    assert not _pip_requirements_satisfied("pip install pytest>=1.0")  # range
+    assert not _pip_requirements_satisfied("pip install -r reqs.txt")
+    assert not _pip_requirements_satisfied("pip install git+https://x/y.git")
+    assert not _pip_requirements_satisfied("echo not pip at all")
+
+
 def test_decorators_build_module():
     @kt.compute(cpus=1)
     @kt.distribute("pytorch", workers=2, num_proc=1)
