@@ -348,6 +348,27 @@ def debug(service: str, port: int = typer.Option(None),
     attach(host, port or 4444)
 
 
+@app.command("tunnel")
+def tunnel(service: str, port: int = typer.Argument(8080),
+           local_port: int = typer.Option(0, "--local-port"),
+           namespace: str = typer.Option("default", "-n")):
+    """Bridge a local TCP port to an in-cluster service through the
+    controller's public port (firewall traversal; no kubectl needed)."""
+    from kubetorch_amd.client.tunnel import TcpTunnel
+
+    t = TcpTunnel(service, port, namespace=namespace,
+                  local_port=local_port).start()
+    console.print(f"[green]tunnel up[/green]: 127.0.0.1:{t.local_port} -> "
+                  f"{service}:{port} (ctrl-c to stop)")
+    import time as _t
+
+    try:
+        while True:
+            _t.sleep(3600)
+    except KeyboardInterrupt:
+        t.stop()
+
+
 @app.command("port-forward")
 def port_forward(target: str, ports: str,
                  namespace: str = typer.Option(None, "-n")):

@@ -365,6 +365,90 @@ async def delete_secret(ns: str, name: str):
     return {"ok": True}
 
 
+# -- TCP tunnel for out-of-cluster clients (reference parity:
+# -- data_store/websocket_tunnel.py — firewall traversal through the one
+# -- public controller port). No WS stack in this image, so the bridge is
+# -- two chunked-HTTP streams per connection: POST .../up carries
+# -- client->service bytes, GET .../down streams service->client.
+TUNNELS = {}
+
+
+@app.post("/controller/tunnel/open")
+async def tunnel_open(request: Request):
+    import socket as _socket
+
+    body = await request.json()
+    ns = body.get("namespace", "default")
+    service = body["service"]
+    port = int(body["port"])
+    host = None
+    if ":" in service:  # explicit host:port target (BYO)
+        host, _, p = service.partition(":")
+        port = int(p)
+    elif HUB.driver_name == "local":
+        pods = HUB.driver.pods(service, ns)
+        if not pods:
+            return JSONResponse({"error": f"no pods for {service}"},
+                                status_code=404)
+        host, _, p = pods[0].partition(":")
+        port = int(p or port)
+    else:
+        host = f"{service}.{ns}.svc.cluster.local"
+    try:
+        sock = _socket.create_connection((host, port), timeout=10)
+    except OSError as e:
+        return JSONResponse({"error": f"connect {host}:{port}: {e}"},
+                            status_code=502)
+    tid = uuid.uuid4().hex
+    TUNNELS[tid] = {"sock": sock, "created": time.time()}
+    return {"tunnel_id": tid, "target": f"{host}:{port}"}
+
+
+@app.post("/controller/tunnel/{tid}/up")
+async def tunnel_up(tid: str, request: Request):
+    t = TUNNELS.get(tid)
+    if t is None:
+        return JSONResponse({"error": "no such tunnel"}, status_code=404)
+    sock = t["sock"]
+    try:
+        async for chunk in request.stream():
+            if chunk:
+                await asyncio.to_thread(sock.sendall, chunk)
+    except Exception:
+        pass
+    try:
+        sock.shutdown(1)  # half-close: service sees EOF upstream
+    except OSError:
+        pass
+    return {"ok": True}
+
+
+@app.get("/controller/tunnel/{tid}/down")
+async def tunnel_down(tid: str):
+    from fastapi.responses import StreamingResponse
+
+    t = TUNNELS.get(tid)
+    if t is None:
+        return JSONResponse({"error": "no such tunnel"}, status_code=404)
+    sock = t["sock"]
+
+    async def gen():
+        try:
+            while True:
+                data = await asyncio.to_thread(sock.recv, 65536)
+                if not data:
+                    break
+                yield data
+        finally:
+            TUNNELS.pop(tid, None)
+            try:
+                sock.close()
+            except OSError:
+                pass
+
+    return StreamingResponse(gen(), media_type="application/octet-stream")
+
+
 @app.post("/controller/workload")
 async def register_workload(request: Request):
     body = await request.json()
