@@ -398,16 +398,81 @@ class RaySupervisor(ExecutionSupervisor):
         super().cleanup()
 
 
-class MonarchSupervisor(ExecutionSupervisor):
-    """Monarch actor meshes are not available in the MI355X image (no
-    monarch wheel for ROCm yet); fail loudly with guidance rather than
-    silently degrading. (Reference: serving/monarch_supervisor.py.)"""
+class MonarchSupervisor(DistributedSupervisor):
+    """Single-controller actor-mesh launcher (reference parity:
+    serving/monarch_supervisor.py). Every pod runs a `process_allocator`
+    service; the controller pod (rank 0) exposes the full allocator
+    address list as KT_MONARCH_HOSTS so user code can build a
+    RemoteAllocator over all pods' meshes. The monarch wheel has no ROCm
+    build in this image yet, so the allocator binary is resolved from
+    KT_MONARCH_ALLOCATOR / PATH and a clear error is raised when absent
+    (user code still needs `import monarch` to drive meshes)."""
 
-    def __init__(self, **kw):
-        raise NotImplementedError(
-            "Monarch is not available on this image; use "
-            "distribute('pytorch'|'spmd'|'ray') instead."
-        )
+    ALLOCATOR_PORT = 26600
+
+    def __init__(self, num_workers=None, num_proc=None, **kw):
+        import shutil
+        import subprocess
+
+        binary = os.environ.get("KT_MONARCH_ALLOCATOR") or \
+            shutil.which("process_allocator")
+        if not binary:
+            raise NotImplementedError(
+                "monarch's process_allocator binary is not on this image "
+                "(no ROCm monarch wheel yet); install torchmonarch or set "
+                "KT_MONARCH_ALLOCATOR, or use "
+                "distribute('pytorch'|'spmd'|'ray')")
+        self.allocator_port = int(os.environ.get("KT_MONARCH_PORT",
+                                                 self.ALLOCATOR_PORT))
+        self._alloc_proc = subprocess.Popen(
+            [binary, f"--port={self.allocator_port}",
+             "--program=monarch_bootstrap"])
+        self._wait_allocator()
+        # single controller process drives the meshes (Ray-style)
+        super().__init__(num_proc=1, num_workers=num_workers,
+                         quorum_timeout=kw.get("quorum_timeout",
+                                               C.QUORUM_TIMEOUT))
+
+    def _wait_allocator(self, timeout=60):
+        import socket
+        import time
+
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            if self._alloc_proc.poll() is not None:
+                raise RuntimeError(
+                    f"process_allocator exited rc={self._alloc_proc.returncode}")
+            try:
+                s = socket.create_connection(("127.0.0.1",
+                                              self.allocator_port), 1)
+                s.close()
+                return
+            except OSError:
+                time.sleep(0.25)
+        raise RuntimeError(
+            f"process_allocator not live on :{self.allocator_port}")
+
+    def call(self, args=(), kwargs=None, method=None, timeout=None,
+             serialized_body=None, **_):
+        body = serialized_body or _encode_call(args, kwargs)
+        hosts = self.worker_hosts()
+        allocators = ",".join(
+            f"{h.split(':')[0]}:{self.allocator_port}" for h in sorted(hosts))
+        env = {"KT_MONARCH_HOSTS": allocators,
+               "MONARCH_ALLOCATOR_PORT": str(self.allocator_port)}
+        resp = self.pool.submit(0, body, method=method, env=env).result(
+            timeout or C.HTTP_TIMEOUT * 10)
+        return _decode_resp(resp)
+
+    def cleanup(self):
+        if getattr(self, "_alloc_proc", None) is not None \
+                and self._alloc_proc.poll() is None:
+            self._alloc_proc.terminate()
+            try:
+                self._alloc_proc.wait(5)
+            except Exception:
+                self._alloc_proc.kill()
+        super().cleanup()
 
 
 def supervisor_factory(distribution_type=None, **kw):
@@ -423,5 +488,7 @@ def supervisor_factory(distribution_type=None, **kw):
     if distribution_type == "ray":
         return RaySupervisor()
     if distribution_type == "monarch":
-        return MonarchSupervisor()
+        return MonarchSupervisor(**{k: v for k, v in kw.items()
+                                    if k in ("num_workers", "num_proc",
+                                             "quorum_timeout")})
     return SPMDSupervisor(framework=distribution_type, **kw)
