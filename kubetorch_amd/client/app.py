@@ -52,19 +52,41 @@ class App(Module):
             super()._wait_ready(timeout=timeout or 900, reloaded=reloaded)
 
 
-    def wait(self, timeout=900, poll=1.0):
-        """Foreground mode: block until the app's pods exit (reference:
-        App deploy --follow / _wait_for_app_exit). Returns True if the
-        app finished within the timeout."""
+    def wait(self, timeout=900, poll=1.0, follow=True, printer=print):
+        """Foreground mode: block until the app's pods exit, streaming the
+        app's stdout/stderr while waiting (reference: App deploy --follow
+        / _wait_for_app_exit). Returns True if the app finished within
+        the timeout."""
         import time as _t
 
         from kubetorch_amd.globals import controller_client
 
+        cc = controller_client()
+        offset = 0
         deadline = _t.time() + timeout
+
+        def drain():
+            nonlocal offset
+            if not follow:
+                return
+            try:
+                r = cc._request(
+                    "GET",
+                    f"/controller/podlogs/{self.namespace}/{self.name}",
+                    params={"offset": offset}).json()
+            except Exception:
+                return
+            if r.get("text"):
+                for line in r["text"].splitlines():
+                    printer(f"[{self.name}] {line}")
+            offset = r.get("offset", offset)
+
         while _t.time() < deadline:
-            w = controller_client().get_workload(self.name, self.namespace)
+            drain()
+            w = cc.get_workload(self.name, self.namespace)
             pods = (w or {}).get("pods") or []
             if not pods:
+                drain()  # final tail
                 return True
             _t.sleep(poll)
         return False

@@ -219,6 +219,8 @@ async def _pod_monitor(interval=None):
             manifest = w.get("manifest") or {}
             if not manifest:
                 continue
+            if (w.get("metadata") or {}).get("module_type") == "app":
+                continue  # apps run to completion; exit is not a fault
             desired = w.get("desired_replicas") or \
                 manifest.get("spec", {}).get("replicas", 1)
             try:
@@ -487,6 +489,19 @@ async def delete_workload(ns: str, name: str):
     except Exception:
         pass
     return {"ok": True}
+
+
+@app.get("/controller/podlogs/{ns}/{name}")
+def pod_logs(ns: str, name: str, offset: int = 0):
+    """Raw pod main-process output (kubectl-logs passthrough; the source
+    for App foreground log-follow)."""
+    if not hasattr(HUB.driver, "pod_logs"):
+        return {"text": "", "offset": offset}
+    try:
+        text, new_off = HUB.driver.pod_logs(name, ns, offset=offset)
+        return {"text": text, "offset": new_off}
+    except Exception as e:
+        return {"text": "", "offset": offset, "error": str(e)}
 
 
 @app.get("/controller/events/{ns}/{name}")

@@ -31,6 +31,7 @@ class LocalPod:
         self.name = name
         self.port = port
         self.proc = proc
+        self.log_path = None
 
     @property
     def host(self):
@@ -160,13 +161,20 @@ class LocalDriver:
                 cmd = [sys.executable, "-m",
                        "kubetorch_amd.serving.http_server",
                        "--port", str(port), "--host", "127.0.0.1"]
-            proc = subprocess.Popen(
-                cmd,
-                env=env,
-                stdout=subprocess.DEVNULL,
-                stderr=subprocess.DEVNULL,
-            )
-            new_pods.append(LocalPod(f"{name}-{i}", port, proc))
+            # pod stdout/stderr goes to a per-pod log file (the stand-in
+            # for `kubectl logs`; /controller/podlogs serves it)
+            import tempfile
+
+            logdir = os.path.join(tempfile.gettempdir(), "kt-pod-logs",
+                                  namespace)
+            os.makedirs(logdir, exist_ok=True)
+            log_path = os.path.join(logdir, f"{name}-{i}.log")
+            logf = open(log_path, "ab")
+            proc = subprocess.Popen(cmd, env=env, stdout=logf, stderr=logf)
+            logf.close()
+            pod = LocalPod(f"{name}-{i}", port, proc)
+            pod.log_path = log_path
+            new_pods.append(pod)
             self._event(namespace, name, "Scheduled",
                         f"assigned 127.0.0.1:{port}", pod=f"{name}-{i}")
             self._event(namespace, name, "Started",
@@ -244,6 +252,22 @@ class LocalDriver:
         return [p.host for p in self.services.get((namespace, name), [])
                 if p.alive()]
 
+    def pod_logs(self, name, namespace, offset=0):
+        """Raw pod process stdout/stderr (the kubectl-logs stand-in; app
+        pods' command output lands here). Returns (text, new_offset)."""
+        chunks = []
+        end = offset
+        for p in self.services.get((namespace, name), []):
+            if not p.log_path or not os.path.exists(p.log_path):
+                continue
+            with open(p.log_path, "rb") as f:
+                f.seek(offset)
+                data = f.read()
+            if data:
+                chunks.append(data.decode(errors="replace"))
+                end = max(end, offset + len(data))
+        return "".join(chunks), end
+
     def teardown_all(self):
         for key in list(self.services):
             self.delete(key[1], key[0])
@@ -314,6 +338,14 @@ class K8sDriver:
             if ip:
                 hosts.append(f"{ip}:{C.SERVER_PORT}")
         return hosts
+
+    def pod_logs(self, name, namespace, offset=0):
+        out = subprocess.run(
+            [self.kubectl, "-n", namespace, "logs", "-l",
+             f"{C.SERVICE_LABEL}={name}", "--tail=1000"],
+            capture_output=True, text=True)
+        text = out.stdout if out.returncode == 0 else ""
+        return text[offset:], len(text)
 
     def apply_secret(self, spec, namespace):
         import base64

@@ -247,12 +247,17 @@ def test_get_if_exists_reuses_service(remote_fn):
 
 
 def test_app_mode_run_and_wait():
-    """App mode: the user command is the pod main process; wait() returns
-    once it exits (reference: kt run + foreground follow)."""
-    a = kt.app("sleep 1 && echo app-done", name="shortapp")
+    """App mode: the user command is the pod main process; wait() streams
+    its output and returns once it exits (reference: kt run + foreground
+    follow / _wait_for_app_exit)."""
+    a = kt.app("echo app-line-one && sleep 1 && echo app-done",
+               name="shortapp")
     a.to(kt.Compute(cpus=1))
+    lines = []
     try:
-        assert a.wait(timeout=60), "app did not finish"
+        assert a.wait(timeout=60, printer=lines.append), "app did not finish"
+        joined = "\n".join(lines)
+        assert "app-line-one" in joined and "app-done" in joined, lines
     finally:
         a.teardown()
 
