@@ -32,7 +32,7 @@ class _Request:
 
 class BatchedGenerator:
     def __init__(self, model, max_batch=8, max_len=None, device=None,
-                 prefill_chunk=None):
+                 prefill_chunk=None, graph=None):
         self.model = model
         cfg = model.cfg
         self.cfg = cfg
@@ -52,6 +52,26 @@ class BatchedGenerator:
         self.pending = []
         self.finished = {}
         self._next_rid = 0
+        # hipGraph-captured decode step (the single-token step is
+        # launch-bound: ~10 kernels x n_layers per step; one graph replay
+        # removes the per-launch host cost — profiles/ROUND2.md lever 5).
+        # Shape-static by construction: full max_batch every step, fixed
+        # max_len horizon, static in/out buffers. KT_DECODE_GRAPH=0 or
+        # graph=False forces the eager path.
+        if graph is None:
+            import os
+
+            graph = (self.device.type == "cuda"
+                     and os.environ.get("KT_DECODE_GRAPH", "1") != "0")
+        self._use_graph = bool(graph)
+        self._graphs = {}      # L bucket -> (CUDAGraph, logits buffer)
+        self._g_pool = None    # shared capture mempool across buckets
+        self._g_toks = None
+        self._g_act = None
+        self._g_rows = torch.arange(max_batch, device=self.device)
+        # host mirror of per-slot lengths: picks the smallest captured
+        # horizon bucket without a device sync
+        self._host_lens = [0] * max_batch
 
     # -- client API ----------------------------------------------------------
     def submit(self, prompt_ids, max_new_tokens=32, temperature=0.0,
@@ -110,6 +130,7 @@ class BatchedGenerator:
             self.k[i][req.slot, :, :S0] = cache.k[i][0, :, :S0]
             self.v[i][req.slot, :, :S0] = cache.v[i][0, :, :S0]
         self.lens[req.slot] = S0
+        self._host_lens[req.slot] = S0
         req.out = req.prompt.tolist()
         self._emit(req, logits[0])
 
@@ -129,8 +150,99 @@ class BatchedGenerator:
             self.finished[req.rid] = req.out
             self.slots[req.slot] = None
             self.lens[req.slot] = 0
+            self._host_lens[req.slot] = 0
 
     def _decode(self, active):
+        if self._use_graph:
+            return self._decode_graph(active)
+        return self._decode_eager(active)
+
+    # -- hipGraph path -------------------------------------------------------
+    def _decode_math(self, L):
+        """One full-batch decode step over horizon L as a pure function of
+        the static buffers (_g_toks, lens, _g_act) — capturable: fixed
+        shapes, no host syncs, no data-dependent control flow. Inactive
+        slots decode garbage (their mask exposes only cache row 0) and are
+        ignored at emit time; their cache rows are overwritten by the next
+        prefill."""
+        cfg, m = self.cfg, self.model
+        B = self.max_batch
+        hd, Hq, Hkv = cfg.head_dim, cfg.n_heads, cfg.n_kv_heads
+        eps = cfg.norm_eps
+        lens = self.lens
+        cos = m.rope_cos[lens].view(B, 1, 1, hd // 2)
+        sin = m.rope_sin[lens].view(B, 1, 1, hd // 2)
+        ar = torch.arange(L, device=self.device)
+        mask = (ar.view(1, L) <= lens.view(B, 1)).view(B, 1, 1, L)
+        h = m.embed(self._g_toks).view(B, 1, cfg.dim)
+        for i, layer in enumerate(m.layers):
+            n1 = ops.rmsnorm(h, layer.attn_norm.weight, eps)
+            qkv = layer.attn.wqkv(n1)
+            q, k, v = qkv.split([Hq * hd, Hkv * hd, Hkv * hd], dim=-1)
+            q = self._rope1(q.view(B, 1, Hq, hd), cos, sin)
+            k = self._rope1(k.view(B, 1, Hkv, hd), cos, sin)
+            v = v.view(B, 1, Hkv, hd)
+            self.k[i][self._g_rows, :, lens] = k[:, 0]
+            self.v[i][self._g_rows, :, lens] = v[:, 0]
+            o = layer.attn._sdpa_masked(q.transpose(1, 2),
+                                        self.k[i][:, :, :L],
+                                        self.v[i][:, :, :L], mask)
+            h = h + layer.attn.wo(o.transpose(1, 2).reshape(B, 1, -1))
+            n2 = ops.rmsnorm(h, layer.mlp_norm.weight, eps)
+            h = h + layer.mlp(n2)
+        logits = m.lm_head(ops.rmsnorm(h, m.norm.weight, eps))[:, -1]
+        self.lens.add_(self._g_act)  # active slots advance one position
+        return logits
+
+    def _bucket_for(self, need):
+        """Smallest power-of-two horizon >= need (>=128), capped at
+        max_len. The per-bucket graphs keep attention reads proportional
+        to the actual active horizon instead of the full cache."""
+        L = 128
+        while L < need:
+            L *= 2
+        return min(L, self.max_len)
+
+    def _ensure_graph(self, L):
+        g = self._graphs.get(L)
+        if g is not None:
+            return g
+        if self._g_toks is None:
+            self._g_toks = torch.zeros(self.max_batch, dtype=torch.long,
+                                       device=self.device)
+            self._g_act = torch.zeros(self.max_batch, dtype=torch.long,
+                                      device=self.device)
+            self._g_pool = torch.cuda.graph_pool_handle()
+        lens_snapshot = self.lens.clone()
+        for _ in range(2):  # warm up allocator/workspaces pre-capture
+            self._decode_math(L)
+        self.lens.copy_(lens_snapshot)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self._g_pool):
+            logits = self._decode_math(L)
+        self._graphs[L] = (graph, logits)
+        return self._graphs[L]
+
+    def _decode_graph(self, active):
+        need = max(self._host_lens[s] for s in active) + 1
+        graph, logits = self._ensure_graph(self._bucket_for(need))
+        toks = [0] * self.max_batch
+        act = [0] * self.max_batch
+        for s in active:
+            toks[s] = self.slots[s].next_tok
+            act[s] = 1
+            self._host_lens[s] += 1
+        self._g_toks.copy_(torch.tensor(toks, dtype=torch.long),
+                           non_blocking=True)
+        self._g_act.copy_(torch.tensor(act, dtype=torch.long),
+                          non_blocking=True)
+        graph.replay()
+        for s in active:
+            self._emit(self.slots[s], logits[s])
+
+    # -- eager path (CPU, or KT_DECODE_GRAPH=0) ------------------------------
+    def _decode_eager(self, active):
         cfg = self.cfg
         m = self.model
         idx = torch.tensor(active, device=self.device)
