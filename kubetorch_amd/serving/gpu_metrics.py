@@ -14,7 +14,16 @@ class AMDGPUCollector:
     per GPU. Cheap no-op on CPU-only nodes."""
 
     def collect(self):
-        stats = _read_gpu_stats()
+        # A collector exception (or a non-numeric sample — amd-smi reports
+        # "N/A" for some fields) surfaces during prometheus TEXT
+        # GENERATION and turns the whole /metrics endpoint into a 500,
+        # which silently starves every consumer of the core metrics (the
+        # autoscaler's concurrency signal, the TTL heartbeat). Coerce to
+        # float and never raise.
+        try:
+            stats = _read_gpu_stats()
+        except Exception:
+            stats = []
         if not stats:
             return
         util = GaugeMetricFamily("kt_gpu_utilization_percent",
@@ -32,8 +41,11 @@ class AMDGPUCollector:
             for fam, key in ((util, "util"), (vram, "vram_used"),
                              (vram_total, "vram_total"), (power, "power"),
                              (temp, "temp")):
-                if s.get(key) is not None:
-                    fam.add_metric(lbl, s[key])
+                try:
+                    val = float(s.get(key))
+                except (TypeError, ValueError):
+                    continue
+                fam.add_metric(lbl, val)
         yield from (util, vram, vram_total, power, temp)
 
 
@@ -110,7 +122,9 @@ def register():
             import torch
 
             if torch.cuda.is_available():
-                REGISTRY.register(AMDGPUCollector())
+                from kubetorch_amd.serving.metrics import register_custom
+
+                register_custom(AMDGPUCollector())
         except Exception:
             pass
         _registered = True

@@ -36,8 +36,29 @@ ACTIVE_REQUESTS = Gauge(
 HEARTBEAT = Counter("kt_heartbeat_sent", "TTL heartbeats", registry=REGISTRY)
 
 
+# custom (optional) collectors — e.g. the amd-smi GPU collector — are
+# tracked so a broken one can be evicted instead of 500-ing /metrics
+CUSTOM_COLLECTORS = []
+
+
+def register_custom(collector):
+    REGISTRY.register(collector)
+    CUSTOM_COLLECTORS.append(collector)
+
+
 def exposition() -> bytes:
-    return generate_latest(REGISTRY)
+    try:
+        return generate_latest(REGISTRY)
+    except Exception:
+        # a failing optional collector must never starve consumers of the
+        # core metrics (autoscaler signal, TTL heartbeat): evict and retry
+        while CUSTOM_COLLECTORS:
+            c = CUSTOM_COLLECTORS.pop()
+            try:
+                REGISTRY.unregister(c)
+            except Exception:
+                pass
+        return generate_latest(REGISTRY)
 
 
 def inactivity_ttl_seconds():

@@ -43,13 +43,19 @@ class CudaTrainer:
 
 
 class CudaInference:
-    def pull(self):
+    def pull(self, delay=0.0):
+        import os
+        import time
+
         import kubetorch_amd as kt
 
         dest = torch.zeros(256, 256, device="cuda", dtype=torch.bfloat16)
         kt.get("rlgpu/w", dest)
         torch.cuda.synchronize()
-        return float(dest.float().sum().item())
+        if delay:
+            time.sleep(delay)  # simulated inference load (autoscale signal)
+        return {"sum": float(dest.float().sum().item()),
+                "pod": os.environ.get("POD_NAME", "?")}
 
 
 def elastic_gpu_step(v, delay=0.0):
