@@ -7,6 +7,7 @@
 // pair it with torch's AITER backward in ops.flash_attention.
 // (Upstream dispatch `fmha_fwd()` is codegen at CK build time and is not
 // shipped in /opt/rocm, so we instantiate the kernel template directly.)
+#include <cstdlib>
 #include <cstring>
 #include <ck_tile/core.hpp>
 #include <ck_tile/host/kernel_launch.hpp>
@@ -144,7 +145,12 @@ extern "C" void kt_attn_fwd_v3(const void* q, const void* k, const void* v,
       (ck_tile::index_t)Hq * S /* batch_stride_lse */,
       (ck_tile::index_t)st.o_b, -1 /* window_left */, 0 /* window_right */,
       (ck_tile::index_t)ck_tile::GenericAttentionMaskEnum::MASK_FROM_TOP_LEFT,
-      2 /* remap_opt (aiter default for this shape) */,
+      [] {  // XCD-aware tile remapping (0=none, 1/2=swizzles); 2 is the
+           // aiter default for this shape — KT_V3_REMAP overrides for
+           // on-box sweeps without a recompile
+        const char* r = std::getenv("KT_V3_REMAP");
+        return (ck_tile::index_t)(r ? atoi(r) : 2);
+      }(),
       nullptr, nullptr);
   dim3 grid = V3Kernel::GridSize(B, Hq, S, D);
   constexpr dim3 blocks = V3Kernel::BlockSize();

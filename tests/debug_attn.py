@@ -431,3 +431,41 @@ def bwd_perf():
 
 if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "bwdperf":
     bwd_perf()
+
+
+def v3_sweep():
+    """Sweep KT_V3_REMAP (XCD tile remap) on the Llama shape.
+    Run: debug_attn.py v3sweep"""
+    import os
+
+    B, Hq, Hkv, S = 4, 32, 8, 4096
+    q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    scale = 128 ** -0.5
+    flops = 4 * B * Hq * S * S * 128 * 0.5
+
+    def timeit(fn, iters=30):
+        for _ in range(5):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    ref = None
+    for remap in ("0", "1", "2"):
+        os.environ["KT_V3_REMAP"] = remap
+        t = timeit(lambda: ops._ext().attn_fwd_v3(q, k, v, scale))
+        o, lse = ops._ext().attn_fwd_v3(q, k, v, scale)
+        if ref is None:
+            ref = o.float()
+        err = (o.float() - ref).abs().max().item()
+        print(f"remap={remap}: {flops/t/1e12:.0f} TF ({t*1e3:.3f} ms) err={err:.1e}")
+    os.environ.pop("KT_V3_REMAP", None)
+
+
+if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "v3sweep":
+    v3_sweep()
