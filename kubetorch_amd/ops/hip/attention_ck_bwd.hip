@@ -45,10 +45,11 @@ using BwdDropout = ck_tile::BlockDropoutBwd<false, false, false>;
 // Only gemm4 (dS@K over kN0 keys) has an explicit k4 loop.
 // Block-warp M dims follow gemm1/3 M=N0 (dV/dK partition keys across
 // warps), gemm0/2/4 M=M0.
-template <bool UseTrLoad, int WtM, int WtN, int WtK, int M0 = 32>
+template <bool UseTrLoad, int WtM, int WtN, int WtK, int M0 = 32,
+          int N0 = 128>
 struct BwdConfig {
   using Shape = ck_tile::TileFmhaBwdShape<
-      ck_tile::sequence<M0, 128, 128, M0, 128, M0, 32, 128, 128>,
+      ck_tile::sequence<M0, N0, 128, M0, 128, M0, 32, 128, 128>,
       ck_tile::sequence<1, 4, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm0 S
       ck_tile::sequence<4, 1, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm1 dV
       ck_tile::sequence<1, 4, 1>, ck_tile::sequence<WtM, WtN, WtK>,  // gemm2 dP
@@ -77,6 +78,9 @@ struct BwdConfig {
 using StdKernel = BwdConfig<false, 32, 32, 16>::Kernel;
 using TrKernel = BwdConfig<true, 16, 16, 32>::Kernel;
 using TrKernel64 = BwdConfig<true, 16, 16, 32, 64>::Kernel;  // pipe=2
+// pipe=3: N0=64 halves the register-resident K/KT/V tiles (occupancy 2
+// candidate; 2x the key-block grid)
+using TrKernelN64 = BwdConfig<true, 16, 16, 32, 32, 64>::Kernel;
 
 // --- D = rowsum(dO*O) -------------------------------------------------------
 using DotTraits = ck_tile::TileFmhaBwdOGradDotOTraits<
@@ -180,7 +184,10 @@ extern "C" void kt_attn_bwd_ck(const void* q, const void* k, const void* v,
                                void* dv, int B, int Hq, int Hkv, int S,
                                float scale, int mask_mode, int pipe,
                                hipStream_t stream) {
-  if (pipe == 2) {
+  if (pipe == 3) {
+    run_bwd<TrKernelN64>(q, k, v, o, do_, lse, d, dq_acc, dq, dk, dv, B, Hq,
+                         Hkv, S, scale, mask_mode, stream);
+  } else if (pipe == 2) {
     run_bwd<TrKernel64>(q, k, v, o, do_, lse, d, dq_acc, dq, dk, dv, B, Hq,
                         Hkv, S, scale, mask_mode, stream);
   } else if (pipe == 1) {

@@ -416,7 +416,7 @@ def bwd_perf():
 
     flops = 4 * B * Hq * S * S * D * 0.5 * 2.5  # bwd ~2.5x fwd matmul work
     ra = aten_path()
-    for pipe in ("0", "1", "2"):
+    for pipe in ("0", "1", "2", "3"):
         os.environ["KT_CKBWD_PIPE"] = pipe
         t = timeit(ck_path)
         rc = ck_path()
