@@ -459,10 +459,31 @@ def secrets_delete(name: str,
 
 
 @volumes_app.command("create")
-def volumes_create(name: str, size: str = typer.Option("10Gi")):
+def volumes_create(name: str, size: str = typer.Option("10Gi"),
+                   existing_pv: str = typer.Option(None, "--pv"),
+                   namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
     from kubetorch_amd.resources.volume import Volume
 
-    console.print_json(json.dumps(Volume(name, size=size).to_pvc_manifest("default")))
+    v = Volume(name, size=size, existing_pv=existing_pv)
+    controller_client().put_volume(v, namespace)
+    console.print(f"[green]pvc {v.claim_name} created[/green]")
+
+
+@volumes_app.command("list")
+def volumes_list(namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
+
+    for v in controller_client().list_volumes(namespace):
+        console.print(f"{v['name']}  {v.get('spec', {})}")
+
+
+@volumes_app.command("delete")
+def volumes_delete(name: str, namespace: str = typer.Option("default", "-n")):
+    from kubetorch_amd.globals import controller_client
+
+    controller_client().delete_volume(name, namespace)
+    console.print(f"[green]pvc {name} deleted[/green]")
 
 
 @server_app.command("start")

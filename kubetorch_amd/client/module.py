@@ -163,10 +163,13 @@ class Module:
         self._exec_token = uuid.uuid4().hex
         md = self.metadata()
         self._sync_workdir(md)
-        # secrets must exist before pods reference them (reference:
-        # _upload_secrets_list before the manifest apply)
+        # secrets and PVCs must exist before pods reference them
+        # (reference: _upload_secrets_list / Volume create before apply)
         for s in (self.compute.secrets or []):
             controller_client().put_secret(s, self.namespace)
+        for v in (self.compute.volumes or []):
+            if getattr(v, "needs_create", False):
+                controller_client().put_volume(v, self.namespace)
         launch_id = uuid.uuid4().hex[:12]
         manifest = self.compute.to_manifest(self.name, username=config.username,
                                             module=self.pointers["name"])

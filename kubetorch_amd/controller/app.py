@@ -336,6 +336,36 @@ async def deploy(request: Request):
             "reloaded_pods": acked, "driver": HUB.driver_name}
 
 
+@app.post("/controller/volumes/{ns}")
+async def put_volume(ns: str, request: Request):
+    """Create a PVC through the control plane (reference: Volume create
+    via controller). Body: the PVC manifest."""
+    manifest = await request.json()
+    try:
+        await asyncio.to_thread(HUB.driver.apply_volume, manifest, ns)
+    except Exception as e:
+        return JSONResponse({"error": f"pvc apply failed: {e}"},
+                            status_code=500)
+    return {"ok": True, "name": manifest.get("metadata", {}).get("name")}
+
+
+@app.get("/controller/volumes/{ns}")
+def list_volumes(ns: str):
+    try:
+        return {"volumes": HUB.driver.list_volumes(ns)}
+    except Exception as e:
+        return JSONResponse({"error": str(e)}, status_code=500)
+
+
+@app.delete("/controller/volumes/{ns}/{name}")
+async def delete_volume(ns: str, name: str):
+    try:
+        await asyncio.to_thread(HUB.driver.delete_volume, name, ns)
+    except Exception as e:
+        return JSONResponse({"error": str(e)}, status_code=500)
+    return {"ok": True}
+
+
 @app.post("/controller/secrets/{ns}")
 async def put_secret(ns: str, request: Request):
     """Create/update a secret through the control plane (reference parity:

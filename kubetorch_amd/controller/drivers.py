@@ -139,6 +139,7 @@ class LocalDriver:
             for k, v in (md.get("env") or {}).items():
                 env[k] = str(v)
             self._inject_secrets(manifest, namespace, env)
+            self._inject_volumes(manifest, namespace, env)
             # module metadata -> env contract (same as controller push)
             if md.get("file_path"):
                 env[C.ENV_FILE_PATH] = md["file_path"]
@@ -204,6 +205,52 @@ class LocalDriver:
         k8s_name = f"kt-secret-{name}".lower().replace("_", "-")
         self.secrets.pop((namespace, k8s_name), None)
         self.secrets.pop((namespace, name), None)
+
+    # -- volumes (local stand-in: host dirs) --------------------------------
+    def _volume_root(self, namespace, claim):
+        import tempfile
+
+        return os.path.join(tempfile.gettempdir(), "kt-local-volumes",
+                            namespace, claim)
+
+    def apply_volume(self, manifest, namespace):
+        claim = manifest.get("metadata", {}).get("name")
+        os.makedirs(self._volume_root(namespace, claim), exist_ok=True)
+        self.volumes = getattr(self, "volumes", {})
+        self.volumes[(namespace, claim)] = manifest
+
+    def list_volumes(self, namespace):
+        vols = getattr(self, "volumes", {})
+        return [{"name": c, "spec": m.get("spec", {})}
+                for (ns, c), m in vols.items() if ns == namespace]
+
+    def delete_volume(self, name, namespace):
+        import shutil as _sh
+
+        getattr(self, "volumes", {}).pop((namespace, name), None)
+        _sh.rmtree(self._volume_root(namespace, name), ignore_errors=True)
+
+    def _inject_volumes(self, manifest, namespace, env):
+        """Expose PVC mounts to local pods as host dirs via
+        KT_VOLUME_MOUNT_<NAME> (a pod subprocess cannot be given a real
+        mount namespace; same convention as secrets)."""
+        try:
+            spec = (manifest.get("spec", {}).get("template", {})
+                    .get("spec", {})) or {}
+            claims = {v["name"]: v["persistentVolumeClaim"]["claimName"]
+                      for v in spec.get("volumes", []) or []
+                      if "persistentVolumeClaim" in v}
+            c = (spec.get("containers") or [{}])[0]
+            for vm in c.get("volumeMounts", []) or []:
+                claim = claims.get(vm["name"])
+                if not claim:
+                    continue
+                d = self._volume_root(namespace, claim)
+                os.makedirs(d, exist_ok=True)
+                env["KT_VOLUME_MOUNT_" + vm["name"].upper()
+                    .replace("-", "_")] = d
+        except Exception:
+            pass
 
     def _inject_secrets(self, manifest, namespace, env):
         """Materialize the pod template's secret references into the local
@@ -338,6 +385,27 @@ class K8sDriver:
             if ip:
                 hosts.append(f"{ip}:{C.SERVER_PORT}")
         return hosts
+
+    def apply_volume(self, manifest, namespace):
+        r = subprocess.run(
+            [self.kubectl, "-n", namespace, "apply", "-f", "-"],
+            input=json.dumps(manifest).encode(), capture_output=True)
+        if r.returncode != 0:
+            raise RuntimeError(r.stderr.decode(errors="replace")[-500:])
+
+    def list_volumes(self, namespace):
+        out = subprocess.run(
+            [self.kubectl, "-n", namespace, "get", "pvc", "-o", "json"],
+            check=True, capture_output=True)
+        return [{"name": it["metadata"]["name"],
+                 "spec": it.get("spec", {})}
+                for it in json.loads(out.stdout).get("items", [])]
+
+    def delete_volume(self, name, namespace):
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "delete", "pvc", name,
+             "--ignore-not-found"],
+            check=True, capture_output=True)
 
     def pod_logs(self, name, namespace, offset=0):
         out = subprocess.run(

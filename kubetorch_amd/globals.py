@@ -68,6 +68,20 @@ class ControllerClient:
             raise LaunchError(str(data))
         return data
 
+    def put_volume(self, volume, namespace):
+        r = self._request("POST", f"/controller/volumes/{namespace}",
+                          json=volume.to_pvc_manifest(namespace))
+        r.raise_for_status()
+        return r.json()
+
+    def list_volumes(self, namespace):
+        return self._request(
+            "GET", f"/controller/volumes/{namespace}").json().get("volumes", [])
+
+    def delete_volume(self, name, namespace):
+        return self._request(
+            "DELETE", f"/controller/volumes/{namespace}/{name}").json()
+
     def put_secret(self, secret, namespace):
         """Create/update a Secret through the controller (reference:
         controller-side kubernetes_secrets_client)."""

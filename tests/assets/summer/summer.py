@@ -82,3 +82,20 @@ def read_secret_file(secret_name, key):
         return None
     with open(os.path.join(d, key)) as f:
         return f.read()
+
+
+def write_volume(vol_env, fname, content):
+    d = os.environ.get(vol_env)
+    if not d:
+        return None
+    with open(os.path.join(d, fname), "w") as f:
+        f.write(content)
+    return d
+
+
+def read_volume(vol_env, fname):
+    d = os.environ.get(vol_env)
+    if not d:
+        return None
+    with open(os.path.join(d, fname)) as f:
+        return f.read()

@@ -95,3 +95,39 @@ def test_ssh_interactive_refused_in_local_mode():
     f.compute = kt.Compute(cpus=1)
     with pytest.raises(RuntimeError, match="local-driver"):
         f.ssh()
+
+
+@pytest.mark.timeout(240)
+def test_volume_persists_across_services():
+    """PVC behavior on the local driver: the controller creates the
+    volume at deploy, two different services mount the same claim
+    (KT_VOLUME_MOUNT_<NAME> stand-in) and share state; existing-PV binds
+    render volumeName in the PVC manifest."""
+    vol = kt.Volume("shared-cache", size="1Gi")
+    writer = kt.fn(summer_mod.write_volume).to(
+        kt.Compute(cpus=1, volumes=[vol]))
+    reader = kt.fn(summer_mod.read_volume).to(
+        kt.Compute(cpus=1, volumes=[vol]))
+    try:
+        d = writer("KT_VOLUME_MOUNT_SHARED_CACHE", "state.txt", "v42",
+                   kt_timeout=60)
+        assert d, "volume mount env missing in pod"
+        assert reader("KT_VOLUME_MOUNT_SHARED_CACHE", "state.txt",
+                      kt_timeout=60) == "v42"
+        from kubetorch_amd.globals import controller_client
+
+        names = {v["name"] for v in controller_client().list_volumes("default")}
+        assert "shared-cache" in names
+        controller_client().delete_volume("shared-cache", "default")
+    finally:
+        writer.teardown()
+        reader.teardown()
+
+
+def test_volume_existing_pv_bind():
+    m = kt.Volume("bindit", existing_pv="pv-123").to_pvc_manifest("default")
+    assert m["spec"]["volumeName"] == "pv-123"
+    assert m["spec"]["storageClassName"] == ""
+    # existing claim -> no creation needed
+    assert not kt.Volume("x", existing_claim="already").needs_create
+    assert kt.Volume("y").needs_create
