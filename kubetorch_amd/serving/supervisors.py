@@ -374,11 +374,14 @@ class RaySupervisor(ExecutionSupervisor):
             self._wait_gcs()
         super().__init__(num_proc=1)
 
-    def _wait_gcs(self, port=6379, timeout=120):
+    def _wait_gcs(self, port=None, timeout=120):
         """Poll the Ray GCS port until live (reference: ray_supervisor.py
         GCS liveness check)."""
         import socket
         import time
+
+        if port is None:
+            port = int(os.environ.get("KT_RAY_GCS_PORT", "6379"))
 
         deadline = time.time() + timeout
         while time.time() < deadline:

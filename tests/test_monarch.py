@@ -64,3 +64,55 @@ def test_allocator_service_and_controller_env(stub_allocator):
     finally:
         sup.cleanup()
     assert sup._alloc_proc.poll() is not None  # torn down
+
+
+GCS_STUB = """#!{python}
+import socket, time
+s = socket.socket()
+s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+s.bind(("127.0.0.1", {port}))
+s.listen(1)
+while True:
+    time.sleep(1)
+"""
+
+
+@pytest.mark.timeout(180)
+def test_ray_supervisor_launch_contract(tmp_path, monkeypatch):
+    """RaySupervisor launch contract with a stub head command (no ray
+    wheel on this image): starts KUBERAY_GEN_RAY_START_CMD, waits for GCS
+    liveness, serves calls from one worker, tears the head down."""
+    import socket as _socket
+
+    from kubetorch_amd.serving.supervisors import RaySupervisor
+
+    s = _socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    stub = tmp_path / "ray_head.py"
+    stub.write_text(GCS_STUB.format(python=sys.executable, port=port))
+    monkeypatch.setenv("KT_RAY_GCS_PORT", str(port))
+    monkeypatch.setenv("KUBERAY_GEN_RAY_START_CMD",
+                       f"{sys.executable} {stub}")
+    monkeypatch.setenv("KT_FILE_PATH", os.path.join(ASSETS, "summer.py"))
+    monkeypatch.setenv("KT_PROJECT_ROOT", ASSETS)
+    monkeypatch.setenv("KT_CLS_OR_FN_NAME", "summer")
+    monkeypatch.setenv("KT_MODULE_TYPE", "fn")
+    sup = RaySupervisor()
+    try:
+        assert sup._ray_proc.poll() is None  # head process live
+        assert sup.call(args=(3, 4)) == 7
+    finally:
+        sup.cleanup()
+    assert sup._ray_proc.poll() is not None
+
+
+@pytest.mark.timeout(120)
+def test_ray_supervisor_head_crash_raises(tmp_path, monkeypatch):
+    from kubetorch_amd.serving.supervisors import RaySupervisor
+
+    monkeypatch.setenv("KT_RAY_GCS_PORT", "1")  # nothing will listen
+    monkeypatch.setenv("KUBERAY_GEN_RAY_START_CMD", "exit 3")
+    with pytest.raises(RuntimeError, match="exited during startup"):
+        RaySupervisor()
