@@ -50,7 +50,13 @@ def test_secrets_volumes_manifests():
                             '{"HF_TOKEN": "x"}'])
     assert r.exit_code == 0 and "kt-secret-hf" in r.output
     r = runner.invoke(app, ["volumes", "create", "cache", "--size", "5Gi"])
-    assert r.exit_code == 0 and "5Gi" in r.output
+    assert r.exit_code == 0 and "pvc cache created" in r.output
+    r = runner.invoke(app, ["volumes", "list"])
+    assert r.exit_code == 0 and "cache" in r.output
+    r = runner.invoke(app, ["volumes", "delete", "cache"])
+    assert r.exit_code == 0
+    r = runner.invoke(app, ["secrets", "delete", "hf"])
+    assert r.exit_code == 0
 
 
 def test_run_bash_and_pip_helpers():
