@@ -101,16 +101,48 @@ class BatchedGenerator:
     @torch.no_grad()
     def step(self):
         """One engine iteration: admit pending prompts into free slots
-        (prefill), then one batched decode step for every active slot."""
+        (prefill — same-length prompts share one batched forward), then
+        one batched decode step for every active slot."""
+        admitted = []
         for slot in range(self.max_batch):
             if self.slots[slot] is None and self.pending:
                 req = self.pending.pop(0)
                 req.slot = slot
                 self.slots[slot] = req
-                self._prefill(req)
+                admitted.append(req)
+        if admitted:
+            by_len = {}
+            for req in admitted:
+                by_len.setdefault(req.prompt.numel(), []).append(req)
+            for group in by_len.values():
+                if len(group) > 1 and not self.prefill_chunk:
+                    self._prefill_batch(group)
+                else:
+                    for req in group:
+                        self._prefill(req)
         active = [s for s in range(self.max_batch) if self.slots[s] is not None]
         if active:
             self._decode(active)
+
+    def _prefill_batch(self, group):
+        """One forward for G same-length prompts (a 1-at-a-time prefill is
+        GEMM-starved: [1, S0] activations; [G, S0] runs the same layer
+        GEMMs at G-fold arithmetic intensity)."""
+        from kubetorch_amd.models.llama import KVCache
+
+        S0 = group[0].prompt.numel()
+        G = len(group)
+        batch = torch.stack([req.prompt for req in group])
+        cache = KVCache(self.cfg, G, S0, self.device, self.dtype)
+        logits = self.model._forward_cached(batch, cache)
+        for j, req in enumerate(group):
+            for i in range(self.cfg.n_layers):
+                self.k[i][req.slot, :, :S0] = cache.k[i][j, :, :S0]
+                self.v[i][req.slot, :, :S0] = cache.v[i][j, :, :S0]
+            self.lens[req.slot] = S0
+            self._host_lens[req.slot] = S0
+            req.out = req.prompt.tolist()
+            self._emit(req, logits[j])
 
     def _prefill(self, req):
         """Run the prompt through a throwaway per-request cache, then copy

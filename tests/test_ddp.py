@@ -327,6 +327,7 @@ def test_clip_and_accum_on_gpu_match_reference():
     m2 = build("cpu", torch.float32)
     opt = torch.optim.AdamW(m2.parameters(), lr=1e-2, betas=(0.9, 0.95),
                             eps=1e-8, weight_decay=0.0)
+    norms = []
     for step in range(4):
         for micro in range(2):  # backward twice; buckets accumulate
             torch.nn.functional.mse_loss(m1(x), y).backward()
@@ -334,6 +335,7 @@ def test_clip_and_accum_on_gpu_match_reference():
         grads = {p: eng._param_view[p].float().cpu() / 2.0
                  for p in m1.parameters()}
         eng.step()
+        norms.append(eng.last_grad_norm)
         opt.zero_grad()
         for p1, p2 in zip(m1.parameters(), m2.parameters()):
             p2.grad = grads[p1].clone()
@@ -344,4 +346,5 @@ def test_clip_and_accum_on_gpu_match_reference():
             torch.testing.assert_close(p1.float().cpu(), p2,
                                        rtol=2e-2, atol=2e-2)
             p2.data.copy_(p1.detach().float().cpu())
-    assert eng.last_grad_norm is not None and eng.last_grad_norm > 0.5
+    # clipping actually engaged at least once during the run
+    assert norms and max(norms) > 0.5, norms
