@@ -284,6 +284,7 @@ class LocalDriver:
                 d = os.path.join(tempfile.gettempdir(), "kt-local-secrets",
                                  namespace, sname)
                 os.makedirs(d, exist_ok=True)
+                os.chmod(d, 0o700)
                 for k, v in sec.get("values", {}).items():
                     path = os.path.join(d, k)
                     with open(path, "w") as f:
