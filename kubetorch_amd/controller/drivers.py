@@ -61,6 +61,11 @@ class LocalDriver:
         self._master_ports = {}  # (ns, name) -> rendezvous port (stable
         # across pod replacement so re-joined ranks find the same master)
         self.secrets = {}  # (ns, k8s_name) -> {name, values, as_env, mount_path}
+        # deploys, the pod monitor and the autoscaler all reconcile
+        # concurrently; serialize applies so a service can't double-spawn
+        import threading
+
+        self._apply_lock = threading.Lock()
 
     def _event(self, namespace, name, reason, message, pod=None):
         self.events.setdefault((namespace, name), []).append({
@@ -94,6 +99,11 @@ class LocalDriver:
         (the K8s-Deployment-controller behavior the reference delegates to
         Kubernetes; here the control plane does its own reconciliation,
         which is what makes mid-step pod death auto-heal)."""
+        with self._apply_lock:
+            return self._apply_locked(manifest, namespace, metadata,
+                                      launch_id)
+
+    def _apply_locked(self, manifest, namespace, metadata, launch_id):
         name = manifest["metadata"]["name"]
         replicas = manifest.get("spec", {}).get("replicas", 1)
         key = (namespace, name)
