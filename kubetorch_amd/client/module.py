@@ -325,6 +325,9 @@ class Module:
                 hosts = []
             self._lb_hosts = hosts or self.service_hosts
             self._lb_ts = now
+        if not self._lb_hosts:
+            return self.http  # no live pods known: fall back to the
+            # deploy-time URL (its retry loop rides out a scale-from-zero)
         self._lb_i = getattr(self, "_lb_i", -1) + 1
         host = self._lb_hosts[self._lb_i % len(self._lb_hosts)]
         return HTTPClient(f"http://{host}", self.pointers["name"])
