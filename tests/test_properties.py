@@ -84,3 +84,74 @@ def test_ttl_parse_total(raw):
     expect = {"30s": 30, "5m": 300, "2h": 7200, "1d": 86400, "90": 90,
               "bogus": None, "": None}[raw]
     assert out == expect
+
+
+@given(st.lists(st.sampled_from(["a", "b/../..", "../x", "ok/sub", "/abs",
+                                 "c/./d", "..", "deep/a/b/c"]),
+                min_size=1, max_size=6))
+@settings(max_examples=60, deadline=None)
+def test_tar_safety_never_escapes(names):
+    """Random archives of benign+hostile member names: extraction either
+    raises or writes only inside the destination."""
+    import io
+    import os
+    import tarfile
+    import tempfile
+
+    from kubetorch_amd.utils.tar import safe_extractall
+
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w") as tar:
+        for i, n in enumerate(names):
+            ti = tarfile.TarInfo(f"{n}/f{i}" if not n.startswith("/") else n)
+            data = b"x"
+            ti.size = 1
+            tar.addfile(ti, io.BytesIO(data))
+    buf.seek(0)
+    with tempfile.TemporaryDirectory() as outer:
+        dest = os.path.join(outer, "inner")
+        os.makedirs(dest)
+        try:
+            with tarfile.open(fileobj=buf) as tar:
+                safe_extractall(tar, dest)
+        except ValueError:
+            pass
+        # nothing escaped into the outer dir
+        assert set(os.listdir(outer)) == {"inner"}
+        for root, _dirs, files in os.walk(outer):
+            assert os.path.realpath(root).startswith(os.path.realpath(outer))
+
+
+@given(st.text(min_size=1, max_size=80))
+@settings(max_examples=80, deadline=None)
+def test_sanitize_name_k8s_valid(raw):
+    import re
+
+    from kubetorch_amd.client.module import sanitize_name
+
+    out = sanitize_name(raw)
+    assert len(out) <= 63
+    assert re.fullmatch(r"[a-z0-9-]*", out)
+    assert sanitize_name(out) == out  # idempotent
+
+
+@given(st.integers(min_value=1, max_value=8),
+       st.integers(min_value=1, max_value=8))
+@settings(max_examples=40, deadline=None)
+def test_autoscaler_duration_and_spec(mins, maxs):
+    from kubetorch_amd.controller.app import _autoscale_spec, _parse_duration
+    from kubetorch_amd.provisioning.manifests import build_knative_manifest
+    from kubetorch_amd.resources.autoscaling import AutoscalingConfig
+
+    lo, hi = sorted((mins, maxs))
+    m = build_knative_manifest(
+        "svc", "default", "img",
+        autoscaling=AutoscalingConfig(target=2, min_scale=lo, max_scale=hi,
+                                      scale_down_delay="5s"))
+    spec = _autoscale_spec(m)
+    assert spec["target"] == 2
+    assert spec["min"] == max(1, lo)
+    assert spec["max"] in (hi, 0)
+    assert _parse_duration("5s") == 5.0
+    assert _parse_duration("2m") == 120.0
+    assert _parse_duration("1h") == 3600.0
