@@ -160,7 +160,7 @@ def test_moe_trains_on_gpu():
     finally:
         torch.set_default_dtype(prev)
     convert_to_moe(model, n_experts=4, top_k=2)
-    model.cuda()
+    model = model.to("cuda", torch.bfloat16)  # new experts join in bf16
     opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
     x = torch.randint(0, cfg.vocab_size, (2, 128), device="cuda")
     y = torch.randint(0, cfg.vocab_size, (2, 128), device="cuda")
