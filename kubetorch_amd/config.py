@@ -29,6 +29,7 @@ class KTConfig:
     def __init__(self, path=CONFIG_PATH):
         self._path = path
         self._file = {}
+        self._cluster = None
         if os.path.exists(path):
             try:
                 with open(path) as f:
@@ -45,7 +46,38 @@ class KTConfig:
             return val
         if key in self._file:
             return self._file[key]
+        cluster = self.cluster_config()
+        if key in cluster:
+            return cluster[key]
         return _DEFAULTS.get(key, default)
+
+    def cluster_config(self):
+        """Cluster-wide defaults published by the controller (the chart's
+        kubetorch-config ConfigMap; reference: service_manager.py:803
+        fetching templates/configmaps/kubetorch-config.yaml). Layered
+        BELOW env and the user's file, fetched once per process; never
+        fetched in local mode or before an api_url is known (no recursive
+        discovery)."""
+        if self._cluster is not None:
+            return self._cluster
+        self._cluster = {}
+        api = (os.environ.get("KT_API_URL") or self._file.get("api_url"))
+        local = os.environ.get("KT_LOCAL_MODE", "").lower() == "true" or \
+            self._file.get("local_mode")
+        if api and not local:
+            try:
+                import httpx
+
+                r = httpx.get(api.rstrip("/") + "/controller/config",
+                              timeout=5)
+                if r.status_code == 200:
+                    got = r.json().get("config", {})
+                    if isinstance(got, dict):
+                        self._cluster = {k: v for k, v in got.items()
+                                         if k in _DEFAULTS}
+            except Exception:
+                pass
+        return self._cluster
 
     def set(self, key, value, persist=False):
         self._file[key] = value

@@ -264,6 +264,32 @@ def health():
     return {"status": "ok", "driver": HUB.driver_name}
 
 
+@app.get("/controller/config")
+def cluster_config():
+    """Cluster-wide client defaults (the chart mounts the kubetorch-config
+    ConfigMap at /etc/kubetorch/config.yaml; KT_CLUSTER_CONFIG JSON
+    overrides for tests/BYO)."""
+    import json as _json
+
+    raw = os.environ.get("KT_CLUSTER_CONFIG")
+    if raw:
+        try:
+            return {"config": _json.loads(raw)}
+        except ValueError:
+            pass
+    path = os.environ.get("KT_CLUSTER_CONFIG_PATH",
+                          "/etc/kubetorch/config.yaml")
+    if os.path.exists(path):
+        try:
+            import yaml
+
+            with open(path) as f:
+                return {"config": yaml.safe_load(f) or {}}
+        except Exception:
+            pass
+    return {"config": {}}
+
+
 @app.post("/controller/deploy")
 async def deploy(request: Request):
     """Apply the manifest and hot-reload the workload's pods.

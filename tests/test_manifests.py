@@ -193,3 +193,32 @@ def test_autoscale_bounds_in_manifest():
     ann = m["spec"]["template"]["metadata"]["annotations"]
     assert ann["autoscaling.knative.dev/min-scale"] == "1"
     assert ann["autoscaling.knative.dev/max-scale"] == "6"
+
+
+def test_cluster_config_layer(tmp_path, monkeypatch):
+    """Cluster-wide defaults from the controller sit BELOW env and the
+    user's file (reference: kubetorch-config ConfigMap fetch)."""
+    import json
+
+    from kubetorch_amd.config import KTConfig
+
+    # controller side: env-injected JSON beats the mounted file
+    monkeypatch.setenv("KT_CLUSTER_CONFIG", json.dumps({"image": "rocm/x:1"}))
+    from kubetorch_amd.controller.app import cluster_config
+
+    assert cluster_config()["config"] == {"image": "rocm/x:1"}
+    monkeypatch.delenv("KT_CLUSTER_CONFIG")
+    cfgfile = tmp_path / "cc.yaml"
+    cfgfile.write_text("image: rocm/y:2\nstream_logs: false\n")
+    monkeypatch.setenv("KT_CLUSTER_CONFIG_PATH", str(cfgfile))
+    assert cluster_config()["config"]["image"] == "rocm/y:2"
+
+    # client side layering: cluster < file < env
+    c = KTConfig(path=str(tmp_path / "none.yaml"))
+    c._cluster = {"image": "cluster-img", "namespace": "team-ns"}
+    assert c.get("image") == "cluster-img"
+    assert c.get("namespace") == "team-ns"
+    c._file["image"] = "file-img"
+    assert c.get("image") == "file-img"
+    monkeypatch.setenv("KT_IMAGE", "env-img")
+    assert c.get("image") == "env-img"
