@@ -322,9 +322,10 @@ std::vector<at::Tensor> attn_bwd_ck(const at::Tensor& grad_out,
   auto d = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   // dq is accumulated atomically across key blocks -> fp32, zero-init
   auto dq_acc = at::zeros({B, Hq, S, 128}, q.options().dtype(at::kFloat));
-  // KT_CKBWD_MASK selects the causal-mask convention at runtime (WIP
-  // flipped-mask A/B, profiles/ROUND2.md): 0=top-left (fwd convention),
-  // 1=bottom-right, 2=swapped lr window. KT_CKBWD_PIPE selects the
+  // KT_CKBWD_MASK selects the causal-mask convention at runtime (the
+  // round-2 root cause turned out to be contraction depth, not the mask;
+  // the knob stays for window-attention experiments): 0=top-left (fwd
+  // convention), 1=bottom-right, 2=swapped lr window. KT_CKBWD_PIPE selects the
   // pipeline: 0=std (KRKTRVR IGLP, 32x32x16 warp tiles), 1=trload
   // (gfx950 transposed-fragment loads, 16x16x32).
   int mask_mode = 0;

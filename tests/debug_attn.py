@@ -469,3 +469,51 @@ def v3_sweep():
 
 if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "v3sweep":
     v3_sweep()
+
+
+def flash_fwd_probe():
+    """Does aten's FLASH forward (AITER asm candidate) beat our CK v3 on
+    the Llama shape? Run: debug_attn.py flashfwd"""
+    import torch.nn.functional as F
+    from torch.nn.attention import SDPBackend, sdpa_kernel
+
+    B, Hq, Hkv, S = 4, 32, 8, 4096
+    q = torch.randn(B, Hq, S, 128, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, 128, dtype=torch.bfloat16, device=dev)
+    scale = 128 ** -0.5
+    flops = 4 * B * Hq * S * S * 128 * 0.5
+
+    def timeit(fn, iters=20):
+        for _ in range(3):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters
+
+    for name, backend in (("flash", SDPBackend.FLASH_ATTENTION),
+                          ("efficient", SDPBackend.EFFICIENT_ATTENTION)):
+        try:
+            with sdpa_kernel(backend):
+                t = timeit(lambda: F.scaled_dot_product_attention(
+                    q, k, v, is_causal=True, scale=scale, enable_gqa=True))
+            print(f"sdpa {name} (gqa-native): {flops/t/1e12:.0f} TF")
+        except RuntimeError as e:
+            print(f"sdpa {name}: unavailable ({str(e)[:80]})")
+    t = timeit(lambda: ops._ext().attn_fwd_v3(q, k, v, scale))
+    print(f"our v3 (w/ lse):  {flops/t/1e12:.0f} TF")
+    # training needs LSE: time the aten flash fwd WITH logsumexp
+    try:
+        t = timeit(lambda: torch.ops.aten._flash_attention_forward(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+            None, None, S, S, 0.0, True, True, scale=scale))
+        print(f"aten _flash_attention_forward (lse): {flops/t/1e12:.0f} TF")
+    except Exception as e:
+        print(f"_flash_attention_forward: {str(e)[:100]}")
+
+
+if __name__ == "__main__" and len(sys.argv) > 1 and sys.argv[1] == "flashfwd":
+    flash_fwd_probe()
