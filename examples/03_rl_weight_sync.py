@@ -4,6 +4,9 @@ tensor store (hipIpc registration, zero copy); inference pods pull them
 BASELINE config 5."""
 import kubetorch_amd as kt
 
+# Requires a GPU (hipIpc tensor store). On a CPU-only machine see
+# tests/test_rl_two_services.py for the CPU analog of this flow.
+
 
 class Trainer:
     def __init__(self):
@@ -52,6 +55,12 @@ class Inference:
 
 
 if __name__ == "__main__":
+    import torch
+
+    if not torch.cuda.is_available():
+        raise SystemExit("example 03 needs a GPU (hipIpc tensor store) — "
+                         "see tests/test_rl_two_services.py for the CPU "
+                         "analog")
     trainer = kt.cls(Trainer).to(kt.Compute(gpus=4, memory="64Gi"))
     infer = kt.cls(Inference).to(
         kt.Compute(gpus=1).autoscale(min_scale=1, max_scale=8, target=4))
