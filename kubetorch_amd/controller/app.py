@@ -10,8 +10,18 @@ from-scratch implementation. Endpoints:
     GET|DELETE /controller/workload/{ns}/{name}
     POST /controller/pods/stream      pod registration -> NDJSON push stream
     POST /controller/pods/reload_ack  ack barrier for hot reload
+    GET  /controller/config           cluster-wide client defaults
+    GET  /controller/events/{ns}/{name}   pod lifecycle events (launch UX)
+    GET  /controller/podlogs/{ns}/{name}  raw pod stdout (app log-follow)
+    POST|GET|DELETE /controller/secrets/{ns}[/{name}]
+    POST|GET|DELETE /controller/volumes/{ns}[/{name}]
+    POST /controller/tunnel/open (+/up,/down)  external-client TCP bridge
     GET  /controller/debug/connections
     GET  /health
+
+Background loops: TTL reaper (inactivity teardown), pod monitor (auto
+re-provision to desired replicas), KPA autoscaler (knative-kind
+workloads on the local driver).
 
 Push semantics: one asyncio queue per connected pod; deploy broadcasts
 {action: reload, metadata, launch_id} to the service's pods and waits for
