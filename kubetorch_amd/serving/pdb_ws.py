@@ -26,9 +26,21 @@ class _SockIO:
         self._file.flush()
 
 
+_last_conn = [None]  # previous session's socket, closed on the next hit
+
+
 def deep_breakpoint(*_, **__):
     """PYTHONBREAKPOINT target: wait for one debugger client, run pdb over
-    the connection, resume on 'continue'."""
+    the connection, resume on 'continue'. set_trace is the LAST statement
+    (no finally, server socket closed beforehand): the first trace event
+    must land on the CALLER's next line, not inside this function's
+    cleanup code."""
+    if _last_conn[0] is not None:
+        try:
+            _last_conn[0].close()
+        except OSError:
+            pass
+        _last_conn[0] = None
     srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
     srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
     srv.bind(("0.0.0.0", DEBUG_PORT))
@@ -36,14 +48,12 @@ def deep_breakpoint(*_, **__):
     sys.stderr.write(
         f"[kt-debug] breakpoint hit; waiting for `kt debug` on :{DEBUG_PORT}\n")
     conn, addr = srv.accept()
+    _last_conn[0] = conn
     io = _SockIO(conn)
-    try:
-        io.write(f"[kt-debug] attached from {addr}\n")
-        dbg = pdb.Pdb(stdin=io, stdout=io)
-        frame = sys._getframe(1)
-        dbg.set_trace(frame)
-    finally:
-        srv.close()
+    io.write(f"[kt-debug] attached from {addr}\n")
+    dbg = pdb.Pdb(stdin=io, stdout=io)
+    srv.close()
+    dbg.set_trace(sys._getframe(1))
 
 
 def attach(host, port=DEBUG_PORT):
