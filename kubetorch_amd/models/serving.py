@@ -63,6 +63,9 @@ class BatchedGenerator:
 
             graph = (self.device.type == "cuda"
                      and os.environ.get("KT_DECODE_GRAPH", "1") != "0")
+        if graph and any(hasattr(m, "ep_world") for m in model.modules()):
+            graph = False  # MoE top-k dispatch is shape-dynamic: the
+            # variable-size expert index_selects cannot be graph-captured
         self._use_graph = bool(graph)
         self._graphs = {}      # L bucket -> (CUDAGraph, logits buffer)
         self._g_pool = None    # shared capture mempool across buckets

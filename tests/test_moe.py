@@ -172,3 +172,20 @@ def test_moe_trains_on_gpu():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0], losses
+
+
+def test_moe_decode_graph_disabled():
+    """MoE models must fall back to the eager decode step (dynamic expert
+    dispatch is not hipGraph-capturable)."""
+    import torch
+
+    from kubetorch_amd.models import BatchedGenerator, Llama, llama_tiny
+    from kubetorch_amd.models.moe import convert_to_moe
+
+    model = Llama(llama_tiny())
+    convert_to_moe(model, n_experts=4, top_k=2)
+    eng = BatchedGenerator(model, max_batch=2, max_len=64, graph=True)
+    assert eng._use_graph is False
+    dense = Llama(llama_tiny())
+    assert BatchedGenerator(dense, max_batch=2, max_len=64,
+                            graph=True)._use_graph is True
