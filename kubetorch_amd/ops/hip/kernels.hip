@@ -231,8 +231,11 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
     const size_t base = ((size_t)tok * Hh + h) * D + qi * 4;
     const size_t sbase = (size_t)tok * src_t_stride +
                          (size_t)h * src_h_stride + qi * 4;
-    vec4u x1 = *reinterpret_cast<const vec4u*>(x + sbase);
-    vec4u x2 = *reinterpret_cast<const vec4u*>(x + sbase + half);
+    // x streamed once (nontemporal); cos/sin tables stay cache-resident
+    vec4u x1 = __builtin_nontemporal_load(
+        reinterpret_cast<const vec4u*>(x + sbase));
+    vec4u x2 = __builtin_nontemporal_load(
+        reinterpret_cast<const vec4u*>(x + sbase + half));
     vec4f c = *reinterpret_cast<const vec4f*>(cost + (size_t)pos * half + qi * 4);
     vec4f s = *reinterpret_cast<const vec4f*>(sint + (size_t)pos * half + qi * 4);
     vec4u o1, o2;
@@ -242,8 +245,9 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
       o1[j] = f2bf((a * c[j] - sign * b * s[j]) * oscale);
       o2[j] = f2bf((b * c[j] + sign * a * s[j]) * oscale);
     }
-    *reinterpret_cast<vec4u*>(o + base) = o1;
-    *reinterpret_cast<vec4u*>(o + base + half) = o2;
+    __builtin_nontemporal_store(o1, reinterpret_cast<vec4u*>(o + base));
+    __builtin_nontemporal_store(
+        o2, reinterpret_cast<vec4u*>(o + base + half));
   }
 }
 
@@ -259,8 +263,10 @@ __global__ void swiglu_fwd_kernel(const u16* __restrict__ gu,
     const long row = t / nvec;
     const int vI = (int)(t - row * nvec);
     const size_t gbase = (size_t)row * 2 * I + vI * 8;
-    vec8u g = *reinterpret_cast<const vec8u*>(gu + gbase);
-    vec8u u = *reinterpret_cast<const vec8u*>(gu + gbase + I);
+    vec8u g = __builtin_nontemporal_load(
+        reinterpret_cast<const vec8u*>(gu + gbase));
+    vec8u u = __builtin_nontemporal_load(
+        reinterpret_cast<const vec8u*>(gu + gbase + I));
     vec8u o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -268,7 +274,8 @@ __global__ void swiglu_fwd_kernel(const u16* __restrict__ gu,
       float sig = 1.f / (1.f + __expf(-gf));
       o[j] = f2bf(gf * sig * bf2f(u[j]));
     }
-    *reinterpret_cast<vec8u*>(out + (size_t)row * I + vI * 8) = o;
+    __builtin_nontemporal_store(
+        o, reinterpret_cast<vec8u*>(out + (size_t)row * I + vI * 8));
   }
 }
 
@@ -282,9 +289,12 @@ __global__ void swiglu_bwd_kernel(const u16* __restrict__ dout,
     const long row = t / nvec;
     const int vI = (int)(t - row * nvec);
     const size_t gbase = (size_t)row * 2 * I + vI * 8;
-    vec8u g = *reinterpret_cast<const vec8u*>(gu + gbase);
-    vec8u u = *reinterpret_cast<const vec8u*>(gu + gbase + I);
-    vec8u dov = *reinterpret_cast<const vec8u*>(dout + (size_t)row * I + vI * 8);
+    vec8u g = __builtin_nontemporal_load(
+        reinterpret_cast<const vec8u*>(gu + gbase));
+    vec8u u = __builtin_nontemporal_load(
+        reinterpret_cast<const vec8u*>(gu + gbase + I));
+    vec8u dov = __builtin_nontemporal_load(
+        reinterpret_cast<const vec8u*>(dout + (size_t)row * I + vI * 8));
     vec8u dg, du;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -296,8 +306,9 @@ __global__ void swiglu_bwd_kernel(const u16* __restrict__ dout,
       dg[j] = f2bf(dof * uf * sig * (1.f + gf * (1.f - sig)));
       du[j] = f2bf(dof * silu);
     }
-    *reinterpret_cast<vec8u*>(dgu + gbase) = dg;
-    *reinterpret_cast<vec8u*>(dgu + gbase + I) = du;
+    __builtin_nontemporal_store(dg, reinterpret_cast<vec8u*>(dgu + gbase));
+    __builtin_nontemporal_store(du,
+                                reinterpret_cast<vec8u*>(dgu + gbase + I));
   }
 }
 
