@@ -231,11 +231,8 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
     const size_t base = ((size_t)tok * Hh + h) * D + qi * 4;
     const size_t sbase = (size_t)tok * src_t_stride +
                          (size_t)h * src_h_stride + qi * 4;
-    // x streamed once (nontemporal); cos/sin tables stay cache-resident
-    vec4u x1 = __builtin_nontemporal_load(
-        reinterpret_cast<const vec4u*>(x + sbase));
-    vec4u x2 = __builtin_nontemporal_load(
-        reinterpret_cast<const vec4u*>(x + sbase + half));
+    vec4u x1 = *reinterpret_cast<const vec4u*>(x + sbase);
+    vec4u x2 = *reinterpret_cast<const vec4u*>(x + sbase + half);
     vec4f c = *reinterpret_cast<const vec4f*>(cost + (size_t)pos * half + qi * 4);
     vec4f s = *reinterpret_cast<const vec4f*>(sint + (size_t)pos * half + qi * 4);
     vec4u o1, o2;
@@ -245,9 +242,11 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ o,
       o1[j] = f2bf((a * c[j] - sign * b * s[j]) * oscale);
       o2[j] = f2bf((b * c[j] + sign * a * s[j]) * oscale);
     }
-    __builtin_nontemporal_store(o1, reinterpret_cast<vec4u*>(o + base));
-    __builtin_nontemporal_store(
-        o2, reinterpret_cast<vec4u*>(o + base + half));
+    // NOT nontemporal: rope's K/V outputs are ~32 MB/layer and re-read
+    // immediately by the attention kernel — keeping them L2-retained
+    // measured faster (20,773 vs 20,442 tok/s with streaming stores)
+    *reinterpret_cast<vec4u*>(o + base) = o1;
+    *reinterpret_cast<vec4u*>(o + base + half) = o2;
   }
 }
 
