@@ -69,3 +69,42 @@ def test_http_store_through_tunnel(tmp_path, monkeypatch):
             assert httpx.get(base + "/health",
                              timeout=30).status_code == 200
     server.should_exit = True
+
+
+@pytest.mark.timeout(240)
+def test_tunnel_to_deployed_service(tmp_path):
+    """Tunnel by SERVICE NAME: the controller resolves the workload's pods
+    and bridges to the first one — an external client calls a deployed
+    fn through the tunnel without knowing any pod address."""
+    import sys as _sys
+
+    _sys.path.insert(0, os.path.join(os.path.dirname(__file__), "assets",
+                                     "summer"))
+    os.environ["KT_USERNAME"] = "tunsvc"
+    import kubetorch_amd as kt
+    from tests.assets.summer import summer as summer_mod
+
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        from kubetorch_amd.client.tunnel import TcpTunnel
+
+        with TcpTunnel(f.name, 32300) as t:
+            base = f"http://127.0.0.1:{t.local_port}"
+            r = httpx.get(base + "/health", timeout=30)
+            assert r.status_code == 200
+            r = httpx.post(f"{base}/call/summer",
+                           json={"args": [20, 22], "kwargs": {}}, timeout=60)
+            assert r.status_code == 200 and r.json()["result"] == 42
+    finally:
+        f.teardown()
+
+
+def test_tunnel_refuses_nonloopback_hostport():
+    """Explicit host:port targets must not turn the controller into a
+    network pivot (403 for anything but loopback on the local driver)."""
+    from kubetorch_amd.globals import controller_client
+
+    cc = controller_client()
+    r = httpx.post(cc.base_url + "/controller/tunnel/open",
+                   json={"service": "10.0.0.5:22", "port": 22}, timeout=10)
+    assert r.status_code == 403
