@@ -34,6 +34,15 @@ def init_distributed(backend=None, timeout_s=300):
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1 and torch.cuda.is_available():
+        # RCCL-over-xGMI defaults (also set by the SPMD launcher; this
+        # covers direct torchrun launches like the driver's SCALE run):
+        # dmabuf-only IPC, no InfiniBand probing intra-node, surfaced
+        # comm errors instead of silent hangs
+        from kubetorch_amd import constants as _C
+
+        for k, v in _C.RCCL_ENV_DEFAULTS.items():
+            os.environ.setdefault(k, v)
     if world > 1 and not dist.is_initialized():
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
