@@ -47,6 +47,15 @@ def hip_available():
         _ext()
         return True
     except Exception:
+        # On a GPU machine a missing extension must FAIL, not silently
+        # run the eager fallback at a fraction of the speed (the gfx950
+        # .so ships in-tree; its absence means a broken build/snapshot).
+        # KT_ALLOW_EAGER_FALLBACK=1 overrides for debugging.
+        import torch
+
+        if torch.cuda.is_available() and \
+                os.environ.get("KT_ALLOW_EAGER_FALLBACK") != "1":
+            raise
         return False
 
 
