@@ -348,6 +348,34 @@ def debug(service: str, port: int = typer.Option(None),
     attach(host, port or 4444)
 
 
+@app.command()
+def bench(steps: int = typer.Option(8), warmup: int = typer.Option(3),
+          gpus: int = typer.Option(1), batch: int = typer.Option(4),
+          seq: int = typer.Option(4096),
+          model: str = typer.Option("llama3-8b")):
+    """Run the flagship Llama-3-8B DDP benchmark THROUGH the dispatch
+    stack (deploy -> SPMD fan-out -> per-rank training loop) and print
+    the tokens/s JSON. `python bench.py` is the direct (launcher-free)
+    variant; the two agree within noise (profiles/r02_launcher.json)."""
+    import json as _json
+
+    import kubetorch_amd as kt
+    from kubetorch_amd.models.benchmark import bench_entry
+
+    f = kt.fn(bench_entry).to(
+        kt.Compute(gpus=gpus).distribute("pytorch", workers=1,
+                                         num_proc=max(1, gpus)))
+    try:
+        results = f(steps=steps, warmup=warmup, batch=batch, seq=seq,
+                    model=model, kt_timeout=3600)
+        if not isinstance(results, list):
+            results = [results]
+        result = next(r for r in results if r)
+        console.print_json(_json.dumps(result))
+    finally:
+        f.teardown()
+
+
 @app.command("tunnel")
 def tunnel(service: str, port: int = typer.Argument(8080),
            local_port: int = typer.Option(0, "--local-port"),
