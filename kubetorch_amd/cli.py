@@ -268,9 +268,10 @@ def run(command: str, name: str = typer.Option("app"),
     a.to(kt.Compute(cpus=cpus, gpus=gpus))
     console.print(f"[green]running[/green] {a.name}")
     if follow:
-        done = a.wait(timeout=timeout)
-        for e in a.logs(limit=200):
-            console.print(e.get("line", ""))
+        # wait() streams the app's stdout/stderr while blocking (app-mode
+        # pods run the user command directly — no HTTP server to query
+        # after exit)
+        done = a.wait(timeout=timeout, printer=console.print)
         a.teardown()
         if not done:
             console.print(f"[red]timed out after {timeout}s[/red]")

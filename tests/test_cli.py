@@ -67,3 +67,12 @@ def test_run_bash_and_pip_helpers():
         assert "hello-from-pod" in out["stdout"]
     finally:
         f.teardown()
+
+
+def test_run_follow_streams_and_exits():
+    r = runner.invoke(app, ["run", "echo from-kt-run && sleep 1",
+                            "--name", "clirun", "--follow",
+                            "--timeout", "120"])
+    assert r.exit_code == 0, r.output[-500:]
+    assert "from-kt-run" in r.output
+    assert "app finished" in r.output
