@@ -166,19 +166,39 @@ class ProcessPool:
                 resp = worker.resp_q.get(timeout=1.0)
             except (pyqueue.Empty, EOFError, OSError):
                 if not worker.alive():
+                    # a crashed worker (segfault, OOM-kill) must FAIL its
+                    # in-flight requests now, not let callers hang until
+                    # the HTTP timeout
+                    self._fail_worker_futures(
+                        worker.idx,
+                        f"worker {worker.idx} died "
+                        f"(exitcode {worker.proc.exitcode})")
                     return
                 continue
             with self._lock:
-                fut = self._futures.pop(resp["rid"], None)
+                entry = self._futures.pop(resp["rid"], None)
+            fut = entry[0] if entry else None
             if fut is not None and not fut.done():
                 fut.set_result(resp)
+
+    def _fail_worker_futures(self, idx, msg):
+        with self._lock:
+            dead = [rid for rid, (f, widx) in self._futures.items()
+                    if widx == idx]
+            entries = [self._futures.pop(rid) for rid in dead]
+        for fut, _w in entries:
+            if not fut.done():
+                fut.set_result(
+                    {"ok": False,
+                     "error": {"error_type": "PodTerminatedError",
+                               "message": msg, "traceback": ""}})
 
     def submit(self, idx, body_b64, method=None, env=None, fresh=False,
                request_id=None):
         rid = uuid.uuid4().hex
         fut = Future()
         with self._lock:
-            self._futures[rid] = fut
+            self._futures[rid] = (fut, idx)
         self.workers[idx].req_q.put(
             {"rid": rid, "body": body_b64, "method": method, "env": env,
              "fresh": fresh, "request_id": request_id}
@@ -212,7 +232,7 @@ class ProcessPool:
         for w in self.workers:
             w.terminate()
         with self._lock:
-            for fut in self._futures.values():
+            for fut, _idx in self._futures.values():
                 if not fut.done():
                     fut.set_result(
                         {"ok": False,

@@ -2,6 +2,7 @@
 worker-selection logic (reference parity: test_distributed.py asserts the
 RANK/WORLD_SIZE/... env on every rank and the worker-filter arg)."""
 import json
+import os
 
 from kubetorch_amd.serving.supervisors import (
     PROCESS_CLASSES,
@@ -11,6 +12,7 @@ from kubetorch_amd.serving.supervisors import (
 )
 
 HOSTS = ["10.0.0.1:32300", "10.0.0.2:32300", "10.0.0.3:32300"]
+ASSETS = os.path.join(os.path.dirname(__file__), "assets", "summer")
 
 
 def test_torch_env_contract():
@@ -197,3 +199,30 @@ def test_config_set_persist(tmp_path):
     assert path.exists()
     cfg2 = KTConfig(path=str(path))
     assert cfg2.namespace == "teamspace"
+
+
+def test_worker_crash_fails_inflight_request(monkeypatch):
+    """A worker killed mid-request fails the caller promptly with
+    PodTerminatedError instead of hanging until the HTTP timeout."""
+    import time
+
+    from kubetorch_amd.serving.process_pool import ProcessPool
+    from kubetorch_amd.serving.supervisors import _encode_call
+
+    monkeypatch.setenv("KT_FILE_PATH", os.path.join(ASSETS, "summer.py"))
+    monkeypatch.setenv("KT_PROJECT_ROOT", ASSETS)
+    monkeypatch.setenv("KT_CLS_OR_FN_NAME", "slow_echo")
+    monkeypatch.setenv("KT_MODULE_TYPE", "fn")
+    pool = ProcessPool(num_proc=1)
+    try:
+        body = _encode_call((1,), {"delay": 60})
+        fut = pool.submit(0, body)
+        time.sleep(2.0)  # request in flight inside the worker
+        pool.workers[0].proc.kill()
+        t0 = time.time()
+        resp = fut.result(timeout=30)
+        assert time.time() - t0 < 15, "crash not detected promptly"
+        assert resp["ok"] is False
+        assert resp["error"]["error_type"] == "PodTerminatedError"
+    finally:
+        pool.terminate()
