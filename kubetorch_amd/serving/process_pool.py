@@ -168,11 +168,18 @@ class ProcessPool:
                 if not worker.alive():
                     # a crashed worker (segfault, OOM-kill) must FAIL its
                     # in-flight requests now, not let callers hang until
-                    # the HTTP timeout
-                    self._fail_worker_futures(
-                        worker.idx,
-                        f"worker {worker.idx} died "
-                        f"(exitcode {worker.proc.exitcode})")
+                    # the HTTP timeout. Only if THIS worker is still the
+                    # pool's current worker for its slot — after a
+                    # deliberate restart() the old router must not fail
+                    # futures that belong to the replacement worker.
+                    with self._lock:
+                        current = (self.workers[worker.idx]
+                                   if worker.idx < len(self.workers) else None)
+                    if current is worker:
+                        self._fail_worker_futures(
+                            worker.idx,
+                            f"worker {worker.idx} died "
+                            f"(exitcode {worker.proc.exitcode})")
                     return
                 continue
             with self._lock:
