@@ -37,10 +37,20 @@ def extract_pointers(obj):
         if obj.__module__ in ("__main__",) and file_path.endswith(
                 ("<stdin>", "<ipython-input>")):
             raise TypeError("interactive source")
-    except TypeError:
+    except (TypeError, OSError):
         # notebook / REPL: dump the cell source to a real file
-        # (reference: resources/callables/utils.py notebook support)
-        src = inspect.getsource(obj)
+        # (reference: resources/callables/utils.py notebook support).
+        # inspect raises TypeError for builtins and OSError when no source
+        # is retrievable (python -c / piped stdin) — the latter cannot be
+        # recovered, so say what to do instead of a raw OSError.
+        try:
+            src = inspect.getsource(obj)
+        except OSError as e:
+            raise ValueError(
+                f"cannot extract source for {name!r}: it was defined in an "
+                "interactive/stdin context with no retrievable source. "
+                "Define it in a .py file (or a notebook cell) and retry."
+            ) from e
         nb_dir = os.path.join(os.getcwd(), ".kt_notebook")
         os.makedirs(nb_dir, exist_ok=True)
         file_path = os.path.join(nb_dir, f"{name}.py")
