@@ -202,6 +202,14 @@ class ProcessPool:
 
     def submit(self, idx, body_b64, method=None, env=None, fresh=False,
                request_id=None):
+        # self-heal: a worker that died while idle is respawned before the
+        # request is enqueued (a dead worker's queue accepts puts but
+        # nothing drains them)
+        if not self.workers[idx].alive():
+            with self._lock:
+                needs = not self.workers[idx].alive()
+            if needs:
+                self._start_worker(idx)
         rid = uuid.uuid4().hex
         fut = Future()
         with self._lock:

@@ -226,3 +226,26 @@ def test_worker_crash_fails_inflight_request(monkeypatch):
         assert resp["error"]["error_type"] == "PodTerminatedError"
     finally:
         pool.terminate()
+
+
+def test_idle_worker_death_self_heals(monkeypatch):
+    """A worker that dies while idle is respawned on the next submit (the
+    reference restarts workers on reload; death between requests must not
+    strand the queue)."""
+    from kubetorch_amd.serving.process_pool import ProcessPool
+    from kubetorch_amd.serving.supervisors import _decode_resp, _encode_call
+
+    monkeypatch.setenv("KT_FILE_PATH", os.path.join(ASSETS, "summer.py"))
+    monkeypatch.setenv("KT_PROJECT_ROOT", ASSETS)
+    monkeypatch.setenv("KT_CLS_OR_FN_NAME", "summer")
+    monkeypatch.setenv("KT_MODULE_TYPE", "fn")
+    pool = ProcessPool(num_proc=1)
+    try:
+        r = pool.submit(0, _encode_call((1, 2), {})).result(60)
+        assert _decode_resp(r) == 3
+        pool.workers[0].proc.kill()
+        pool.workers[0].proc.join(10)
+        r = pool.submit(0, _encode_call((5, 6), {})).result(60)
+        assert _decode_resp(r) == 11
+    finally:
+        pool.terminate()
