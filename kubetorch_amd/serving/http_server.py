@@ -433,11 +433,19 @@ async def exec_route(request: Request):
             status_code=403,
         )
     payload = await request.json()
-    cmd = payload["command"]
-    res = await asyncio.to_thread(
-        subprocess.run, ["bash", "-lc", cmd],
-        capture_output=True, text=True, timeout=payload.get("timeout", 600),
-    )
+    cmd = payload.get("command")
+    if not cmd:
+        return JSONResponse({"error": {"error_type": "ValueError",
+                                       "message": "missing 'command'",
+                                       "traceback": ""}}, status_code=400)
+    try:
+        res = await asyncio.to_thread(
+            subprocess.run, ["bash", "-lc", cmd],
+            capture_output=True, text=True,
+            timeout=payload.get("timeout", 600),
+        )
+    except subprocess.TimeoutExpired as e:
+        return JSONResponse({"error": package_exception(e)}, status_code=500)
     return {"returncode": res.returncode, "stdout": res.stdout[-20000:],
             "stderr": res.stderr[-20000:]}
 
