@@ -57,7 +57,14 @@ def reform_process_group(rank, world, master_addr, master_port,
 class FileRendezvous:
     """Directory-based survivor rendezvous for single-host worlds (the
     local driver, tests, one-node jobs). Each reform generation g uses
-    fresh files `g{g}-{uid}` and a fresh master port (base + g)."""
+    fresh files `g{g}-{uid}` and a fresh master port (base + g).
+
+    `settle` is how long the member set must be stable before concluding
+    — it MUST exceed the worst-case spread between survivors detecting
+    the failure (≈ the process-group timeout plus scheduling jitter), or
+    a fast detector can conclude a smaller world before slow detectors
+    arrive and the group splits. ElasticStepper wires this automatically
+    (settle >= pg_timeout_s + 2)."""
 
     def __init__(self, root, uid=None, master_addr="127.0.0.1",
                  port_base=29700, settle=2.0, timeout=60.0):
@@ -148,6 +155,13 @@ class ElasticStepper:
         self.backend = backend
         self.pg_timeout_s = pg_timeout_s
         self.reforms = 0
+        # survivors detect a failure up to ~pg_timeout apart; the
+        # rendezvous must wait at least that long for stragglers
+        min_settle = pg_timeout_s + 2.0
+        if getattr(rendezvous, "settle", None) is not None                 and rendezvous.settle < min_settle:
+            rendezvous.settle = min_settle
+        if getattr(rendezvous, "timeout", 0) < 4 * min_settle:
+            rendezvous.timeout = 4 * min_settle
 
     def step(self, forward_backward, lr=None):
         """Run forward_backward() + engine.step(); on a collective failure
