@@ -243,8 +243,14 @@ async def _pod_monitor(interval=None):
                 continue
             backoff[(ns, name)] = time.time()
             try:
+                # reconcile to the CURRENT desired count — knative-kind
+                # manifests carry no spec.replicas (the autoscaler owns
+                # desired_replicas), so stamp it into the applied copy
+                scaled = dict(manifest)
+                scaled["spec"] = dict(scaled.get("spec", {}),
+                                      replicas=desired)
                 await asyncio.to_thread(
-                    HUB.driver.apply, manifest, ns, w.get("metadata"),
+                    HUB.driver.apply, scaled, ns, w.get("metadata"),
                     w.get("launch_id"))
                 HUB.driver._event(ns, name, "Respawned",
                                   f"re-provisioned {desired - alive} pod(s)")

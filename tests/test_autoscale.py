@@ -81,3 +81,37 @@ def test_load_balanced_calls_spread_over_pods():
         assert len(seen) >= 2, f"calls pinned to one pod: {seen}"
     finally:
         f.teardown()
+
+
+@pytest.mark.timeout(300)
+def test_autoscaled_service_heals_dead_pod():
+    """Pod death in an autoscaled service: the reconciliation loops
+    (autoscaler owns desired replicas, monitor heals to them) cooperate —
+    the pod comes back and calls keep succeeding."""
+    f = kt.fn(summer_mod.pod_name).to(
+        kt.Compute(cpus=1).autoscale(target=1, min_scale=2, max_scale=2,
+                                     scale_down_delay="60s"))
+    from kubetorch_amd.controller.app import HUB
+
+    try:
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            if len(HUB.driver.pods(f.name, "default")) >= 2:
+                break
+            time.sleep(0.5)
+        pods = HUB.driver.services[("default", f.name)]
+        victim = pods[1]
+        victim.kill()
+        deadline = time.time() + 60
+        healed = []
+        while time.time() < deadline:
+            healed = HUB.driver.pods(f.name, "default")
+            if len(healed) == 2 and victim.host not in healed:
+                break
+            time.sleep(0.5)
+        assert len(healed) == 2 and victim.host not in healed, healed
+        time.sleep(2.5)  # lb cache refresh
+        seen = {f(kt_timeout=60) for _ in range(6)}
+        assert len(seen) >= 2, seen
+    finally:
+        f.teardown()
