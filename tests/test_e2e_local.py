@@ -277,3 +277,35 @@ def test_examples_run():
             env=env, capture_output=True, text=True, timeout=240)
         assert r.returncode == 0, (ex, r.stdout[-500:], r.stderr[-500:])
         assert needle in r.stdout, (ex, r.stdout[-300:])
+
+
+def test_reload_during_inflight_call():
+    """Hot reload while a call is executing: the in-flight request gets a
+    structured error (worker pool restarted), never a hang, and the next
+    call lands on the reloaded service."""
+    import threading
+    import time as _t
+
+    f = kt.fn(summer_mod.slow_echo).to(kt.Compute(cpus=1))
+    try:
+        result = {}
+
+        def call():
+            try:
+                result["value"] = f(7, delay=20, kt_timeout=120)
+            except Exception as e:  # noqa: BLE001
+                result["error"] = e
+        t = threading.Thread(target=call)
+        t.start()
+        _t.sleep(2.0)
+        f.to()  # hot reload into the warm pod (supervisor recreated)
+        t.join(60)
+        assert not t.is_alive(), "in-flight call hung through the reload"
+        # either the old worker finished first or a clean error came back
+        if "error" in result:
+            msg = str(result["error"])
+            assert "terminated" in msg or "died" in msg or \
+                "Terminated" in msg, msg
+        assert f(9, delay=0, kt_timeout=60) == 9
+    finally:
+        f.teardown()
