@@ -73,3 +73,55 @@ def test_300_subcalls_run_concurrently():
     assert STATE["peak"] > 50, f"peak concurrency only {STATE['peak']}"
     assert STATE["peak"] <= MAX_CONCURRENT_SUBCALLS + 1
     server.should_exit = True
+
+
+@pytest.mark.timeout(60)
+def test_subcall_error_reconstructed():
+    """A peer's packaged exception comes back as the real type."""
+    import socket as _socket
+    import threading as _threading
+    import time as _time
+
+    import httpx
+    import uvicorn
+    from fastapi import FastAPI
+
+    from kubetorch_amd.exceptions import package_exception
+
+    err_app = FastAPI()
+
+    @err_app.get("/health")
+    def _h():
+        return {"status": "ok"}
+
+    @err_app.post("/spmd/subcall")
+    async def _s():
+        from fastapi.responses import JSONResponse
+
+        try:
+            raise ValueError("remote-kaboom")
+        except ValueError as e:
+            return JSONResponse({"error": package_exception(e)},
+                                status_code=500)
+
+    s = _socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    server = uvicorn.Server(uvicorn.Config(err_app, host="127.0.0.1",
+                                           port=port, log_level="error"))
+    _threading.Thread(target=server.run, daemon=True).start()
+    deadline = _time.time() + 15
+    while _time.time() < deadline:
+        try:
+            if httpx.get(f"http://127.0.0.1:{port}/health",
+                         timeout=1).status_code == 200:
+                break
+        except Exception:
+            _time.sleep(0.05)
+    from kubetorch_amd.serving.remote_pool import fanout
+
+    fut = fanout().submit(f"127.0.0.1:{port}", "", None, [], None, 30)
+    with pytest.raises(ValueError, match="remote-kaboom"):
+        fut.result(30)
+    server.should_exit = True
