@@ -231,8 +231,9 @@ async def _pod_monitor(interval=None):
                 continue
             if (w.get("metadata") or {}).get("module_type") == "app":
                 continue  # apps run to completion; exit is not a fault
-            desired = w.get("desired_replicas") or \
-                manifest.get("spec", {}).get("replicas", 1)
+            from kubetorch_amd.controller.drivers import desired_replicas
+
+            desired = w.get("desired_replicas") or desired_replicas(manifest)
             try:
                 alive = len(HUB.driver.pods(name, ns))
             except Exception:

@@ -18,6 +18,20 @@ import time
 from kubetorch_amd import constants as C
 
 
+def desired_replicas(manifest):
+    """Replica count across manifest kinds: plain spec.replicas, or the
+    sum over a Kubeflow training job's replica specs."""
+    spec = manifest.get("spec", {})
+    if "replicas" in spec:
+        return spec["replicas"] or 1
+    for key in ("pytorchReplicaSpecs", "tfReplicaSpecs", "mxReplicaSpecs",
+                "xgbReplicaSpecs"):
+        if key in spec:
+            return sum(rs.get("replicas", 1)
+                       for rs in spec[key].values()) or 1
+    return 1
+
+
 def _free_port():
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
@@ -105,7 +119,7 @@ class LocalDriver:
 
     def _apply_locked(self, manifest, namespace, metadata, launch_id):
         name = manifest["metadata"]["name"]
-        replicas = manifest.get("spec", {}).get("replicas", 1)
+        replicas = desired_replicas(manifest)
         key = (namespace, name)
         pods = self.services.get(key, [])
         alive = [p for p in pods if p.alive()]

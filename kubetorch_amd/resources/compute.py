@@ -110,7 +110,12 @@ class Compute:
 
     def to_manifest(self, service_name, username=None, module=None):
         if self._raw_manifest is not None:
-            return copy.deepcopy(self._raw_manifest)
+            m = copy.deepcopy(self._raw_manifest)
+            # the service owns the name: the driver registers pods under
+            # metadata.name, and the client queries by service name — a
+            # BYO manifest keeping its own name would orphan its pods
+            m.setdefault("metadata", {})["name"] = service_name
+            return m
         pod_kw = dict(
             env=self.pod_env(), cpus=self.cpus, memory=self.memory,
             gpus=self.gpus, gpu_type=self.gpu_type,

@@ -131,3 +131,25 @@ def test_volume_existing_pv_bind():
     # existing claim -> no creation needed
     assert not kt.Volume("x", existing_claim="already").needs_create
     assert kt.Volume("y").needs_create
+
+
+@pytest.mark.timeout(300)
+def test_byo_pytorchjob_manifest_runs_distributed():
+    """BYO: a prebuilt PyTorchJob manifest deploys through
+    Compute.from_manifest and the local driver spawns the job's full
+    replica count; the auto-derived SPMD config fans the call out to
+    every rank (reference: test_byo_manifest.py with PyTorchJob CRDs)."""
+    base = kt.Compute(cpus=1)
+    base.kind = "pytorchjob"
+    base.distributed_config = {"type": "pytorch", "workers": 2, "num_proc": 1}
+    manifest = base.to_manifest("byojob")
+    comp = kt.Compute.from_manifest(manifest)
+    assert comp.kind == "pytorchjob"
+    assert comp.distributed_config["workers"] == 2
+    f = kt.fn(summer_mod.rank_env).to(comp)
+    try:
+        results = f(kt_timeout=180)
+        assert len(results) == 2, results
+        assert sorted(r["rank"] for r in results) == [0, 1]
+    finally:
+        f.teardown()
