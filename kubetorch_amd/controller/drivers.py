@@ -29,6 +29,9 @@ def desired_replicas(manifest):
         if key in spec:
             return sum(rs.get("replicas", 1)
                        for rs in spec[key].values()) or 1
+    if "headGroupSpec" in spec:  # RayCluster: head + worker groups
+        return 1 + sum(g.get("replicas", 0)
+                       for g in spec.get("workerGroupSpecs", []))
     return 1
 
 

@@ -145,7 +145,14 @@ class ExecutionSupervisor:
     def _base_env(self, idx):
         return {}
 
-    def call(self, args=(), kwargs=None, method=None, timeout=None, **_):
+    def call(self, args=(), kwargs=None, method=None, timeout=None,
+             serialized_body=None, **_):
+        if serialized_body is not None:
+            # distributed-flagged supervisors without SPMD fan-out (Ray,
+            # Monarch) receive the pre-serialized body from run_callable
+            resp = self.pool.submit(0, serialized_body, method=method).result(
+                timeout or C.HTTP_TIMEOUT * 10)
+            return _decode_resp(resp)
         resp = self.pool.call(0, args, kwargs, method=method, timeout=timeout)
         return _decode_resp(resp)
 

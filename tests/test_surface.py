@@ -153,3 +153,20 @@ def test_byo_pytorchjob_manifest_runs_distributed():
         assert sorted(r["rank"] for r in results) == [0, 1]
     finally:
         f.teardown()
+
+
+@pytest.mark.timeout(240)
+def test_raycluster_local_pod_count():
+    """distribute('ray', workers=3) renders a RayCluster (head + 2 worker
+    groups); the local driver spawns all 3 pods (head included)."""
+    f = kt.fn(summer_mod.summer).to(
+        kt.Compute(cpus=1).distribute("ray", workers=3))
+    from kubetorch_amd.controller.app import HUB
+
+    try:
+        assert len(HUB.driver.pods(f.name, "default")) == 3
+        # no ray wheel on this image: the head-only supervisor degrades to
+        # a single-worker call path, which still serves
+        assert f(20, 22, kt_timeout=120) == 42
+    finally:
+        f.teardown()
