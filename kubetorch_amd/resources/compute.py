@@ -59,6 +59,10 @@ class Compute:
                    quorum_timeout=C.QUORUM_TIMEOUT):
         """Configure the SPMD launcher: `workers` pods x `num_proc` local
         ranks (auto = one per MI355X GPU via torch.cuda.device_count())."""
+        if self.autoscaling is not None:
+            raise ValueError(
+                "distribute() and autoscale() are mutually exclusive: SPMD "
+                "ranks need a stable worker set")
         if framework == "ray":
             self.kind = "raycluster"
             self.replicas = workers
@@ -74,6 +78,11 @@ class Compute:
         return self
 
     def autoscale(self, autoscaling=None, **kw):
+        if self.distributed_config is not None:
+            raise ValueError(
+                "autoscale() and distribute() are mutually exclusive: SPMD "
+                "ranks need a stable worker set (scale jobs by re-deploying "
+                "with a different workers=)")
         self.kind = "knative"
         self.autoscaling = autoscaling or AutoscalingConfig(**kw)
         return self

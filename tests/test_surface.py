@@ -170,3 +170,10 @@ def test_raycluster_local_pod_count():
         assert f(20, 22, kt_timeout=120) == 42
     finally:
         f.teardown()
+
+
+def test_autoscale_distribute_mutually_exclusive():
+    with pytest.raises(ValueError, match="mutually exclusive"):
+        kt.Compute(cpus=1).distribute("pytorch", workers=2).autoscale(target=1)
+    with pytest.raises(ValueError, match="mutually exclusive"):
+        kt.Compute(cpus=1).autoscale(target=1).distribute("pytorch")
