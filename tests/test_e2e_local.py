@@ -309,3 +309,16 @@ def test_reload_during_inflight_call():
         assert f(9, delay=0, kt_timeout=60) == 9
     finally:
         f.teardown()
+
+
+def test_distributed_cls_per_rank_state():
+    """kt.cls + distribute: each rank holds its own instance; a method
+    call fans out and returns the per-rank results."""
+    c = kt.cls(summer_mod.Counter, init_args=((), {"start": 100})).to(
+        kt.Compute(cpus=1).distribute("pytorch", workers=2, num_proc=1))
+    try:
+        assert c.add(5, kt_timeout=120) == [105, 105]
+        assert c.add(1, kt_timeout=120) == [106, 106]  # state persists
+        assert c.get(kt_timeout=120) == [106, 106]
+    finally:
+        c.teardown()
