@@ -46,10 +46,38 @@ class App(Module):
         return super().to()
 
     def _wait_ready(self, timeout=None, reloaded=False):
-        if self.compute.local:
-            return  # local driver runs the command; no HTTP server to poll
-        if self.health_path:
-            super()._wait_ready(timeout=timeout or 900, reloaded=reloaded)
+        if not self.health_path:
+            return  # fire-and-forget command (kt run)
+        # an app serves ITS OWN http endpoint: poll health_path for any
+        # non-5xx answer (our /ready route doesn't exist there)
+        import time as _t
+
+        import httpx
+
+        from kubetorch_amd.globals import controller_client
+
+        deadline = _t.time() + (timeout or 900)
+        path = "/" + self.health_path.lstrip("/")
+        while _t.time() < deadline:
+            try:
+                w = controller_client().get_workload(self.name, self.namespace)
+                pods = (w or {}).get("pods") or self.service_hosts
+            except Exception:
+                pods = self.service_hosts
+            for host in pods:
+                port = self.port or host.split(":")[-1]
+                url = f"http://{host.split(':')[0]}:{port}{path}"
+                try:
+                    if httpx.get(url, timeout=3).status_code < 500:
+                        return
+                except httpx.HTTPError:
+                    pass
+            _t.sleep(0.5)
+        from kubetorch_amd.exceptions import LaunchError
+
+        raise LaunchError(
+            f"app {self.name} health path {path} not answering "
+            f"after {timeout or 900}s")
 
 
     def wait(self, timeout=900, poll=1.0, follow=True, printer=print):

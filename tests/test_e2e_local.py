@@ -322,3 +322,28 @@ def test_distributed_cls_per_rank_state():
         assert c.get(kt_timeout=120) == [106, 106]
     finally:
         c.teardown()
+
+
+def test_app_health_path_readiness():
+    """An HTTP app with health_path: .to() blocks until the app's OWN
+    endpoint answers (not our /ready route, which apps don't serve)."""
+    import socket as _socket
+
+    s = _socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    a = kt.app(f"sleep 2 && exec python -m http.server {port} "
+               f"--bind 127.0.0.1", name="httpapp", health_path="/",
+               port=port)
+    t0 = time.time()
+    a.to(kt.Compute(cpus=1))
+    waited = time.time() - t0
+    try:
+        assert waited >= 1.5, f"did not wait for the app ({waited:.1f}s)"
+        import httpx
+
+        assert httpx.get(f"http://127.0.0.1:{port}/",
+                         timeout=10).status_code == 200
+    finally:
+        a.teardown()
