@@ -76,3 +76,21 @@ def test_run_follow_streams_and_exits():
     assert r.exit_code == 0, r.output[-500:]
     assert "from-kt-run" in r.output
     assert "app finished" in r.output
+
+
+def test_apply_manifest_file(tmp_path):
+    """kt apply: a raw Deployment manifest file becomes running pods
+    through the controller."""
+    import yaml
+
+    manifest = kt.Compute(cpus=1).to_manifest("cli-applied")
+    path = tmp_path / "dep.yaml"
+    path.write_text(yaml.safe_dump(manifest))
+    r = runner.invoke(app, ["apply", str(path)])
+    assert r.exit_code == 0, r.output[-400:]
+    from kubetorch_amd.controller.app import HUB
+
+    try:
+        assert len(HUB.driver.pods("cli-applied", "default")) == 1
+    finally:
+        runner.invoke(app, ["teardown", "cli-applied"])
