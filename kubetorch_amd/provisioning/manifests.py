@@ -40,7 +40,7 @@ def build_pod_spec(service_name, image, command=None, env=None, cpus=None,
     container = {
         "name": "kubetorch",
         "image": image,
-        "ports": [{"containerPort": port}],
+        "ports": [{"name": "http", "containerPort": port}],
         "env": env_list,
         "resources": resources,
         "command": command or [
