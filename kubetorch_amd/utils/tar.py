@@ -21,9 +21,19 @@ def safe_extractall(tar: tarfile.TarFile, dest: str):
     dest = os.path.abspath(dest)
     members = []
     for m in tar.getmembers():
-        name = m.name
-        if os.path.isabs(name) or not _resolves_inside(dest, name):
+        # normalize "./a", "a/./b" and collapse inner "a/../b"; anything that
+        # still *leads* with ".." after normpath points above dest → reject.
+        # (names with dot segments also break tarfile mechanically even when
+        # they resolve inside: "0/../y" hits makedirs("0/..") EEXIST)
+        name = m.name = os.path.normpath(m.name)
+        if os.path.isabs(name) or name == ".." or name.startswith("../") \
+                or not _resolves_inside(dest, name):
             raise ValueError(f"tar member escapes destination: {name!r}")
+        if m.isfile() and os.path.realpath(
+                os.path.join(dest, name)) == os.path.realpath(dest):
+            # a REGULAR file named "." would try to open the dest dir itself
+            raise ValueError(f"tar file member resolves to destination root: "
+                             f"{name!r}")
         if m.issym() or m.islnk():
             link = m.linkname
             base = os.path.dirname(name)
