@@ -183,3 +183,53 @@ def test_bucket_for_invariants(need, max_len):
 def test_bucket_for_monotone(need):
     b = _Bucketer(8192)
     assert b._bucket_for(need) <= b._bucket_for(need + 1)
+
+
+# -- autoscaling annotation round-trip ---------------------------------------
+
+@SETTINGS
+@given(st.integers(min_value=1, max_value=10_000),
+       st.integers(min_value=0, max_value=64),
+       st.integers(min_value=0, max_value=128),
+       st.integers(min_value=1, max_value=3600))
+def test_autoscale_spec_roundtrip(target, min_scale, max_scale, window_s):
+    """AutoscalingConfig -> knative manifest annotations -> the controller's
+    _autoscale_spec must agree on every scaling parameter."""
+    from kubetorch_amd.controller.app import _autoscale_spec
+    from kubetorch_amd.provisioning.manifests import build_knative_manifest
+    from kubetorch_amd.resources.autoscaling import AutoscalingConfig
+
+    cfg = AutoscalingConfig(target=target, min_scale=min_scale,
+                            max_scale=max_scale, window=f"{window_s}s")
+    m = build_knative_manifest("svc", "ns", "img", autoscaling=cfg)
+    spec = _autoscale_spec(m)
+    assert spec is not None
+    assert spec["target"] == target
+    assert spec["min"] == max(1, min_scale)   # controller floors at 1
+    assert spec["max"] == max_scale           # 0 = unlimited
+    assert spec["window"] == float(window_s)
+
+
+@SETTINGS
+@given(st.floats(min_value=0, max_value=10_000,
+                 allow_nan=False, allow_infinity=False),
+       st.sampled_from(["s", "m", "h", ""]))
+def test_parse_duration(value, unit):
+    from kubetorch_amd.controller.app import _parse_duration
+    mult = {"s": 1, "m": 60, "h": 3600, "": 1}[unit]
+    got = _parse_duration(f"{value}{unit}")
+    assert abs(got - value * mult) < 1e-6
+
+
+def test_parse_duration_garbage_defaults():
+    from kubetorch_amd.controller.app import _parse_duration
+    assert _parse_duration("", 7.0) == 7.0
+    assert _parse_duration(None, 7.0) == 7.0
+    assert _parse_duration("not-a-time", 7.0) == 7.0
+
+
+def test_autoscale_spec_none_for_plain_deployment():
+    from kubetorch_amd.controller.app import _autoscale_spec
+    from kubetorch_amd.provisioning.manifests import build_deployment_manifest
+    m = build_deployment_manifest("svc", "ns", "img")
+    assert _autoscale_spec(m) is None
