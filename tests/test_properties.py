@@ -233,3 +233,36 @@ def test_autoscale_spec_none_for_plain_deployment():
     from kubetorch_amd.provisioning.manifests import build_deployment_manifest
     m = build_deployment_manifest("svc", "ns", "img")
     assert _autoscale_spec(m) is None
+
+
+# -- Image dockerfile round-trip ---------------------------------------------
+
+_PAYLOAD = st.text(
+    alphabet=st.characters(blacklist_categories=("Cs", "Cc"),
+                           blacklist_characters="\n\r"),
+    min_size=1, max_size=60,
+).map(str.strip).filter(lambda s: s and not s.startswith("#"))
+
+
+@SETTINGS
+@given(st.lists(st.tuples(st.sampled_from(["RUN", "ENV", "COPY", "CMD",
+                                           "SYNC"]),
+                          _PAYLOAD), max_size=8))
+def test_image_dockerfile_roundtrip(steps):
+    """Image.contents() -> from_dockerfile() preserves base image and the
+    ordered step list (the reload differ depends on this being stable)."""
+    from kubetorch_amd.resources.image import Image
+
+    img = Image("rocm/pytorch:latest")
+    img.steps = list(steps)
+    back = Image.from_dockerfile(img.contents())
+    assert back.image_id == "rocm/pytorch:latest"
+    assert back.steps == list(steps)
+
+
+def test_image_dockerfile_skips_comments_and_blanks():
+    from kubetorch_amd.resources.image import Image
+    img = Image.from_dockerfile(
+        "# a comment\n\nFROM base\nRUN echo hi\n  \n# more\nENV A=1\n")
+    assert img.image_id == "base"
+    assert img.steps == [("RUN", "echo hi"), ("ENV", "A=1")]
