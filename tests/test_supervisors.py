@@ -249,3 +249,28 @@ def test_idle_worker_death_self_heals(monkeypatch):
         assert _decode_resp(r) == 11
     finally:
         pool.terminate()
+
+
+def test_pool_concurrent_routing_integrity(monkeypatch):
+    """Many interleaved in-flight requests across 2 workers: every future
+    must get ITS OWN answer (the response router matches request ids under
+    concurrency — advisor-flagged lock path)."""
+    from kubetorch_amd.serving.process_pool import ProcessPool
+    from kubetorch_amd.serving.supervisors import _decode_resp, _encode_call
+
+    monkeypatch.setenv("KT_FILE_PATH", os.path.join(ASSETS, "summer.py"))
+    monkeypatch.setenv("KT_PROJECT_ROOT", ASSETS)
+    monkeypatch.setenv("KT_CLS_OR_FN_NAME", "summer")
+    monkeypatch.setenv("KT_MODULE_TYPE", "fn")
+    pool = ProcessPool(num_proc=2)
+    try:
+        futs = []
+        for i in range(40):
+            body = _encode_call((i, 1000 * i), {})
+            futs.append((i, pool.submit(i % 2, body)))
+        for i, fut in futs:
+            resp = fut.result(timeout=120)
+            assert resp["ok"], resp
+            assert _decode_resp(resp) == i + 1000 * i
+    finally:
+        pool.terminate()
