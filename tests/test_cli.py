@@ -94,3 +94,46 @@ def test_apply_manifest_file(tmp_path):
         assert len(HUB.driver.pods("cli-applied", "default")) == 1
     finally:
         runner.invoke(app, ["teardown", "cli-applied"])
+
+
+def test_data_verbs_roundtrip(tmp_path):
+    """kt put/ls/get/rm through the CLI against the local store."""
+    src = tmp_path / "payload.txt"
+    src.write_text("cli data plane")
+    r = runner.invoke(app, ["put", "clitest/data1", str(src)])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(app, ["ls", "clitest"])
+    assert r.exit_code == 0 and "data1" in r.output
+    dest = tmp_path / "out"
+    r = runner.invoke(app, ["get", "clitest/data1", str(dest)])
+    assert r.exit_code == 0, r.output
+    got = dest / "payload.txt"
+    assert (got.read_text() if got.exists() else dest.read_text()) \
+        == "cli data plane"
+    r = runner.invoke(app, ["rm", "clitest/data1"])
+    assert r.exit_code == 0
+    r = runner.invoke(app, ["ls", "clitest"])
+    assert "data1" not in r.output
+
+
+def test_logs_and_workload_verbs():
+    """kt logs tails a deployed fn's captured stdout; kt workload shows the
+    registered module metadata."""
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        f(2, 3)  # prints "summing 2+3" inside the pod
+        r = runner.invoke(app, ["workload", f.name])
+        assert r.exit_code == 0, r.output
+        assert f.name in r.output
+        # LogCapture batches (1 s flush): poll until the line lands
+        import time
+        deadline = time.time() + 15
+        out = ""
+        while time.time() < deadline and "summing" not in out:
+            r = runner.invoke(app, ["logs", f.name])
+            assert r.exit_code == 0, r.output
+            out = r.output
+            time.sleep(0.5)
+        assert "summing" in out
+    finally:
+        f.teardown()
