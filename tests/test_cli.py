@@ -137,3 +137,21 @@ def test_logs_and_workload_verbs():
         assert "summing" in out
     finally:
         f.teardown()
+
+
+def test_bench_verb_through_dispatch_stack():
+    """kt bench --model tiny on CPU: the flagship bench runs THROUGH the
+    deploy -> SPMD fan-out path and prints the tokens/s JSON line."""
+    import json
+    import re
+
+    r = runner.invoke(app, ["bench", "--model", "tiny", "--gpus", "0",
+                            "--steps", "2", "--warmup", "1",
+                            "--batch", "1", "--seq", "256"])
+    assert r.exit_code == 0, r.output
+    m = re.search(r"\{.*\}", r.output.replace("\n", ""))
+    assert m, r.output
+    result = json.loads(m.group(0))
+    assert result["steps"] == 2
+    assert result["value"] > 0
+    assert result["metric"] == "tiny_ddp_tokens_per_sec"
