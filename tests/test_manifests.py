@@ -1,5 +1,9 @@
 """Manifest/unit tests without a cluster (reference parity: test_compute.py,
 test_autodown.py, test_kueue.py style — assert the generated YAML)."""
+import os
+
+import pytest
+
 import kubetorch_amd as kt
 from kubetorch_amd import constants as C
 from kubetorch_amd.resources.autoscaling import AutoscalingConfig
@@ -222,3 +226,44 @@ def test_cluster_config_layer(tmp_path, monkeypatch):
     assert c.get("image") == "file-img"
     monkeypatch.setenv("KT_IMAGE", "env-img")
     assert c.get("image") == "env-img"
+
+
+def test_image_setup_env_and_run_and_cmd(monkeypatch, tmp_path):
+    """run_step semantics: ENV expands $vars into os.environ, RUN executes
+    with expansion and raises on failure, CMD manages the app process
+    (old process terminated on re-CMD)."""
+    import time
+
+    from kubetorch_amd.serving import image_setup
+
+    monkeypatch.setenv("KT_BASE", str(tmp_path))
+    image_setup.run_step("ENV", "KT_SUB=$KT_BASE/sub")
+    assert os.environ["KT_SUB"] == f"{tmp_path}/sub"
+
+    image_setup.run_step("RUN", "mkdir -p $KT_SUB && echo hi > $KT_SUB/f")
+    assert (tmp_path / "sub" / "f").read_text().strip() == "hi"
+
+    with pytest.raises(RuntimeError, match="image step failed"):
+        image_setup.run_step("RUN", "exit 3")
+
+    state = {}
+    image_setup.run_step("CMD", "sleep 30", app_state=state)
+    first = state["app_proc"]
+    assert first.poll() is None
+    image_setup.run_step("CMD", "sleep 30", app_state=state)
+    second = state["app_proc"]
+    assert second is not first
+    deadline = time.time() + 10
+    while first.poll() is None and time.time() < deadline:
+        time.sleep(0.1)
+    assert first.poll() is not None, "old CMD process not terminated"
+    second.terminate()
+    second.wait(10)
+
+
+def test_image_setup_copy_missing_target_is_nonfatal(capsys):
+    from kubetorch_amd.serving import image_setup
+
+    image_setup.run_step("COPY", "src /definitely/not/present")
+    out = capsys.readouterr().out
+    assert "not present" in out
