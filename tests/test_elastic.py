@@ -70,3 +70,58 @@ def test_kill_pod_mid_call_then_rejoin():
         assert any(e["reason"] == "Respawned" for e in evs)
     finally:
         f.teardown()
+
+
+def test_peers_rendezvous_rank_order_and_generation(monkeypatch):
+    """PeersRendezvous concludes rank/world from the controller's live peer
+    list: sorted-host rank order, master = rank 0's host, fresh port per
+    generation, and a clear error when self is missing (we were replaced)."""
+    from kubetorch_amd.parallel.elastic import PeersRendezvous
+    from kubetorch_amd.serving import discovery, supervisors
+
+    peers = ["10.0.0.3:32300", "10.0.0.1:32300", "10.0.0.2:32300"]
+    monkeypatch.setattr(discovery, "current_peers",
+                        lambda *a, **k: list(peers))
+    monkeypatch.setattr(supervisors, "_self_host", lambda: "10.0.0.2:32300")
+
+    rdv = PeersRendezvous(service_name="svc", settle=0.3, quorum_timeout=10)
+    rank, world, addr, port = rdv()
+    assert (rank, world) == (1, 3)        # sorted order, me second
+    assert addr == "10.0.0.1"             # master = rank 0's host
+    rank2, world2, addr2, port2 = rdv()
+    assert port2 == port + 1              # fresh port per generation
+
+    # a shrunk group re-ranks the survivors
+    peers[:] = ["10.0.0.3:32300", "10.0.0.2:32300"]
+    rank3, world3, addr3, _ = rdv()
+    assert (rank3, world3) == (0, 2)
+    assert addr3 == "10.0.0.2"
+
+    # self evicted from the peer list -> loud error, not a silent hang
+    monkeypatch.setattr(supervisors, "_self_host", lambda: "10.9.9.9:32300")
+    with pytest.raises(RuntimeError, match="not in peer list"):
+        rdv()
+
+
+def test_peers_rendezvous_waits_for_settle(monkeypatch):
+    """A flapping peer list (pod still being respawned) delays conclusion
+    until the membership is stable for `settle` seconds."""
+    import itertools
+
+    from kubetorch_amd.parallel.elastic import PeersRendezvous
+    from kubetorch_amd.serving import discovery, supervisors
+
+    t0 = time.time()
+    flap_until = t0 + 1.0
+
+    def peers(*a, **k):
+        if time.time() < flap_until:  # membership still churning
+            return ["10.0.0.1:1", f"10.0.0.{int(time.time()*10)%5+2}:1"]
+        return ["10.0.0.1:1", "10.0.0.2:1"]
+
+    monkeypatch.setattr(discovery, "current_peers", peers)
+    monkeypatch.setattr(supervisors, "_self_host", lambda: "10.0.0.1:1")
+    rdv = PeersRendezvous(service_name="svc", settle=0.5, quorum_timeout=20)
+    rank, world, _, _ = rdv()
+    assert (rank, world) == (0, 2)
+    assert time.time() - t0 >= 1.4  # flap window + settle, not instant
