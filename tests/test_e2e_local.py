@@ -271,7 +271,8 @@ def test_examples_run():
                    os.path.abspath(__file__))))
     root = env["PYTHONPATH"]
     for ex, needle in (("01_hello_world.py", "hello"),
-                       ("05_serving_decode.py", "generated:")):
+                       ("05_serving_decode.py", "generated:"),
+                       ("08_production_surface.py", "'image_env': 'prod'")):
         r = subprocess.run(
             [sys.executable, os.path.join(root, "examples", ex)],
             env=env, capture_output=True, text=True, timeout=240)
