@@ -107,24 +107,45 @@ def apply_metadata(md: dict):
         os.environ[C.ENV_DISTRIBUTED_CONFIG] = json.dumps(md["distributed_config"])
 
 
+_SUP_SIG_KEYS = (C.ENV_DISTRIBUTED_CONFIG, C.ENV_MODULE_NAME,
+                 C.ENV_CALLABLE_NAME, C.ENV_FILE_PATH, C.ENV_PROJECT_ROOT,
+                 C.ENV_INIT_ARGS, C.ENV_MODULE_TYPE, "KT_WORKDIR_KEY",
+                 "KT_REL_PATH")
+
+
+def _sup_sig():
+    return tuple(os.environ.get(k) for k in _SUP_SIG_KEYS)
+
+
 def do_reload(md: dict, launch_id=None):
     """Hot reload: apply metadata, re-sync code, recreate supervisor.
-    The launch_id is only set after success so /ready gates correctly."""
-    apply_metadata(md or {})
-    try:
-        from kubetorch_amd.data_store import commands as ds
+    The launch_id is only set after success so /ready gates correctly.
 
-        ds.sync_workdir_from_store()
-    except Exception:
-        pass
+    A genuine reload (launch_id set) always recreates; a plain metadata
+    push recreates only if the supervisor-relevant config actually changed
+    (reference parity: hash of distributed config decides recreate,
+    http_server.py:878-1137) — otherwise the registration-time metadata
+    echo would terminate the pool under an in-flight first call."""
+    before = _sup_sig()
+    apply_metadata(md or {})
+    rebuild = bool(launch_id) or STATE["supervisor"] is None \
+        or _sup_sig() != before
+    if rebuild:
+        try:
+            from kubetorch_amd.data_store import commands as ds
+
+            ds.sync_workdir_from_store()
+        except Exception:
+            pass
     if md and md.get("image_setup"):
         from kubetorch_amd.serving import image_setup
 
         image_setup.cached_image_setup(md["image_setup"], app_state=STATE)
-    from kubetorch_amd.serving import loading
+    if rebuild:
+        from kubetorch_amd.serving import loading
 
-    loading.clear_cache()
-    get_supervisor(recreate=True)
+        loading.clear_cache()
+        get_supervisor(recreate=True)
     if launch_id:
         STATE["launch_id"] = str(launch_id)
         os.environ[C.ENV_LAUNCH_ID] = str(launch_id)

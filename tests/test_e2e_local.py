@@ -350,16 +350,21 @@ def test_app_health_path_readiness():
         a.teardown()
 
 
-def test_concurrent_calls_same_service(remote_fn):
+def test_concurrent_calls_same_service():
     """20 threads calling the same deployed fn concurrently: every caller
     gets its own correct answer (HTTPClient + supervisor + process pool
-    are thread-safe end to end)."""
+    are thread-safe end to end). Deploys its own service: earlier tests
+    restart the controller, which kills the module fixture's pods."""
     import concurrent.futures as cf
 
-    with cf.ThreadPoolExecutor(max_workers=20) as ex:
-        futs = {ex.submit(remote_fn, i, 1000 * i): i for i in range(20)}
-        for fut, i in futs.items():
-            assert fut.result(timeout=120) == i + 1000 * i
+    f = kt.fn(summer_mod.summer, name="concur").to(kt.Compute(cpus=1))
+    try:
+        with cf.ThreadPoolExecutor(max_workers=20) as ex:
+            futs = {ex.submit(f, i, 1000 * i): i for i in range(20)}
+            for fut, i in futs.items():
+                assert fut.result(timeout=120) == i + 1000 * i
+    finally:
+        f.teardown()
 
 
 def test_concurrent_deploys_same_name():

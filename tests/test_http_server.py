@@ -136,3 +136,30 @@ def test_metrics_and_logs_endpoints(client):
     logs = client.get("/logs/tail", params={"limit": 10})
     assert logs.status_code == 200
     assert isinstance(logs.json()["entries"], list)
+
+
+def test_metadata_echo_keeps_supervisor(client):
+    """A registration-time metadata push that changes nothing must NOT
+    recreate the supervisor (it would terminate the pool under an
+    in-flight first call); a changed config or a launch_id must."""
+    from kubetorch_amd.serving import http_server
+
+    sup0 = http_server.get_supervisor()
+    md = {"callable_name": "summer", "module_name": "summer",
+          "file_path": os.environ["KT_FILE_PATH"],
+          "project_root": os.environ["KT_PROJECT_ROOT"],
+          "module_type": "fn"}
+    http_server.do_reload(md, launch_id=None)       # metadata echo
+    assert http_server.STATE["supervisor"] is sup0, \
+        "unchanged metadata push recreated the supervisor"
+
+    http_server.do_reload(md, launch_id="unit-lid-2")  # genuine reload
+    sup1 = http_server.STATE["supervisor"]
+    assert sup1 is not sup0
+    assert http_server.STATE["launch_id"] == "unit-lid-2"
+
+    md2 = dict(md, callable_name="slow_echo", module_name="slow_echo")
+    http_server.do_reload(md2, launch_id=None)      # changed config
+    assert http_server.STATE["supervisor"] is not sup1
+    # restore for any later tests in this module
+    http_server.do_reload(md, launch_id="unit-lid-1")
