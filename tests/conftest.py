@@ -74,6 +74,13 @@ def pytest_collection_modifyitems(config, items):
             if ok:
                 keep.append(item)
         items[:] = keep
+    # GPU tests keep the old 600 s budget (first-touch kernel compiles,
+    # pg-timeout waits in the elastic drills); the 300 s pytest.ini default
+    # is sized for the CPU suite, whose slowest test is ~31 s
+    for item in items:
+        if "gpu" in item.keywords and item.get_closest_marker(
+                "timeout") is None:
+            item.add_marker(pytest.mark.timeout(600))
     try:
         import torch
 
