@@ -266,3 +266,51 @@ def test_image_dockerfile_skips_comments_and_blanks():
         "# a comment\n\nFROM base\nRUN echo hi\n  \n# more\nENV A=1\n")
     assert img.image_id == "base"
     assert img.steps == [("RUN", "echo hi"), ("ENV", "A=1")]
+
+
+# -- config layering ---------------------------------------------------------
+
+@SETTINGS
+@given(st.sampled_from(["namespace", "username", "image"]),
+       st.text(alphabet="abcdef012", min_size=1, max_size=8),
+       st.text(alphabet="abcdef012", min_size=1, max_size=8),
+       st.booleans())
+def test_config_env_beats_file(tmp_path_factory, key, envval, fileval,
+                               set_env):
+    """Layering invariant: env KT_* always wins over the config file; the
+    file wins over defaults."""
+    import yaml
+
+    from kubetorch_amd.config import KTConfig
+
+    path = str(tmp_path_factory.mktemp("cfg") / "config")
+    with open(path, "w") as f:
+        yaml.safe_dump({key: fileval}, f)
+    envkey = "KT_" + key.upper()
+    old = os.environ.pop(envkey, None)
+    try:
+        if set_env:
+            os.environ[envkey] = envval
+        cfg = KTConfig(path=path)
+        cfg._cluster = {}  # no cluster fetch in a unit test
+        assert cfg.get(key) == (envval if set_env else fileval)
+    finally:
+        if old is None:
+            os.environ.pop(envkey, None)
+        else:
+            os.environ[envkey] = old
+
+
+def test_config_cluster_below_file_above_defaults(tmp_path):
+    from kubetorch_amd.config import KTConfig
+
+    cfg = KTConfig(path=str(tmp_path / "missing"))
+    cfg._cluster = {"namespace": "from-cluster"}
+    old = os.environ.pop("KT_NAMESPACE", None)
+    try:
+        assert cfg.get("namespace") == "from-cluster"   # cluster > defaults
+        cfg.set("namespace", "from-file")
+        assert cfg.get("namespace") == "from-file"      # file > cluster
+    finally:
+        if old is not None:
+            os.environ["KT_NAMESPACE"] = old
