@@ -28,6 +28,9 @@ _LAZY = {
     "Secret": ("kubetorch_amd.resources.secret", "Secret"),
     "Endpoint": ("kubetorch_amd.resources.endpoint", "Endpoint"),
     "AutoscalingConfig": ("kubetorch_amd.resources.autoscaling", "AutoscalingConfig"),
+    "LoggingConfig": ("kubetorch_amd.workload_configs", "LoggingConfig"),
+    "MetricsConfig": ("kubetorch_amd.workload_configs", "MetricsConfig"),
+    "DebugConfig": ("kubetorch_amd.workload_configs", "DebugConfig"),
     "images": ("kubetorch_amd.resources.images", None),
     "compute": ("kubetorch_amd.resources.decorators", "compute"),
     "distribute": ("kubetorch_amd.resources.decorators", "distribute"),

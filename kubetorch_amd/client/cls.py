@@ -16,6 +16,10 @@ class Cls(Module):
         @functools.wraps(lambda: None)
         def remote_method(*args, **kwargs):
             opts = {}
+            if "kt_config" in kwargs:
+                from kubetorch_amd.workload_configs import expand_config
+
+                opts.update(expand_config(kwargs.pop("kt_config")))
             for key in ("workers", "restart_procs", "stream_logs",
                         "stream_metrics", "timeout", "serialization",
                         "debug"):

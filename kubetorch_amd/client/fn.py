@@ -8,6 +8,10 @@ class Fn(Module):
 
     def __call__(self, *args, **kwargs):
         opts = {}
+        if "kt_config" in kwargs:
+            from kubetorch_amd.workload_configs import expand_config
+
+            opts.update(expand_config(kwargs.pop("kt_config")))
         for key in ("workers", "restart_procs", "stream_logs",
                     "stream_metrics", "timeout", "serialization", "debug"):
             if f"kt_{key}" in kwargs:

@@ -177,3 +177,38 @@ def test_autoscale_distribute_mutually_exclusive():
         kt.Compute(cpus=1).distribute("pytorch", workers=2).autoscale(target=1)
     with pytest.raises(ValueError, match="mutually exclusive"):
         kt.Compute(cpus=1).autoscale(target=1).distribute("pytorch")
+
+
+def test_workload_configs_expand_and_flat_override():
+    """kt.LoggingConfig/MetricsConfig/DebugConfig bundle the per-call
+    options; flat kt_* kwargs win over a bundle (reference parity:
+    workload_configs public API)."""
+    import kubetorch_amd as kt
+    from kubetorch_amd.workload_configs import expand_config
+
+    assert expand_config(None) == {}
+    assert expand_config(kt.LoggingConfig()) == {"stream_logs": True}
+    assert expand_config(kt.LoggingConfig(stream_logs=False)) == \
+        {"stream_logs": False}
+    merged = expand_config([kt.MetricsConfig(), kt.DebugConfig()])
+    assert merged == {"stream_metrics": True, "debug": True}
+
+
+def test_workload_config_flows_through_call(monkeypatch):
+    import kubetorch_amd as kt
+    from kubetorch_amd.client.fn import Fn
+
+    seen = {}
+
+    def fake_call(self, args, kwargs, method=None, **opts):
+        seen.update(opts)
+        return "ok"
+
+    monkeypatch.setattr(Fn, "_call", fake_call)
+    f = Fn({"name": "x", "file_path": "", "rel_path": "",
+            "project_root": "."})
+    f(1, kt_config=[kt.LoggingConfig(stream_logs=False),
+                    kt.MetricsConfig()],
+      kt_stream_metrics=False)  # flat kwarg beats the bundle
+    assert seen["stream_logs"] is False
+    assert seen["stream_metrics"] is False
