@@ -105,12 +105,14 @@ async def _ttl_reaper(interval=None):
             last = w.get("updated", 0)
             try:
                 pods = HUB.driver.pods(name, ns)
-                if pods:
-                    async with httpx.AsyncClient(timeout=5) as client:
-                        r = await client.get(f"http://{pods[0]}/metrics")
-                    for line in r.text.splitlines():
-                        if line.startswith("kt_last_activity_timestamp"):
-                            last = max(last, float(line.split()[-1]))
+                # activity is per-pod: the service is idle only when EVERY
+                # pod is (one busy replica must keep the workload alive)
+                async with httpx.AsyncClient(timeout=5) as client:
+                    for pod in pods:
+                        r = await client.get(f"http://{pod}/metrics")
+                        for line in r.text.splitlines():
+                            if line.startswith("kt_last_activity_timestamp"):
+                                last = max(last, float(line.split()[-1]))
             except Exception:
                 continue
             if time.time() - last > ttl:
