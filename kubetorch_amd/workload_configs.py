@@ -19,8 +19,14 @@ from dataclasses import dataclass
 @dataclass
 class LoggingConfig:
     """Per-call log streaming (client tails the pod's captured stdout for
-    this request_id while the call runs)."""
+    this request_id while the call runs).
+
+    Fields mirror the reference constructor (globals.py LoggingConfig):
+    `level` filters streamed lines client-side; `include_events` controls
+    K8s launch-event streaming during `.to()`."""
     stream_logs: bool = True
+    level: str = "info"
+    include_events: bool = True
 
     def call_opts(self):
         return {"stream_logs": self.stream_logs}
@@ -29,8 +35,10 @@ class LoggingConfig:
 @dataclass
 class MetricsConfig:
     """Per-call hardware metric streaming (CPU/mem + amd-smi GPU util/VRAM
-    sampled from the pod's /metrics while the call runs)."""
+    sampled from the pod's /metrics while the call runs). `interval` is the
+    seconds between metric samples (reference: MetricsConfig.interval)."""
     stream_metrics: bool = True
+    interval: int = 30
 
     def call_opts(self):
         return {"stream_metrics": self.stream_metrics}
@@ -39,8 +47,12 @@ class MetricsConfig:
 @dataclass
 class DebugConfig:
     """Remote debugging: breakpoints hit during this call wait for
-    ``kt debug <service>`` to attach over the pod's debug port."""
+    ``kt debug <service>`` to attach over the pod's debug port
+    (reference: DebugConfig mode/port; mode is always the pdb-over-socket
+    transport here)."""
     debug: bool = True
+    mode: str = "pdb"
+    port: int = 5678
 
     def call_opts(self):
         return {"debug": self.debug}
