@@ -398,6 +398,44 @@ def tunnel(service: str, port: int = typer.Argument(8080),
         t.stop()
 
 
+@app.command()
+def dashboard(namespace: str = typer.Option(None, "-n"),
+              port: int = typer.Option(9090)):
+    """Open the cluster metrics dashboard (reference parity: kt dashboard).
+    On Kubernetes: port-forward the bundled Prometheus
+    (kubetorch-amd-metrics) and print its URL. On the local driver: print
+    the live pods' /metrics endpoints."""
+    from kubetorch_amd.config import config as cfg
+    from kubetorch_amd.globals import controller_client
+
+    ns = namespace or cfg.get("install_namespace") or cfg.namespace
+    cc = controller_client()
+    import httpx
+
+    try:
+        h = httpx.get(cc.base_url + "/health", timeout=10).json()
+    except Exception as e:
+        console.print(f"[red]controller unreachable:[/red] {e}")
+        raise typer.Exit(1)
+    if h.get("driver") == "local":
+        for w in cc.list_workloads(cfg.namespace).get("workloads", []):
+            full = cc.get_workload(w["name"], cfg.namespace) or {}
+            for pod in full.get("pods") or []:
+                console.print(f"{w['name']}: http://{pod}/metrics")
+        console.print("local driver: per-pod Prometheus text above "
+                      "(no cluster dashboard)")
+        return
+    from kubetorch_amd.globals import PortForward
+
+    pf = PortForward("svc/kubetorch-amd-metrics", ns, port, port).start()
+    console.print(f"[green]metrics dashboard[/green]: "
+                  f"http://127.0.0.1:{port} (ctrl-c to stop)")
+    try:
+        pf.proc.wait()
+    except KeyboardInterrupt:
+        pf.stop()
+
+
 @app.command("port-forward")
 def port_forward(target: str, ports: str,
                  namespace: str = typer.Option(None, "-n")):

@@ -155,3 +155,14 @@ def test_bench_verb_through_dispatch_stack():
     assert result["steps"] == 2
     assert result["value"] > 0
     assert result["metric"] == "tiny_ddp_tokens_per_sec"
+
+
+def test_dashboard_local_lists_metrics_endpoints():
+    f = kt.fn(summer_mod.summer).to(kt.Compute(cpus=1))
+    try:
+        r = runner.invoke(app, ["dashboard"])
+        assert r.exit_code == 0, r.output
+        assert "/metrics" in r.output
+        assert f.name in r.output
+    finally:
+        f.teardown()
