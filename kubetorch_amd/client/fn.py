@@ -24,6 +24,31 @@ class Fn(Module):
         return await asyncio.to_thread(self.__call__, *args, **kwargs)
 
 
-def fn(callable_obj, name=None):
-    """kt.fn(train) -> Fn proxy; deploy with .to(kt.Compute(...))."""
+def fn(callable_obj=None, name=None, remote_dir=None,
+       remote_import_path=None):
+    """kt.fn(train) -> Fn proxy; deploy with .to(kt.Compute(...)).
+
+    remote_dir mode (reference parity: fn(sync_dir/remote_dir/
+    remote_import_path)): dispatch to a function whose code is already on
+    the pod image — no client-side code sync. Pass
+    ``remote_import_path="pkg.module:func"`` and ``remote_dir="/app"``;
+    the pod imports /app/pkg/module.py and serves ``func``."""
+    if remote_dir is not None:
+        if not remote_import_path or ":" not in remote_import_path:
+            raise ValueError(
+                'remote_dir needs remote_import_path="pkg.module:func"')
+        module_path, _, func = remote_import_path.partition(":")
+        import os
+
+        rel = module_path.replace(".", "/") + ".py"
+        pointers = {
+            "name": func,
+            "file_path": os.path.join(remote_dir, rel),
+            "rel_path": rel,
+            "project_root": remote_dir,
+            "remote": True,  # code lives on the image: skip workdir sync
+        }
+        return Fn(pointers, name=name or func)
+    if callable_obj is None:
+        raise TypeError("kt.fn() needs a callable (or remote_dir=...)")
     return Fn(extract_pointers(callable_obj), name=name)

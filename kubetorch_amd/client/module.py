@@ -77,6 +77,8 @@ class Module:
         return md
 
     def _sync_workdir(self, md):
+        if self.pointers.get("remote"):
+            return  # remote_dir mode: the code is baked into the image
         if self.compute is not None and self.compute.local:
             return  # same filesystem: pods import the original paths
         try:

@@ -212,3 +212,34 @@ def test_workload_config_flows_through_call(monkeypatch):
       kt_stream_metrics=False)  # flat kwarg beats the bundle
     assert seen["stream_logs"] is False
     assert seen["stream_metrics"] is False
+
+
+def test_fn_remote_dir_mode():
+    """kt.fn(remote_dir=..., remote_import_path=...) dispatches to code
+    already present on the pod image — no client sync (reference:
+    fn(remote_dir/remote_import_path)). Simulated by pointing remote_dir
+    at a dir that exists on the 'image' (this host)."""
+    import os
+
+    import kubetorch_amd as kt
+
+    assets = os.path.join(os.path.dirname(__file__), "assets", "summer")
+    f = kt.fn(remote_dir=assets, remote_import_path="summer:summer",
+              name="baked")
+    assert f.pointers["file_path"] == os.path.join(assets, "summer.py")
+    f.to(kt.Compute(cpus=1))
+    try:
+        assert f(20, 22) == 42
+    finally:
+        f.teardown()
+
+
+def test_fn_remote_dir_requires_import_path():
+    import pytest as _pytest
+
+    import kubetorch_amd as kt
+
+    with _pytest.raises(ValueError, match="remote_import_path"):
+        kt.fn(remote_dir="/app")
+    with _pytest.raises(TypeError):
+        kt.fn()
