@@ -31,16 +31,43 @@ class Cls(Module):
         return remote_method
 
 
-def cls(class_obj, init_args=None, name=None):
+def _norm_init_args(init_args):
+    if init_args is None:
+        return None
+    is_pair = (isinstance(init_args, tuple) and len(init_args) == 2
+               and isinstance(init_args[0], (tuple, list))
+               and isinstance(init_args[1], dict))
+    if is_pair:
+        return init_args
+    if isinstance(init_args, (list, tuple)):
+        return (tuple(init_args), {})
+    return ((), dict(init_args))
+
+
+def cls(class_obj=None, init_args=None, name=None, remote_dir=None,
+        remote_import_path=None):
     """kt.cls(MyClass, init_args=((...), {...})) -> Cls proxy. The class is
-    instantiated once per worker process with init_args at load time."""
-    if init_args is not None:
-        is_pair = (isinstance(init_args, tuple) and len(init_args) == 2
-                   and isinstance(init_args[0], (tuple, list))
-                   and isinstance(init_args[1], dict))
-        if not is_pair:
-            if isinstance(init_args, (list, tuple)):
-                init_args = (tuple(init_args), {})
-            else:
-                init_args = ((), dict(init_args))
-    return Cls(extract_pointers(class_obj), name=name, init_args=init_args)
+    instantiated once per worker process with init_args at load time.
+    remote_dir mode mirrors kt.fn: serve a class already baked into the
+    image via remote_import_path="pkg.module:Class"."""
+    if remote_dir is not None:
+        if not remote_import_path or ":" not in remote_import_path:
+            raise ValueError(
+                'remote_dir needs remote_import_path="pkg.module:Class"')
+        import os
+
+        module_path, _, symbol = remote_import_path.partition(":")
+        rel = module_path.replace(".", "/") + ".py"
+        pointers = {
+            "name": symbol,
+            "file_path": os.path.join(remote_dir, rel),
+            "rel_path": rel,
+            "project_root": remote_dir,
+            "remote": True,
+        }
+        return Cls(pointers, name=name or symbol.lower(),
+                   init_args=_norm_init_args(init_args))
+    if class_obj is None:
+        raise TypeError("kt.cls() needs a class (or remote_dir=...)")
+    return Cls(extract_pointers(class_obj), name=name,
+               init_args=_norm_init_args(init_args))

@@ -243,3 +243,18 @@ def test_fn_remote_dir_requires_import_path():
         kt.fn(remote_dir="/app")
     with _pytest.raises(TypeError):
         kt.fn()
+
+
+def test_cls_remote_dir_mode():
+    import os
+
+    import kubetorch_amd as kt
+
+    assets = os.path.join(os.path.dirname(__file__), "assets", "summer")
+    c = kt.cls(remote_dir=assets, remote_import_path="summer:Counter",
+               init_args={"start": 5}, name="baked-counter")
+    c.to(kt.Compute(cpus=1))
+    try:
+        assert c.add(3) == 8
+    finally:
+        c.teardown()
