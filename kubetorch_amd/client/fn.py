@@ -24,7 +24,7 @@ class Fn(Module):
         return await asyncio.to_thread(self.__call__, *args, **kwargs)
 
 
-def fn(callable_obj=None, name=None, remote_dir=None,
+def fn(callable_obj=None, name=None, sync_dir=None, remote_dir=None,
        remote_import_path=None):
     """kt.fn(train) -> Fn proxy; deploy with .to(kt.Compute(...)).
 
@@ -33,6 +33,8 @@ def fn(callable_obj=None, name=None, remote_dir=None,
     the pod image — no client-side code sync. Pass
     ``remote_import_path="pkg.module:func"`` and ``remote_dir="/app"``;
     the pod imports /app/pkg/module.py and serves ``func``."""
+    if sync_dir and remote_dir:
+        raise ValueError("sync_dir and remote_dir are mutually exclusive")
     if remote_dir is not None:
         if not remote_import_path or ":" not in remote_import_path:
             raise ValueError(
@@ -51,4 +53,17 @@ def fn(callable_obj=None, name=None, remote_dir=None,
         return Fn(pointers, name=name or func)
     if callable_obj is None:
         raise TypeError("kt.fn() needs a callable (or remote_dir=...)")
-    return Fn(extract_pointers(callable_obj), name=name)
+    pointers = extract_pointers(callable_obj)
+    if sync_dir:
+        # sync THIS directory instead of the auto-detected project root
+        # (reference: Module sync_dir — the fn's file must live under it)
+        import os
+
+        root = os.path.abspath(sync_dir)
+        fp = os.path.abspath(pointers["file_path"])
+        if not fp.startswith(root + os.sep):
+            raise ValueError(
+                f"{pointers['name']} at {fp} is not under sync_dir {root}")
+        pointers["project_root"] = root
+        pointers["rel_path"] = os.path.relpath(fp, root)
+    return Fn(pointers, name=name)

@@ -258,3 +258,21 @@ def test_cls_remote_dir_mode():
         assert c.add(3) == 8
     finally:
         c.teardown()
+
+
+def test_fn_sync_dir_override():
+    """sync_dir narrows the synced project root to a chosen directory
+    (reference: Module sync_dir); the fn's file must live under it."""
+    import os
+
+    import kubetorch_amd as kt
+    from tests.assets.summer import summer as summer_mod
+
+    assets = os.path.join(os.path.dirname(__file__), "assets", "summer")
+    f = kt.fn(summer_mod.summer, sync_dir=assets)
+    assert f.pointers["project_root"] == os.path.abspath(assets)
+    assert f.pointers["rel_path"] == "summer.py"
+    with pytest.raises(ValueError, match="not under sync_dir"):
+        kt.fn(summer_mod.summer, sync_dir="/nonexistent-root")
+    with pytest.raises(ValueError, match="mutually exclusive"):
+        kt.fn(summer_mod.summer, sync_dir=assets, remote_dir="/app")
