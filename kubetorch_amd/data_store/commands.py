@@ -40,8 +40,9 @@ def _is_gpu_data(obj):
 
 
 def _local_path(key):
-    p = os.path.abspath(os.path.join(LOCAL_STORE_ROOT, key.strip("/")))
-    if not p.startswith(os.path.abspath(LOCAL_STORE_ROOT)):
+    root = os.path.abspath(LOCAL_STORE_ROOT)
+    p = os.path.abspath(os.path.join(root, key.strip("/")))
+    if p != root and not p.startswith(root + os.sep):
         raise ValueError("key escapes store root")
     return p
 

@@ -38,8 +38,10 @@ _LOG_RING = 50000
 
 
 def _path_for(key):
-    p = os.path.abspath(os.path.join(DATA_ROOT, key.strip("/")))
-    if not p.startswith(os.path.abspath(DATA_ROOT)):
+    root = os.path.abspath(DATA_ROOT)
+    p = os.path.abspath(os.path.join(root, key.strip("/")))
+    # separator-aware: "/data/storeX" must not pass a "/data/store" root
+    if p != root and not p.startswith(root + os.sep):
         raise ValueError("key escapes store root")
     return p
 
@@ -114,7 +116,7 @@ async def delete_files(key: str, request: Request):
     base = _path_for(key)
     for rel in body.get("paths", []):
         p = os.path.abspath(os.path.join(base, rel))
-        if p.startswith(base) and os.path.exists(p):
+        if (p == base or p.startswith(base + os.sep)) and os.path.isfile(p):
             os.remove(p)
     return {"ok": True}
 

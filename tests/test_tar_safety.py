@@ -77,3 +77,18 @@ def test_setuid_bit_stripped(tmp_path):
     with _tar_with([_file("s.bin", mode=0o4755)]) as tar:
         safe_extractall(tar, str(tmp_path))
     assert (os.stat(tmp_path / "s.bin").st_mode & 0o7777) == 0o755
+
+
+def test_store_path_guard_separator_aware(tmp_path, monkeypatch):
+    """'/data/storeX' must not pass a '/data/store' root (prefix check
+    without the separator)."""
+    from kubetorch_amd.data_store import server
+
+    monkeypatch.setattr(server, "DATA_ROOT", str(tmp_path / "store"))
+    import pytest as _pytest
+
+    assert server._path_for("ns/key").startswith(str(tmp_path / "store"))
+    with _pytest.raises(ValueError):
+        server._path_for("../storeX/evil")
+    with _pytest.raises(ValueError):
+        server._path_for("../../etc/passwd")
