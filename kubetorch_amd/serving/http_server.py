@@ -519,14 +519,27 @@ def _install_sigterm_drain():
     TerminationCheckMiddleware, http_server.py:1184-1236)."""
     import signal
 
-    def handler(signum, frame):
-        STATE["terminating"] = True
+    def _drain_then_exit():
+        # runs OFF the main thread: the asyncio loop must stay free to
+        # deliver the in-flight responses we are waiting for
+        drain_s = float(os.environ.get("KT_TERM_DRAIN_S", "25"))
+        deadline = time.time() + drain_s
+        while time.time() < deadline and \
+                metrics.ACTIVE_REQUESTS._value.get() > 0:
+            time.sleep(0.2)
         if STATE["supervisor"] is not None:
             try:
                 STATE["supervisor"].cleanup()
             except Exception:
                 pass
-        raise SystemExit(0)
+        os._exit(0)
+
+    def handler(signum, frame):
+        # flip readiness FIRST (new traffic stops routing here), then let
+        # in-flight calls finish before tearing the pool down — a worker
+        # killed mid-request fails the caller for no reason
+        STATE["terminating"] = True
+        threading.Thread(target=_drain_then_exit, daemon=True).start()
 
     try:
         signal.signal(signal.SIGTERM, handler)

@@ -386,3 +386,46 @@ def test_concurrent_deploys_same_name():
         assert len(HUB.driver.pods(fns[0].name, "default")) == 1
     finally:
         fns[0].teardown()
+
+
+def test_sigterm_drains_inflight_call():
+    """SIGTERM on a pod mid-call: readiness flips 503 immediately but the
+    in-flight request finishes before the pool is torn down (graceful
+    drain — reference: TerminationCheckMiddleware)."""
+    import signal
+    import threading
+
+    import httpx
+
+    from kubetorch_amd.controller.app import HUB
+
+    f = kt.fn(summer_mod.slow_echo).to(kt.Compute(cpus=1))
+    try:
+        assert f("warm", delay=0, kt_timeout=60) == "warm"
+        result = {}
+
+        def call():
+            try:
+                result["value"] = f("drained", kt_timeout=120, delay=5)
+            except Exception as e:  # noqa: BLE001
+                result["error"] = e
+
+        t = threading.Thread(target=call)
+        t.start()
+        time.sleep(1.5)  # request in flight inside the worker
+        pods = HUB.driver.services[("default", f.name)]
+        pods[0].proc.send_signal(signal.SIGTERM)
+        # readiness flips while the call drains
+        host = f.service_hosts[0]
+        deadline = time.time() + 10
+        ready = 200
+        while time.time() < deadline and ready == 200:
+            try:
+                ready = httpx.get(f"http://{host}/ready", timeout=3).status_code
+            except httpx.HTTPError:
+                break
+            time.sleep(0.2)
+        t.join(60)
+        assert result.get("value") == "drained", result.get("error")
+    finally:
+        f.teardown()
