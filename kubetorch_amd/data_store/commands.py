@@ -138,7 +138,16 @@ def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True,
     locale="local" registers the key WITHOUT copying (zero-copy): this
     pod serves the file on demand and getters fetch it p2p, falling back
     to the store (and de-registering the source) if this pod dies.
-    (Reference parity: data_store_client.py local locale, :175/:325.)"""
+    (Reference parity: data_store_client.py local locale, :175/:325.)
+
+    Batch mode (reference: put(key=[...], src=[...])): lists of keys and
+    sources are stored pairwise."""
+    if isinstance(key, (list, tuple)):
+        srcs = src if isinstance(src, (list, tuple)) else [src] * len(key)
+        if len(srcs) != len(key):
+            raise ValueError(f"{len(key)} keys but {len(srcs)} sources")
+        return [put(k, s, window=window, timeout=timeout, _delta=_delta,
+                    locale=locale) for k, s in zip(key, srcs)]
     if _is_gpu_data(src):
         from kubetorch_amd.data_store import gpu_store
 

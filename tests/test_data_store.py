@@ -487,3 +487,24 @@ def test_fs_broadcast_concurrent(http_store, tmp_store):
         if srv is not None:
             srv.close()
             fileserve._server = None
+
+
+def test_put_batch_keys(tmp_path):
+    """Batch mode: put(key=[...], src=[...]) stores pairwise (reference:
+    data_store_cmds.put list keys)."""
+    import kubetorch_amd as kt
+
+    a = tmp_path / "a.txt"; a.write_text("A")
+    b = tmp_path / "b.txt"; b.write_text("B")
+    out = kt.put(["batch/ka", "batch/kb"], [str(a), str(b)])
+    assert len(out) == 2
+    da = tmp_path / "outa"; db = tmp_path / "outb"
+    kt.get("batch/ka", str(da)); kt.get("batch/kb", str(db))
+    ra = da / "a.txt" if (da / "a.txt").exists() else da
+    rb = db / "b.txt" if (db / "b.txt").exists() else db
+    assert ra.read_text() == "A" and rb.read_text() == "B"
+    kt.rm("batch/ka"); kt.rm("batch/kb")
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="keys but"):
+        kt.put(["k1", "k2"], [str(a)])
