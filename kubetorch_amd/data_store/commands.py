@@ -196,9 +196,31 @@ def put(key, src, window=None, timeout=C.HTTP_TIMEOUT, _delta=True,
     return {"key": key, "locale": "store"}
 
 
-def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT):
+def get(key, dest=None, window=None, timeout=C.HTTP_TIMEOUT,
+        contents=False):
     """Fetch a stored file/dir to `dest` path, or receive GPU tensors into
-    `dest` tensors/state dict."""
+    `dest` tensors/state dict.
+
+    Batch mode (reference: get(key=[...])): a list of keys fetches each
+    (dest may be a matching list or None). `contents=True` returns a
+    single file's bytes directly instead of writing a path."""
+    if isinstance(key, (list, tuple)):
+        dests = dest if isinstance(dest, (list, tuple)) else [dest] * len(key)
+        if len(dests) != len(key):
+            raise ValueError(f"{len(key)} keys but {len(dests)} dests")
+        return [get(k, d, window=window, timeout=timeout, contents=contents)
+                for k, d in zip(key, dests)]
+    if contents:
+        import tempfile
+
+        with tempfile.TemporaryDirectory() as td:
+            out = get(key, os.path.join(td, "f"), window=window,
+                      timeout=timeout)
+            if os.path.isdir(out):
+                raise ValueError(
+                    f"contents=True needs a file key; {key!r} is a dir")
+            with open(out, "rb") as f:
+                return f.read()
     if dest is not None and _is_gpu_data(dest):
         from kubetorch_amd.data_store import gpu_store
 

@@ -508,3 +508,15 @@ def test_put_batch_keys(tmp_path):
 
     with _pytest.raises(ValueError, match="keys but"):
         kt.put(["k1", "k2"], [str(a)])
+
+
+def test_get_batch_and_contents(tmp_path):
+    import kubetorch_amd as kt
+
+    a = tmp_path / "x.txt"; a.write_text("XYZ")
+    kt.put("gb/k1", str(a)); kt.put("gb/k2", str(a))
+    outs = kt.get(["gb/k1", "gb/k2"],
+                  [str(tmp_path / "o1"), str(tmp_path / "o2")])
+    assert len(outs) == 2
+    assert kt.get("gb/k1", contents=True) == b"XYZ"
+    kt.rm("gb/k1"); kt.rm("gb/k2")
