@@ -355,7 +355,13 @@ def ls(prefix=""):
     return r.json()["entries"]
 
 
-def rm(key):
+def rm(key, prefix=False):
+    """Delete a key (dirs recurse). prefix=True deletes every key under
+    the given prefix (reference: rm(prefix=) bulk cleanup)."""
+    if prefix:
+        for e in ls(key):
+            rm(e["key"])
+        return {"ok": True, "prefix": key}
     url = _store_url()
     if url is None:
         p = _local_path(key)

@@ -520,3 +520,14 @@ def test_get_batch_and_contents(tmp_path):
     assert len(outs) == 2
     assert kt.get("gb/k1", contents=True) == b"XYZ"
     kt.rm("gb/k1"); kt.rm("gb/k2")
+
+
+def test_rm_prefix_bulk(tmp_path):
+    import kubetorch_amd as kt
+
+    a = tmp_path / "x.txt"; a.write_text("1")
+    kt.put("bulk/a", str(a)); kt.put("bulk/b", str(a)); kt.put("keepme/c", str(a))
+    kt.rm("bulk", prefix=True)
+    assert not [e for e in kt.ls("") if e["key"].startswith("bulk")]
+    assert [e for e in kt.ls("keepme")]
+    kt.rm("keepme", prefix=True)
