@@ -515,8 +515,15 @@ def app_status():
 
 def _install_sigterm_drain():
     """SIGTERM: mark terminating (readiness flips 503) and let in-flight
-    requests drain before uvicorn exits (reference:
-    TerminationCheckMiddleware, http_server.py:1184-1236)."""
+    requests drain before exiting (reference: TerminationCheckMiddleware,
+    http_server.py:1184-1236).
+
+    Once uvicorn's event loop is up it installs its own SIGTERM handler,
+    whose graceful shutdown (stop accepting -> wait for in-flight
+    responses -> run lifespan teardown, which cleans up the supervisor)
+    gives the same drain semantics — validated by the kill-mid-call e2e
+    drill. This handler covers the window before uvicorn starts and
+    non-uvicorn embeddings of the app."""
     import signal
 
     def _drain_then_exit():
