@@ -12,8 +12,17 @@ def train(steps: int = 50):
     from kubetorch_amd.parallel import FlatDDP, init_distributed
 
     rank, world, local_rank = init_distributed()
-    dev = torch.device("cuda", local_rank)
-    cfg = llama3_8b(max_seq_len=4096)
+    # production shape on MI355X; degrades to a tiny model on gloo/CPU so
+    # the same file demos the full path without a GPU
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        dev = torch.device("cuda", local_rank)
+        cfg = llama3_8b(max_seq_len=4096)
+    else:
+        from kubetorch_amd.models import llama_tiny
+
+        dev = torch.device("cpu")
+        cfg = llama_tiny(max_seq_len=256)
     prev = torch.get_default_dtype()
     torch.set_default_dtype(torch.bfloat16)
     with torch.device(dev):
@@ -25,9 +34,11 @@ def train(steps: int = 50):
     from kubetorch_amd.data import ShardedLoader, TokenDataset, synthetic_tokens
     from kubetorch_amd.parallel import warmup_cosine
 
-    ds = TokenDataset(synthetic_tokens(cfg.vocab_size, 4096 * 512), 4096)
-    loader = ShardedLoader(ds, batch=4, rank=rank, world=world,
-                           device=dev)  # pinned one-ahead H2D prefetch
+    seq = cfg.max_seq_len
+    ds = TokenDataset(synthetic_tokens(cfg.vocab_size, seq * (512 if on_gpu
+                                                              else 16)), seq)
+    loader = ShardedLoader(ds, batch=4 if on_gpu else 1, rank=rank,
+                           world=world, device=dev)  # pinned H2D prefetch
     it = iter(loader)
     for step in range(steps):
         try:
@@ -48,8 +59,14 @@ def train(steps: int = 50):
 
 
 if __name__ == "__main__":
-    remote = kt.fn(train).to(
-        kt.Compute(gpus=8, memory="640Gi").distribute("pytorch", workers=2)
-    )
-    results = remote(50, kt_timeout=3600)   # 2 pods x 8 ranks
+    import torch
+
+    if torch.cuda.is_available():
+        compute = kt.Compute(gpus=8, memory="640Gi").distribute(
+            "pytorch", workers=2)     # 2 pods x 8 ranks over RCCL/xGMI
+    else:
+        compute = kt.Compute(cpus=1).distribute(
+            "pytorch", workers=2, num_proc=1)  # local demo: 2 ranks, gloo
+    remote = kt.fn(train).to(compute)
+    results = remote(50, kt_timeout=3600)
     print(results)
