@@ -271,6 +271,7 @@ def test_examples_run():
                    os.path.abspath(__file__))))
     root = env["PYTHONPATH"]
     for ex, needle in (("01_hello_world.py", "hello"),
+                       ("02_ddp_train.py", "final_loss"),
                        ("05_serving_decode.py", "generated:"),
                        ("08_production_surface.py", "'image_env': 'prod'")):
         r = subprocess.run(
