@@ -68,5 +68,8 @@ if __name__ == "__main__":
         compute = kt.Compute(cpus=1).distribute(
             "pytorch", workers=2, num_proc=1)  # local demo: 2 ranks, gloo
     remote = kt.fn(train).to(compute)
-    results = remote(50, kt_timeout=3600)
+    import os
+
+    steps = int(os.environ.get("KT_EX_STEPS", "50"))
+    results = remote(steps, kt_timeout=3600)
     print(results)

@@ -267,6 +267,7 @@ def test_examples_run():
     import subprocess
 
     env = dict(os.environ, KT_LOCAL_MODE="true", KT_USERNAME="exs",
+               KT_EX_STEPS="5",
                PYTHONPATH=os.path.dirname(os.path.dirname(
                    os.path.abspath(__file__))))
     root = env["PYTHONPATH"]
