@@ -266,6 +266,15 @@ from contextlib import asynccontextmanager
 
 @asynccontextmanager
 async def _lifespan(app):
+    if hasattr(HUB.driver, "load_workloads"):
+        # controller restart: rehydrate the registry from the durable
+        # KubetorchWorkload CRs (reference parity: the CRD IS the registry)
+        try:
+            for w in await asyncio.to_thread(HUB.driver.load_workloads):
+                HUB.workloads.setdefault(
+                    (w["namespace"], w["name"]), w)
+        except Exception:
+            pass
     task = asyncio.create_task(_ttl_reaper())
     mon = asyncio.create_task(_pod_monitor())
     scaler = asyncio.create_task(_autoscaler())
@@ -343,6 +352,14 @@ async def deploy(request: Request):
             return JSONResponse(
                 {"error": f"driver apply failed: {e}"}, status_code=500
             )
+        if hasattr(HUB.driver, "persist_workload"):
+            # durable registry: a controller restart rehydrates from the
+            # KubetorchWorkload CRs (best-effort — CRD may not be installed)
+            try:
+                await asyncio.to_thread(
+                    HUB.driver.persist_workload, HUB.workloads[key])
+            except Exception:
+                pass
         if HUB.driver_name == "k8s" and manifest.get("kind") == "Deployment":
             # route + discovery Services (the reference controller creates
             # these; a custom endpoint selector narrows call routing)
@@ -582,6 +599,11 @@ async def delete_workload(ns: str, name: str):
         await asyncio.to_thread(HUB.driver.delete, name, ns)
     except Exception:
         pass
+    if hasattr(HUB.driver, "delete_workload_cr"):
+        try:
+            await asyncio.to_thread(HUB.driver.delete_workload_cr, name, ns)
+        except Exception:
+            pass
     return {"ok": True}
 
 

@@ -400,6 +400,65 @@ class K8sDriver:
             check=True, capture_output=True,
         )
 
+    # -- KubetorchWorkload CRs: durable workload registry ---------------------
+    # The reference controller persists its registry as KubetorchWorkload
+    # objects so a controller restart loses nothing; HUB rehydrates from
+    # these at startup (charts/.../crds.yaml is the schema).
+    def persist_workload(self, w):
+        md = w.get("metadata") or {}
+        cr = {
+            "apiVersion": "kubetorch.amd.com/v1",
+            "kind": "KubetorchWorkload",
+            "metadata": {
+                "name": w["name"], "namespace": w["namespace"],
+                "labels": {C.SERVICE_LABEL: w["name"]},
+                # full registry record (schema-free) for exact rehydration
+                "annotations": {C.LABEL_PREFIX + "/state": json.dumps(w)},
+            },
+            "spec": {
+                "serviceConfig": {
+                    "kind": (w.get("service_config") or {}).get("kind", ""),
+                    "name": w["name"],
+                },
+                "module": {
+                    "type": md.get("module_type", "fn"),
+                    "dispatch": ("spmd" if md.get("distributed_config")
+                                 else "regular"),
+                    "pointers": {k: str(md[k]) for k in
+                                 ("callable_name", "file_path",
+                                  "project_root") if md.get(k)},
+                },
+            },
+        }
+        r = subprocess.run(
+            [self.kubectl, "-n", w["namespace"], "apply", "-f", "-"],
+            input=json.dumps(cr).encode(), capture_output=True)
+        if r.returncode != 0:
+            raise RuntimeError(r.stderr.decode(errors="replace")[-500:])
+
+    def load_workloads(self):
+        """Rehydrate the registry records from KubetorchWorkload CRs."""
+        out = subprocess.run(
+            [self.kubectl, "get", "kubetorchworkloads", "-A", "-o", "json"],
+            capture_output=True)
+        if out.returncode != 0:
+            return []
+        records = []
+        for it in json.loads(out.stdout or b"{}").get("items", []):
+            raw = (it.get("metadata", {}).get("annotations", {})
+                   .get(C.LABEL_PREFIX + "/state"))
+            if raw:
+                try:
+                    records.append(json.loads(raw))
+                except ValueError:
+                    pass
+        return records
+
+    def delete_workload_cr(self, name, namespace):
+        subprocess.run(
+            [self.kubectl, "-n", namespace, "delete", "kubetorchworkload",
+             name, "--ignore-not-found"], capture_output=True)
+
     def pods(self, name, namespace):
         out = subprocess.run(
             [self.kubectl, "-n", namespace, "get", "pods", "-l",

@@ -63,6 +63,16 @@ def main():
         print(json.dumps({"items": items}))
         return 0
 
+    if verb == "get" and args[1] == "kubetorchworkloads":
+        # latest apply wins per (ns, name)
+        latest = {}
+        for rec in load("applied.jsonl"):
+            m = rec["manifest"]
+            if m.get("kind") == "KubetorchWorkload":
+                latest[(rec["ns"], m["metadata"]["name"])] = m
+        print(json.dumps({"items": list(latest.values())}))
+        return 0
+
     if verb == "get" and args[1] == "secrets":
         items = [rec["manifest"] for rec in load("applied.jsonl")
                  if rec["manifest"].get("kind") == "Secret" and rec["ns"] == ns]

@@ -76,3 +76,33 @@ def test_delete_and_pods(driver):
     drv.delete("svc-ok", "default")
     deleted = [json.loads(l) for l in open(state / "deleted.jsonl")]
     assert deleted[-1]["args"][1] == "svc-ok"
+
+
+def test_workload_cr_persist_and_rehydrate(driver):
+    """Deploys persist a KubetorchWorkload CR; a fresh controller registry
+    rehydrates from them (durable across controller restarts)."""
+    drv, state = driver
+    w = {"name": "svc-a", "namespace": "default",
+         "manifest": _deployment("svc-a"),
+         "metadata": {"module_type": "fn", "callable_name": "train",
+                      "file_path": "/w/train.py", "project_root": "/w",
+                      "distributed_config": {"type": "pytorch",
+                                             "workers": 2}},
+         "service_config": {"kind": "deployment"}, "launch_id": "lid-1",
+         "created": 1.0, "updated": 2.0}
+    drv.persist_workload(w)
+    applied = [json.loads(l) for l in
+               open(state / "applied.jsonl") if l.strip()]
+    crs = [a["manifest"] for a in applied
+           if a["manifest"].get("kind") == "KubetorchWorkload"]
+    assert len(crs) == 1
+    assert crs[0]["spec"]["module"]["dispatch"] == "spmd"
+    assert crs[0]["spec"]["module"]["pointers"]["callable_name"] == "train"
+
+    records = drv.load_workloads()
+    assert records == [w]
+
+    drv.delete_workload_cr("svc-a", "default")
+    deleted = [json.loads(l) for l in
+               open(state / "deleted.jsonl") if l.strip()]
+    assert any("kubetorchworkload" in d["args"] for d in deleted)
