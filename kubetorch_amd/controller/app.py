@@ -575,6 +575,12 @@ async def register_workload(request: Request):
         "launch_id": body.get("launch_id"),
         "created": time.time(), "updated": time.time(),
     }
+    if hasattr(HUB.driver, "persist_workload"):
+        try:
+            await asyncio.to_thread(
+                HUB.driver.persist_workload, HUB.workloads[(ns, name)])
+        except Exception:
+            pass
     return {"ok": True}
 
 
