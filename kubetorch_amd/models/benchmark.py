@@ -8,7 +8,6 @@ Rank/world comes from the standard env contract either way; the SPMD
 supervisor sets it for launched workers, torch.distributed.run for direct
 bench.py runs.
 """
-import json
 import os
 import sys
 import time

@@ -6,7 +6,6 @@ import subprocess
 
 from prometheus_client.core import GaugeMetricFamily
 
-from kubetorch_amd.serving.metrics import REGISTRY
 
 
 class AMDGPUCollector:

@@ -12,14 +12,10 @@ import json
 import os
 import pickle
 import socket
-import threading
 from concurrent.futures import FIRST_EXCEPTION, ThreadPoolExecutor, wait
 
 from kubetorch_amd import constants as C
-from kubetorch_amd.exceptions import (
-    WorkerMembershipChanged,
-    reconstruct_exception,
-)
+from kubetorch_amd.exceptions import reconstruct_exception
 from kubetorch_amd.serving import discovery
 from kubetorch_amd.serving.process_pool import ProcessPool
 
