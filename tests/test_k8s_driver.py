@@ -106,3 +106,32 @@ def test_workload_cr_persist_and_rehydrate(driver):
     deleted = [json.loads(l) for l in
                open(state / "deleted.jsonl") if l.strip()]
     assert any("kubetorchworkload" in d["args"] for d in deleted)
+
+
+def test_controller_rehydrates_registry_on_startup(driver, monkeypatch):
+    """Controller restart: HUB.workloads is rebuilt from the persisted
+    KubetorchWorkload CRs during lifespan startup."""
+    from fastapi.testclient import TestClient
+
+    from kubetorch_amd.controller import app as capp
+
+    drv, _state = driver
+    w = {"name": "revived", "namespace": "rehydrate-ns",
+         "manifest": _deployment("revived"),
+         "metadata": {"module_type": "fn", "callable_name": "f"},
+         "service_config": {"kind": "deployment"}, "launch_id": "lid-9",
+         "created": 1.0, "updated": 2.0}
+    drv.persist_workload(w)
+
+    monkeypatch.setattr(capp.HUB, "driver", drv)
+    monkeypatch.setattr(capp.HUB, "driver_name", "k8s")
+    try:
+        with TestClient(capp.app) as c:
+            r = c.get("/controller/workloads/rehydrate-ns").json()
+            names = [x["name"] for x in r["workloads"]]
+            assert "revived" in names
+            full = c.get(
+                "/controller/workload/rehydrate-ns/revived").json()
+            assert full["launch_id"] == "lid-9"
+    finally:
+        capp.HUB.workloads.pop(("rehydrate-ns", "revived"), None)
