@@ -65,5 +65,9 @@ def expand_config(cfg):
     items = cfg if isinstance(cfg, (list, tuple)) else [cfg]
     opts = {}
     for c in items:
+        if not hasattr(c, "call_opts"):
+            raise TypeError(
+                f"kt_config takes LoggingConfig/MetricsConfig/DebugConfig "
+                f"(or a list of them), got {type(c).__name__}")
         opts.update(c.call_opts())
     return opts
