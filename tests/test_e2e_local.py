@@ -431,3 +431,27 @@ def test_sigterm_drains_inflight_call():
         assert result.get("value") == "drained", result.get("error")
     finally:
         f.teardown()
+
+
+def test_ten_services_parallel_lifecycle():
+    """Controller scale drill: 10 services deployed in parallel, called in
+    parallel, all registered, all torn down — no cross-talk between
+    registries, ports, or pools."""
+    import concurrent.futures as cf
+
+    fns = [kt.fn(summer_mod.summer, name=f"mini{i}") for i in range(10)]
+    try:
+        with cf.ThreadPoolExecutor(10) as ex:
+            list(ex.map(lambda f: f.to(kt.Compute(cpus=1)), fns))
+        with cf.ThreadPoolExecutor(10) as ex:
+            res = list(ex.map(lambda p: p[1](p[0], p[0]),
+                              list(enumerate(fns))))
+        assert res == [2 * i for i in range(10)]
+        from kubetorch_amd.globals import controller_client
+
+        ws = controller_client().list_workloads("default")["workloads"]
+        mine = [w["name"] for w in ws if w["name"].startswith("citest-mini")]
+        assert len(mine) == 10
+    finally:
+        with cf.ThreadPoolExecutor(10) as ex:
+            list(ex.map(lambda f: f.teardown(), fns))
