@@ -450,7 +450,8 @@ def test_ten_services_parallel_lifecycle():
         from kubetorch_amd.globals import controller_client
 
         ws = controller_client().list_workloads("default")["workloads"]
-        mine = [w["name"] for w in ws if w["name"].startswith("citest-mini")]
+        prefix = fns[0].name.rsplit("mini", 1)[0] + "mini"
+        mine = [w["name"] for w in ws if w["name"].startswith(prefix)]
         assert len(mine) == 10
     finally:
         with cf.ThreadPoolExecutor(10) as ex:
