@@ -225,9 +225,11 @@ def test_get_if_exists_reuses_service(remote_fn):
     Module.to(get_if_exists=...) with username -> branch -> prod order)."""
     from kubetorch_amd.client.fn import fn as make_fn
 
+    from kubetorch_amd.config import config as _cfg
+
     f2 = make_fn(summer_mod.summer)
     f2.to(kt.Compute(cpus=1), get_if_exists=True,
-          reload_prefixes=["nosuch", "citest"])
+          reload_prefixes=["nosuch", _cfg.username])
     try:
         assert f2.name == remote_fn.name  # bound, not re-prefixed
         assert f2.service_hosts, "expected the existing service's pods"
